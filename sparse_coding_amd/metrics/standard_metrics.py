@@ -1,0 +1,350 @@
+"""Evaluation metrics suite (parity with reference standard_metrics.py, C20).
+
+Implements: FVU (:310), top-k FVU split (:316), r² (:344), mean nonzero
+activations (:305), ever-active / dead counts (:441-454), streaming moments
+(:456-511), the MMCS family (:270-303), Hungarian-matched MMCS (:811-842),
+neurons-per-feature Simpson index (:347), capacity (Scherlis et al., :354),
+logistic/ridge AUROC probes (:254-268), k-means/hierarchical clustering of
+directions (:534-579), plot helpers, and perplexity-under-reconstruction
+(:621-709 — via the HF hooked model of sparse_coding_amd.data.activation_dataset
+instead of TransformerLens).
+"""
+
+from __future__ import annotations
+
+from typing import Any, Dict, List, Optional, Tuple
+
+import numpy as np
+import torch
+
+from sparse_coding_amd.models.learned_dict import LearnedDict
+
+# ---------------------------------------------------------------------------
+# cosine-similarity family
+# ---------------------------------------------------------------------------
+
+def mcs_duplicates(ground: LearnedDict, model: LearnedDict) -> torch.Tensor:
+    """Max cosine sim of each `model` atom against all `ground` atoms."""
+    cos = model.get_learned_dict() @ ground.get_learned_dict().T
+    return cos.max(dim=-1).values
+
+
+def mmcs(model: LearnedDict, model2: LearnedDict) -> torch.Tensor:
+    return mcs_duplicates(model, model2).mean()
+
+
+def mcs_to_fixed(model: LearnedDict, truth: torch.Tensor) -> torch.Tensor:
+    cos = model.get_learned_dict() @ truth.T
+    return cos.max(dim=-1).values
+
+
+def mmcs_to_fixed(model: LearnedDict, truth: torch.Tensor) -> torch.Tensor:
+    return mcs_to_fixed(model, truth).mean()
+
+
+def mmcs_from_list(ld_list: List[LearnedDict]) -> torch.Tensor:
+    n = len(ld_list)
+    out = torch.eye(n)
+    for i in range(n):
+        for j in range(i):
+            out[i, j] = out[j, i] = mmcs(ld_list[i], ld_list[j])
+    return out
+
+
+def representedness(features: torch.Tensor, model: LearnedDict) -> torch.Tensor:
+    cos = features @ model.get_learned_dict().T
+    return cos.max(dim=-1).values
+
+
+def hungarian_mmcs(ground: torch.Tensor, learned: torch.Tensor) -> torch.Tensor:
+    """One-to-one (Hungarian) matched mean cosine sim, reference :811-842."""
+    from scipy.optimize import linear_sum_assignment
+
+    g = ground / torch.clamp(torch.norm(ground, dim=-1, keepdim=True), 1e-8)
+    l = learned / torch.clamp(torch.norm(learned, dim=-1, keepdim=True), 1e-8)
+    cos = (g @ l.T).cpu().numpy()
+    row, col = linear_sum_assignment(-cos)
+    return torch.tensor(cos[row, col]).mean()
+
+
+# ---------------------------------------------------------------------------
+# sparsity / variance metrics
+# ---------------------------------------------------------------------------
+
+def mean_nonzero_activations(model: LearnedDict, batch: torch.Tensor) -> torch.Tensor:
+    c = model.encode(model.center(batch))
+    return (c != 0).float().mean(dim=0)
+
+
+def mean_l0(model: LearnedDict, batch: torch.Tensor) -> torch.Tensor:
+    """Mean number of active features per example."""
+    c = model.encode(model.center(batch))
+    return (c != 0).float().sum(dim=-1).mean()
+
+
+def fraction_variance_unexplained(model: LearnedDict, batch: torch.Tensor) -> torch.Tensor:
+    x_hat = model.predict(batch)
+    residuals = (batch - x_hat).pow(2).mean()
+    total = (batch - batch.mean(dim=0)).pow(2).mean()
+    return residuals / total
+
+
+def fraction_variance_unexplained_top_activating(
+    model: LearnedDict, batch: torch.Tensor, n_top: int = 2
+) -> Tuple[torch.Tensor, torch.Tensor]:
+    c = model.encode(model.center(batch))
+    order = torch.argsort(c.mean(dim=0), descending=True)
+    top_idx, rest_idx = order[:n_top], order[n_top:]
+
+    c_top = torch.zeros_like(c)
+    c_top[:, top_idx] = c[:, top_idx]
+    c_rest = torch.zeros_like(c)
+    c_rest[:, rest_idx] = c[:, rest_idx]
+
+    x_hat_top = model.center(model.decode(c_top))
+    x_hat_rest = model.center(model.decode(c_rest))
+    var = (batch - batch.mean(dim=0)).pow(2).mean()
+    return (batch - x_hat_top).pow(2).mean() / var, (batch - x_hat_rest).pow(2).mean() / var
+
+
+def r_squared(model: LearnedDict, batch: torch.Tensor) -> torch.Tensor:
+    return 1.0 - fraction_variance_unexplained(model, batch)
+
+
+def neurons_per_feature(model: LearnedDict) -> torch.Tensor:
+    """Simpson-diversity count of neurons per learned direction (:347)."""
+    c = model.get_learned_dict()
+    c = c / c.abs().sum(dim=-1, keepdim=True)
+    return (1.0 / c.pow(2).sum(dim=-1)).mean()
+
+
+def capacity_per_feature(model: LearnedDict) -> torch.Tensor:
+    """Scherlis et al. 2022 capacities (:354-362)."""
+    d = model.get_learned_dict()
+    sq = (d @ d.T).pow(2)
+    return torch.diag(sq) / sq.sum(dim=-1)
+
+
+def calc_expected_interference(dictionary: torch.Tensor, batch: torch.Tensor) -> torch.Tensor:
+    """Per-feature capacity weighted by usage (reference big_sweep.py:43-57)."""
+    normed = dictionary / torch.clamp(torch.norm(dictionary, 2, dim=-1), 1e-8)[:, None]
+    cos2 = (normed @ normed.T).pow(2)
+    totals = batch @ cos2.T
+    capacities = batch / torch.clamp(totals, min=1e-8)
+    nonzero_count = batch.count_nonzero(dim=0).float()
+    return capacities.sum(dim=0) / torch.clamp(nonzero_count, min=1.0)
+
+
+# ---------------------------------------------------------------------------
+# activity counts & streaming moments
+# ---------------------------------------------------------------------------
+
+def calc_feature_n_active(batch: torch.Tensor) -> torch.Tensor:
+    return torch.sum(batch != 0, dim=0)
+
+
+def batched_calc_feature_n_ever_active(
+    model: LearnedDict, activations: torch.Tensor, batch_size: int = 1000, threshold: int = 10
+) -> int:
+    counts = torch.zeros(model.n_feats, device=activations.device)
+    for i in range(0, len(activations), batch_size):
+        counts += calc_feature_n_active(model.encode(activations[i : i + batch_size]))
+    return int((counts > threshold).sum().item())
+
+
+def dead_feature_fraction(model: LearnedDict, activations: torch.Tensor, batch_size: int = 1000) -> float:
+    """Fraction of features that never fire on `activations`."""
+    counts = torch.zeros(model.n_feats, device=activations.device)
+    for i in range(0, len(activations), batch_size):
+        counts += calc_feature_n_active(model.encode(activations[i : i + batch_size]))
+    return float((counts == 0).float().mean().item())
+
+
+def calc_feature_mean(batch):
+    return batch.mean(dim=0)
+
+
+def calc_feature_variance(batch):
+    return batch.var(dim=0)
+
+
+def calc_feature_skew(batch):
+    """Asymmetric (zero-centered) skew, reference :482-487."""
+    var = batch.var(dim=0)
+    return (batch**3).mean(dim=0) / torch.clamp(var**1.5, min=1e-8)
+
+
+def calc_feature_kurtosis(batch):
+    var = batch.var(dim=0)
+    return (batch**4).mean(dim=0) / torch.clamp(var**2, min=1e-8)
+
+
+def calc_moments_streaming(learned_dict: LearnedDict, activations: torch.Tensor, batch_size: int = 1000):
+    """Streaming raw moments m1..m4 over encode outputs (reference :456-511).
+
+    Returns (times_active, mean, var, skew, kurtosis, m4).
+    """
+    n_feats = learned_dict.n_feats
+    dev = activations.device
+    times_active = torch.zeros(n_feats, device=dev)
+    m1 = torch.zeros(n_feats, device=dev)
+    m2 = torch.zeros(n_feats, device=dev)
+    m3 = torch.zeros(n_feats, device=dev)
+    m4 = torch.zeros(n_feats, device=dev)
+
+    n = 0
+    for i in range(0, len(activations), batch_size):
+        batch = activations[i : i + batch_size]
+        c = learned_dict.encode(batch)
+        b = c.shape[0]
+        total = n + b
+        times_active += (c.mean(dim=0) != 0).float()
+        m1 = (n * m1 + b * c.mean(dim=0)) / total
+        m2 = (n * m2 + b * (c**2).mean(dim=0)) / total
+        m3 = (n * m3 + b * (c**3).mean(dim=0)) / total
+        m4 = (n * m4 + b * (c**4).mean(dim=0)) / total
+        n = total
+
+    var = m2 - m1**2
+    skew = m3 / torch.clamp(var**1.5, min=1e-8)
+    kurtosis = m4 / torch.clamp(var**2, min=1e-8)
+    return times_active, m1, var, skew, kurtosis, m4
+
+
+# ---------------------------------------------------------------------------
+# probes (sklearn)
+# ---------------------------------------------------------------------------
+
+def logistic_regression_auroc(activations: torch.Tensor, labels: torch.Tensor, **kwargs) -> float:
+    from sklearn import metrics
+    from sklearn.linear_model import LogisticRegression
+
+    clf = LogisticRegression(**kwargs)
+    a, l = activations.cpu().numpy(), labels.cpu().numpy()
+    clf.fit(a, l)
+    return metrics.roc_auc_score(l, clf.predict_proba(a)[:, 1])
+
+
+def ridge_regression_auroc(activations: torch.Tensor, labels: torch.Tensor, **kwargs) -> float:
+    from sklearn import metrics
+    from sklearn.linear_model import RidgeClassifier
+
+    clf = RidgeClassifier(**kwargs)
+    a, l = activations.cpu().numpy(), labels.cpu().numpy()
+    clf.fit(a, l)
+    return metrics.roc_auc_score(l, clf.predict(a))
+
+
+# ---------------------------------------------------------------------------
+# clustering of dictionary directions (reference :534-579)
+# ---------------------------------------------------------------------------
+
+def cluster_directions_kmeans(model: LearnedDict, n_clusters: int = 16):
+    from sklearn.cluster import KMeans
+
+    d = model.get_learned_dict().cpu().numpy()
+    km = KMeans(n_clusters=n_clusters, n_init=4).fit(d)
+    return km.labels_, km.cluster_centers_
+
+
+def cluster_directions_hierarchical(model: LearnedDict, n_clusters: int = 16):
+    from sklearn.cluster import AgglomerativeClustering
+
+    d = model.get_learned_dict().cpu().numpy()
+    ag = AgglomerativeClustering(n_clusters=n_clusters).fit(d)
+    return ag.labels_
+
+
+# ---------------------------------------------------------------------------
+# perplexity under reconstruction (host-LM; HF hooks instead of TL)
+# ---------------------------------------------------------------------------
+
+def calculate_perplexity(
+    model,
+    tokenizer,
+    learned_dict: Optional[LearnedDict],
+    layer: int,
+    layer_loc: str,
+    token_ids: torch.Tensor,
+    device: str = "cuda:0",
+    batch_size: int = 8,
+) -> float:
+    """Mean LM cross-entropy (exp'd) with the hooked activation replaced by
+    ``learned_dict.predict`` (reference standard_metrics.py:621-709).
+
+    `token_ids`: [N, seq] int64. `learned_dict=None` gives the clean baseline.
+    """
+    from sparse_coding_amd.data.activation_dataset import replace_activation_hook
+
+    model.eval()
+    total_nll, total_tok = 0.0, 0
+    with torch.no_grad():
+        for i in range(0, token_ids.shape[0], batch_size):
+            ids = token_ids[i : i + batch_size].to(device)
+            with replace_activation_hook(model, layer, layer_loc, learned_dict):
+                out = model(input_ids=ids)
+            logits = out.logits if hasattr(out, "logits") else out[0]
+            logp = torch.log_softmax(logits[:, :-1].float(), dim=-1)
+            nll = -logp.gather(-1, ids[:, 1:, None]).squeeze(-1)
+            total_nll += nll.sum().item()
+            total_tok += nll.numel()
+    return float(np.exp(total_nll / max(total_tok, 1)))
+
+
+def perplexity_under_reconstruction(model, tokenizer, learned_dict, layer, layer_loc, token_ids, **kw):
+    return calculate_perplexity(model, tokenizer, learned_dict, layer, layer_loc, token_ids, **kw)
+
+
+# ---------------------------------------------------------------------------
+# plotting helpers (reference :364-439,514)
+# ---------------------------------------------------------------------------
+
+def plot_grid(scores: np.ndarray, first_ticks, second_ticks, first_label, second_label, **kwargs):
+    import matplotlib
+
+    matplotlib.use("Agg")
+    import matplotlib.pyplot as plt
+
+    fig, ax = plt.subplots()
+    ax.imshow(scores, **kwargs)
+    ax.set_xticks(np.arange(len(first_ticks)))
+    ax.set_yticks(np.arange(len(second_ticks)))
+    ax.set_xticklabels([str(t) for t in first_ticks])
+    ax.set_yticklabels([str(t) for t in second_ticks])
+    ax.set_xlabel(first_label)
+    ax.set_ylabel(second_label)
+    return fig
+
+
+def plot_hist(values: torch.Tensor, xlabel: str, ylabel: str, **kwargs):
+    import matplotlib
+
+    matplotlib.use("Agg")
+    import matplotlib.pyplot as plt
+
+    fig, ax = plt.subplots()
+    ax.hist(values.detach().cpu().numpy(), **kwargs)
+    ax.set_xlabel(xlabel)
+    ax.set_ylabel(ylabel)
+    return fig
+
+
+def plot_capacities(dicts: List[Tuple[LearnedDict, Dict[str, Any]]], show: bool = False, save_name: str = "capacities"):
+    import matplotlib
+
+    matplotlib.use("Agg")
+    import matplotlib.pyplot as plt
+
+    max_capacity = dicts[0][0].activation_size
+    sums = [capacity_per_feature(d[0]).sum().item() for d in dicts]
+    l1s = [d[1]["l1_alpha"] for d in dicts]
+    fig, ax = plt.subplots()
+    ax.scatter(l1s, sums)
+    ax.set_xlabel("L1 alpha")
+    ax.set_ylabel("Sum of capacities")
+    ax.set_xscale("log")
+    ax.axhline(max_capacity, color="red", linestyle="--")
+    ax.set_ylim(0, max_capacity * 1.1)
+    ax.set_title(f"Sum of capacities vs L1 alpha - {save_name}")
+    fig.savefig(save_name + ".png")
+    return fig

@@ -1,0 +1,146 @@
+"""Data parallelism over activation shards with RCCL over xGMI.
+
+Replaces the reference's two multi-GPU paths:
+  P1 cluster process-per-ensemble  → kept in sweep/cluster_runs.py
+  P3 DDP-over-gloo big-SAE trainer → THIS module: one process per GPU,
+     torch.distributed backend "nccl" (= RCCL on ROCm), gradient all-reduce
+     bucketed and overlapped with the grad kernels on a side stream.
+
+Topology note: xGMI is 7 p2p links × ≈153 GB/s per GPU; the
+[M,n,d] fp32 gradient set is tens-to-hundreds of MB, so we bucket to ~64 MB
+and launch all-reduces as soon as a bucket's producers complete, letting RCCL
+schedule rings across links while later grad GEMMs still run.
+"""
+
+from __future__ import annotations
+
+import os
+from typing import List, Optional, Tuple
+
+import torch
+import torch.distributed as dist
+
+from sparse_coding_amd.utils.tree import tree_flatten
+
+
+def init_distributed(backend: Optional[str] = None) -> Tuple[int, int, int]:
+    """Initialize torch.distributed from torchrun env; returns
+    (rank, local_rank, world_size).  Safe to call when WORLD_SIZE is unset
+    (returns a single-process layout without init)."""
+    world_size = int(os.environ.get("WORLD_SIZE", "1"))
+    rank = int(os.environ.get("RANK", "0"))
+    local_rank = int(os.environ.get("LOCAL_RANK", "0"))
+    if world_size > 1 and not dist.is_initialized():
+        if backend is None:
+            backend = "nccl" if torch.cuda.is_available() else "gloo"
+        os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+        os.environ.setdefault("MASTER_PORT", "29517")
+        dist.init_process_group(backend=backend, world_size=world_size, rank=rank)
+        if torch.cuda.is_available():
+            torch.cuda.set_device(local_rank)
+    return rank, local_rank, world_size
+
+
+class GradBucketAllReducer:
+    """Flattens a fixed tree of gradient tensors into persistent buckets and
+    all-reduces them (avg) on a dedicated stream.
+
+    The bucket layout is computed once (sorted-key tree flatten order is
+    deterministic across ranks — utils/tree.py), so every step is: copy
+    grads into buckets (device-side, async) → all_reduce per bucket →
+    copy back.  With the HIP step the copies are fused away by reducing the
+    grad workspaces directly when they are bucket-contiguous.
+    """
+
+    def __init__(self, grads_template, bucket_bytes: int = 64 << 20, group=None):
+        self.group = group
+        leaves, _ = tree_flatten(grads_template)
+        self.shapes = [g.shape for g in leaves]
+        self.numels = [g.numel() for g in leaves]
+        device = leaves[0].device
+        self.buckets: List[torch.Tensor] = []
+        self.assignments: List[List[Tuple[int, int, int]]] = []  # per bucket: (leaf_idx, offset, numel)
+        cur: List[Tuple[int, int, int]] = []
+        cur_size = 0
+        for i, n in enumerate(self.numels):
+            nb = n * 4
+            if cur and cur_size + nb > bucket_bytes:
+                self._seal(cur, device)
+                cur, cur_size = [], 0
+            cur.append((i, cur_size // 4, n))
+            cur_size += nb
+        if cur:
+            self._seal(cur, device)
+        self.comm_stream = torch.cuda.Stream() if device.type == "cuda" else None
+
+    def _seal(self, assignment, device):
+        total = sum(n for _, _, n in assignment)
+        self.buckets.append(torch.empty(total, device=device, dtype=torch.float32))
+        self.assignments.append(list(assignment))
+
+    def all_reduce_(self, grads) -> None:
+        """In-place average of `grads` across ranks."""
+        if not dist.is_initialized() or dist.get_world_size(self.group) == 1:
+            return
+        leaves, _ = tree_flatten(grads)
+        world = dist.get_world_size(self.group)
+
+        works = []
+        if self.comm_stream is not None:
+            self.comm_stream.wait_stream(torch.cuda.current_stream())
+        for bucket, assignment in zip(self.buckets, self.assignments):
+            for li, off, n in assignment:
+                bucket[off : off + n].copy_(leaves[li].reshape(-1), non_blocking=True)
+            if self.comm_stream is not None:
+                with torch.cuda.stream(self.comm_stream):
+                    works.append((dist.all_reduce(bucket, async_op=True), bucket, assignment))
+            else:
+                works.append((dist.all_reduce(bucket, async_op=True), bucket, assignment))
+
+        for work, bucket, assignment in works:
+            work.wait()
+            bucket.div_(world)
+            for li, off, n in assignment:
+                leaves[li].reshape(-1).copy_(bucket[off : off + n], non_blocking=True)
+        if self.comm_stream is not None:
+            torch.cuda.current_stream().wait_stream(self.comm_stream)
+
+
+class DataParallelEnsembleTrainer:
+    """DP over activation shards: every rank holds a replica of the SAME
+    ensemble; each step consumes a per-rank shard of the global batch,
+    all-reduces the [M, n, d] gradients, and applies the (deterministic)
+    Adam update locally — states stay bit-identical across ranks without a
+    broadcast."""
+
+    def __init__(self, ensemble, bucket_bytes: int = 64 << 20, group=None):
+        self.ensemble = ensemble
+        self.group = group
+        self._reducer: Optional[GradBucketAllReducer] = None
+        self.bucket_bytes = bucket_bytes
+        self.world_size = dist.get_world_size(group) if dist.is_initialized() else 1
+
+    def broadcast_state(self) -> None:
+        """One-time parameter/optimizer broadcast from rank 0 (use when
+        replicas were not constructed from the same seed)."""
+        if not dist.is_initialized():
+            return
+        for tree in (self.ensemble.params, self.ensemble.buffers, self.ensemble.optim_states):
+            for leaf in tree_flatten(tree)[0]:
+                if leaf.dtype.is_floating_point or leaf.dtype in (torch.int32, torch.int64):
+                    dist.broadcast(leaf, src=0, group=self.group)
+
+    def step(self, local_batch: torch.Tensor):
+        grads, (loss_data, aux) = self.ensemble.compute_grads(local_batch)
+        if self.world_size > 1:
+            if self._reducer is None:
+                self._reducer = GradBucketAllReducer(grads, self.bucket_bytes, self.group)
+            self._reducer.all_reduce_(grads)
+        self.ensemble.apply_grads(grads)
+        return loss_data, aux
+
+
+def shard_batch(batch: torch.Tensor, rank: int, world_size: int) -> torch.Tensor:
+    """Contiguous equal shard of a global batch."""
+    n = batch.shape[0] // world_size
+    return batch[rank * n : (rank + 1) * n]

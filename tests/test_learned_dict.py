@@ -1,0 +1,109 @@
+"""LearnedDict family semantics + pickle checkpoint compatibility."""
+
+import io
+import pickle
+
+import torch
+
+from sparse_coding_amd.models.learned_dict import (
+    Identity,
+    IdentityPositive,
+    IdentityReLU,
+    RandomDict,
+    ReverseSAE,
+    Rotation,
+    TiedSAE,
+    UntiedSAE,
+    normalize_rows,
+)
+
+
+def test_normalize_rows():
+    w = torch.randn(8, 4) * 5
+    n = normalize_rows(w)
+    assert torch.allclose(torch.norm(n, dim=-1), torch.ones(8), atol=1e-6)
+    # tiny rows: clamped denominator, not NaN
+    w2 = torch.zeros(2, 4)
+    assert not normalize_rows(w2).isnan().any()
+
+
+def test_untied_sae_encode_decode():
+    torch.manual_seed(0)
+    enc = torch.randn(16, 8)
+    dec = torch.randn(16, 8)
+    bias = torch.randn(16)
+    sae = UntiedSAE(enc, dec, bias)
+    x = torch.randn(5, 8)
+    c = sae.encode(x)
+    # einsum("nd,bd->bn") + bias, clamped
+    ref = torch.clamp(torch.einsum("nd,bd->bn", enc, x) + bias, min=0)
+    assert torch.allclose(c, ref, atol=1e-5)
+    x_hat = sae.decode(c)
+    ref_dec = torch.einsum("nd,bn->bd", normalize_rows(dec), c)
+    assert torch.allclose(x_hat, ref_dec, atol=1e-5)
+    assert torch.allclose(sae.predict(x), ref_dec, atol=1e-5)
+
+
+def test_tied_sae_centering_roundtrip():
+    torch.manual_seed(1)
+    d = 6
+    enc = torch.randn(12, d)
+    bias = torch.zeros(12)
+    # random orthogonal rotation
+    q, _ = torch.linalg.qr(torch.randn(d, d))
+    trans = torch.randn(d)
+    scale = torch.rand(d) + 0.5
+    sae = TiedSAE(enc, bias, centering=(trans, q, scale))
+    x = torch.randn(7, d)
+    assert torch.allclose(sae.uncenter(sae.center(x)), x, atol=1e-5)
+
+
+def test_identity_variants():
+    x = torch.randn(4, 6)
+    assert torch.equal(Identity(6).encode(x), x)
+    pos = IdentityPositive(6)
+    c = pos.encode(x)
+    assert c.shape == (4, 12)
+    assert (c >= 0).all()
+    assert torch.allclose(pos.decode(c), x, atol=1e-6)
+    relu = IdentityReLU(6)
+    assert torch.equal(relu.encode(x), torch.clamp(x, min=0))
+
+
+def test_reverse_sae_bias_removal():
+    torch.manual_seed(2)
+    enc = torch.randn(10, 5)
+    bias = torch.rand(10) + 0.1
+    sae = ReverseSAE(enc, bias, norm_encoder=True)
+    x = torch.randn(3, 5)
+    c = sae.encode(x)
+    x_hat = sae.decode(c.clone())
+    on = c > 0
+    c_adj = torch.where(on, c - bias, c)
+    assert torch.allclose(x_hat, c_adj @ normalize_rows(enc), atol=1e-5)
+
+
+def test_pickle_module_path_is_reference_compatible():
+    """Checkpoints must unpickle as autoencoders.learned_dict.* (SURVEY §2.3)."""
+    sae = TiedSAE(torch.randn(4, 3), torch.zeros(4))
+    assert type(sae).__module__ == "autoencoders.learned_dict"
+    buf = io.BytesIO()
+    torch.save([(sae, {"l1_alpha": 1e-3, "dict_size": 4})], buf)
+    buf.seek(0)
+    loaded = torch.load(buf, weights_only=False)
+    ld, hp = loaded[0]
+    assert isinstance(ld, TiedSAE)
+    assert hp["dict_size"] == 4
+    # raw pickle path string check
+    buf2 = io.BytesIO()
+    pickle.dump(sae.__class__, buf2)
+    assert b"autoencoders" in buf2.getvalue()
+
+
+def test_rotation_and_random_dict():
+    q, _ = torch.linalg.qr(torch.randn(5, 5))
+    rot = Rotation(q)
+    x = torch.randn(3, 5)
+    assert torch.allclose(rot.encode(x), x @ q.T, atol=1e-6)
+    rd = RandomDict(5, 9)
+    assert rd.encode(x).shape == (3, 9)

@@ -1,0 +1,161 @@
+"""Distributed-path tests on CPU: gloo world_size=2 DP equivalence, cluster
+dispatch, big-SAE resampling."""
+
+import os
+
+import numpy as np
+import pytest
+import torch
+import torch.distributed as dist
+import torch.multiprocessing as mp
+
+from sparse_coding_amd.engine.ensemble import FunctionalEnsemble
+from sparse_coding_amd.functional.optim import adam
+from sparse_coding_amd.models.sae_signatures import FunctionalTiedSAE
+
+D, N, B = 16, 32, 128
+
+
+def _make_ensemble(seed=0, device="cpu"):
+    torch.manual_seed(seed)
+    models = [FunctionalTiedSAE.init(D, N, 1e-3) for _ in range(2)]
+    return FunctionalEnsemble(models, FunctionalTiedSAE, adam, {"lr": 1e-3}, device=device, backend="torch")
+
+
+def _dp_worker(rank, world_size, port, batch, out_q):
+    try:
+        os.environ["MASTER_ADDR"] = "127.0.0.1"
+        os.environ["MASTER_PORT"] = str(port)
+        os.environ["WORLD_SIZE"] = str(world_size)
+        os.environ["RANK"] = str(rank)
+        os.environ["LOCAL_RANK"] = str(rank)
+        dist.init_process_group("gloo", rank=rank, world_size=world_size)
+
+        from sparse_coding_amd.parallel.dp import DataParallelEnsembleTrainer, shard_batch
+
+        ens = _make_ensemble(seed=0)  # same seed → identical replicas
+        trainer = DataParallelEnsembleTrainer(ens, bucket_bytes=1 << 16)
+        for _ in range(3):
+            local = shard_batch(batch, rank, world_size)
+            trainer.step(local)
+        if rank == 0:
+            # numpy → pickled by value: the child may exit before the parent
+            # drains the queue (torch tensors would ship an fd to a dead sharer)
+            out_q.put({k: v.detach().cpu().numpy() for k, v in ens.params.items()})
+        dist.destroy_process_group()
+    except Exception:  # noqa: BLE001
+        import traceback
+
+        out_q.put({"_error": f"rank {rank}: {traceback.format_exc()}"})
+        raise
+
+
+@pytest.mark.timeout(120)
+def test_dp_matches_single_process():
+    """2-rank sharded training == single-process full-batch training."""
+    torch.manual_seed(42)
+    batch = torch.randn(B, D)
+
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    port = 29531
+    procs = [ctx.Process(target=_dp_worker, args=(r, 2, port, batch, q)) for r in range(2)]
+    for p in procs:
+        p.start()
+    dp_params = q.get(timeout=100)
+    assert "_error" not in dp_params, dp_params.get("_error")
+    for p in procs:
+        p.join(timeout=60)
+        assert p.exitcode == 0
+
+    ens = _make_ensemble(seed=0)
+    for _ in range(3):
+        # full-batch grad == mean of shard grads (all losses are means)
+        ens.step_batch(batch)
+
+    for k in dp_params:
+        assert torch.allclose(torch.from_numpy(dp_params[k]), ens.params[k], atol=1e-5), k
+
+
+def _cluster_job(ensemble, cfg, args, name, sampler, dataset, progress_counter):
+    for i, idxs in enumerate(sampler):
+        batch = dataset[idxs].to(args["device"])
+        ensemble.step_batch(batch)
+        progress_counter.value = i
+
+
+@pytest.mark.timeout(120)
+def test_cluster_dispatch_two_ensembles():
+    from types import SimpleNamespace
+
+    from sparse_coding_amd.sweep.cluster_runs import dispatch_job_on_chunk
+
+    mp.set_start_method("spawn", force=True)
+    chunk = torch.randn(512, D)
+    e1, e2 = _make_ensemble(seed=1), _make_ensemble(seed=2)
+    before = e1.params["encoder"].clone()
+    cfg = SimpleNamespace(batch_size=128, show_progress=False)
+    ensembles = [
+        (e1, {"batch_size": 128, "device": "cpu"}, "a"),
+        (e2, {"batch_size": 128, "device": "cpu"}, "b"),
+    ]
+    dispatch_job_on_chunk(ensembles, cfg, chunk, _cluster_job)
+    # children trained through shared memory → parent sees updated params
+    assert not torch.allclose(before, e1.params["encoder"])
+
+
+def test_big_sae_resampling(tmp_path):
+    from sparse_coding_amd.parallel.big_sae import (
+        BigSAE,
+        WorstExampleTracker,
+        resample_dead_features,
+        train_big_sae,
+    )
+
+    torch.manual_seed(0)
+    model = BigSAE(D, N, 1e-3)
+    opt = torch.optim.Adam(model.parameters(), lr=1e-3)
+    x = torch.randn(64, D)
+    for _ in range(3):
+        opt.zero_grad()
+        loss, mse, l1, c, per_ex = model(x)
+        loss.backward()
+        opt.step()
+
+    tracker = WorstExampleTracker(N, D, "cpu")
+    tracker.update(x, per_ex.detach())
+    c_totals = (c > 0).float().sum(0).detach()
+    c_totals[:5] = 0  # force 5 dead features
+    n = resample_dead_features(model, opt, c_totals, tracker)
+    assert n == 5
+    # adam state zeroed on those rows
+    st = opt.state[model.encoder]
+    assert (st["exp_avg"][:5] == 0).all()
+    assert (st["exp_avg"][5:] != 0).any()
+    # replaced encoder rows have the scaled norm
+    enc_norms = torch.norm(model.encoder[:5], dim=-1)
+    assert (enc_norms > 0).all()
+
+    # single-process training loop over chunk files
+    for i in range(2):
+        torch.save(torch.randn(256, D, dtype=torch.float16), tmp_path / f"{i}.pt")
+    m = train_big_sae(
+        [str(tmp_path / f"{i}.pt") for i in range(2)],
+        activation_size=D, n_features=N, batch_size=64,
+        reinit_every_chunks=1, device="cpu", log_fn=lambda *a: None,
+    )
+    assert isinstance(m, BigSAE)
+
+
+def test_grad_bucket_allreducer_layout():
+    from sparse_coding_amd.parallel.dp import GradBucketAllReducer
+
+    grads = {"a": torch.randn(100), "b": torch.randn(300), "c": torch.randn(50)}
+    red = GradBucketAllReducer(grads, bucket_bytes=1024)  # 256 floats per bucket
+    total = sum(b.numel() for b in red.buckets)
+    assert total == 450
+    # without dist init, all_reduce_ is a no-op
+    before = {k: v.clone() for k, v in grads.items()}
+    red.all_reduce_(grads)
+    for k in grads:
+        assert torch.equal(before[k], grads[k])
