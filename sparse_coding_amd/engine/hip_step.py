@@ -1,26 +1,32 @@
 """Fused CDNA4 training step for SAE ensembles.
 
 Implements the reference's vmapped grad+Adam `step_batch`
-(autoencoders/ensemble.py:119-123,175-193 + sae_ensemble.py losses) as a
-pipeline of hand-written gfx950 HIP kernels (sources in
-``sparse_coding_amd/ops/hip/``):
+(autoencoders/ensemble.py:119-123,175-193 + the FunctionalSAE /
+FunctionalTiedSAE losses in sae_ensemble.py) as a pipeline of hand-written
+gfx950 HIP kernels (sources in ``sparse_coding_amd/ops/hip/sae_kernels.hip``):
 
-  k_row_norms    : per-row 1/max(||w||, 1e-8) of the dictionary      [M,n]
-  k_fwd_gc       : encoder-GEMM → +bias → ReLU → decoder-GEMM →
-                   residual, MSE/L1 partials, fired counts, and the
-                   code-gradient GEMM → relu-mask → g_pre   (f32 MFMA)
-  k_grad_w       : grad GEMMs over the batch dim → g_Ŵ (+ g_Wenc) + g_bias
+  k_row_norms    : per-row ||w|| and 1/max(||w||,1e-8) of the dictionary
+  k_enc_fwd      : encoder GEMM -> +bias -> ReLU -> c  (+L1 partial, fired)
+  k_dec_fwd      : decoder GEMM (rows scaled by inv-norm) -> residual r
+                   (+MSE partial)
+  k_gc           : code-grad GEMM -> relu mask -> +l1 term -> g_pre
+                   (+bias-grad column sums)
+  k_grad_w       : batch-contraction GEMMs -> dL/d(W_hat) (and encoder grad)
   k_project_adam : analytic gradient of w/max(||w||,eps) (the in-forward
-                   renormalization), fused Adam update of all params
+                   decoder renormalization) + fused Adam
+  k_bias_adam    : Adam on the bias (+ L2-decay gradient)
 
-Numerics: fp32 end to end (the reference trains fp32 — BASELINE.md), using
-the exact-f32 MFMA path (`v_mfma_f32_*_f32`).  Validated against the
+All GEMMs run on the exact-f32 MFMA path (v_mfma_f32_32x32x2_f32): fp32 end
+to end, the reference's training dtype.  Validated against the
 torch/vmap oracle in tests/test_hip_numerics.py.
+
+For multi-GPU DP the step splits at the gradient boundary
+(`grads_phase` / `update_phase`) so the RCCL all-reduce of [M,n,d] grads
+overlaps with nothing locally but lands between grad GEMMs and Adam.
 """
 
 from __future__ import annotations
 
-import math
 from typing import Optional
 
 import torch
@@ -28,8 +34,10 @@ import torch
 from sparse_coding_amd import ops as _ops
 from sparse_coding_amd.models import sae_signatures as sigs
 
+EPS_NORM = 1e-8
 
-def _identity_centering(buffers, n_models: int) -> bool:
+
+def _identity_centering(buffers) -> bool:
     rot = buffers.get("center_rot")
     scale = buffers.get("center_scale")
     trans = buffers.get("center_trans")
@@ -37,20 +45,14 @@ def _identity_centering(buffers, n_models: int) -> bool:
         return True
     d = rot.shape[-1]
     eye = torch.eye(d, device=rot.device, dtype=rot.dtype).expand_as(rot)
-    return (
-        torch.equal(rot, eye)
-        and bool((scale == 1).all())
-        and bool((trans == 0).all())
-    )
+    return torch.equal(rot, eye) and bool((scale == 1).all()) and bool((trans == 0).all())
 
 
 def maybe_make_step(ensemble, required: bool = False) -> Optional["HipSAEStep"]:
     """Return a HipSAEStep if (sig, device, buffers) are supported.
 
     On a CUDA device with a supported signature, a missing extension is a
-    hard error (required or not): GPU runs must not silently fall back to
-    eager (round-end check records which .so files were loaded).
-    """
+    hard error: GPU runs must not silently fall back to eager."""
     import os
 
     if os.environ.get("SPARSE_CODING_AMD_FORCE_TORCH") == "1":
@@ -69,7 +71,7 @@ def maybe_make_step(ensemble, required: bool = False) -> Optional["HipSAEStep"]:
         if required:
             raise RuntimeError(f"backend='hip' requested but signature {sig.__name__} has no fused step yet")
         return None
-    if tied and not _identity_centering(ensemble.buffers, ensemble.n_models):
+    if tied and not _identity_centering(ensemble.buffers):
         if required:
             raise RuntimeError("fused tied step requires identity centering buffers")
         return None
@@ -79,7 +81,7 @@ def maybe_make_step(ensemble, required: bool = False) -> Optional["HipSAEStep"]:
 
 
 class HipSAEStep:
-    """Holds workspaces + launches the fused step for one ensemble."""
+    """Workspaces + kernel-sequence launcher for one ensemble."""
 
     def __init__(self, ensemble, ext, tied: bool):
         self.ens = ensemble
@@ -92,91 +94,126 @@ class HipSAEStep:
         if not tied:
             assert p["decoder"].shape == p["encoder"].shape
 
-        # per-model hyperparams as [M] fp32 device tensors
+        dev = p["encoder"].device
         b = ensemble.buffers
-        self.l1_alpha = b["l1_alpha"].to(torch.float32).reshape(self.n_models).contiguous()
+        self.l1_alpha = b["l1_alpha"].detach().to(dev, torch.float32).reshape(self.n_models).contiguous()
         bd = b.get("bias_decay")
         if bd is None:
-            bd = torch.zeros(self.n_models, device=p["encoder"].device)
-        self.bias_decay = bd.to(torch.float32).reshape(self.n_models).contiguous()
+            bd = torch.zeros(self.n_models, device=dev)
+        self.bias_decay = bd.detach().to(dev, torch.float32).reshape(self.n_models).contiguous()
 
         opt = ensemble.optimizer_kwargs
         self.lr = float(opt.get("lr", 1e-3))
         betas = opt.get("betas", (0.9, 0.999))
         self.beta1, self.beta2 = float(betas[0]), float(betas[1])
         self.eps = float(opt.get("eps", 1e-8))
-        if ensemble.optimizer_func is not None:
-            name = getattr(ensemble.optimizer_func, "__name__", "adam")
-            if name not in ("adam",):
-                raise RuntimeError(f"fused step supports adam only, got {name}")
+        name = getattr(ensemble.optimizer_func, "__name__", "adam")
+        if name != "adam":
+            raise RuntimeError(f"fused HIP step supports adam only, got {name}")
 
-    def _workspace(self, name, shape, dtype=torch.float32):
-        ws = self._ws.get(name)
-        if ws is None or ws.shape != torch.Size(shape):
-            ws = torch.empty(shape, device=self.ens.params["encoder"].device, dtype=dtype)
-            self._ws[name] = ws
-        return ws
+        # persistent workspaces, sized lazily on first batch
+        self._B = None
 
-    def step(self, minibatches: torch.Tensor, expand_dims: bool = True):
-        ens = self.ens
+    # -- workspace management -------------------------------------------------
+    def _alloc(self, B: int):
         M, n, d = self.n_models, self.n_dict, self.d_act
-        if expand_dims:
-            x = minibatches  # [B, d] shared across models
-            shared_x = True
-        else:
-            x = minibatches  # [M, B, d]
-            shared_x = False
-            raise NotImplementedError("per-model batches not yet supported by the HIP step")
-        B = x.shape[0]
-        x = x.contiguous()
+        dev = self.ens.params["encoder"].device
+        f = lambda *shape: torch.empty(shape, device=dev, dtype=torch.float32)
+        self.c = f(M, B, n)
+        self.gpre = f(M, B, n)
+        self.r = f(M, B, d)
+        self.norms = f(M, n)
+        self.inv_norms = f(M, n)
+        self.loss_parts = f(M, 2)
+        self.fired = torch.zeros(M, n, device=dev)
+        self.g_bias = f(M, n)
+        self.gw = f(M, n, d)
+        if not self.tied:
+            self.gw_enc = f(M, n, d)
+        self._B = B
 
+    # -- phases ---------------------------------------------------------------
+    def grads_phase(self, x: torch.Tensor):
+        """Forward + backward: fills self.gw (+gw_enc), g_bias, loss parts."""
+        ens, ext = self.ens, self.ext
+        M, n, d = self.n_models, self.n_dict, self.d_act
+        B = x.shape[0]
+        if self._B != B:
+            self._alloc(B)
+
+        x = x.contiguous()
         p = ens.params
-        st = ens.optim_states
         enc = p["encoder"]
         bias = p["encoder_bias"]
-        dec = enc if self.tied else p["decoder"]
+        dict_w = enc if self.tied else p["decoder"]
 
-        c = self._workspace("c", (M, B, n))
-        gpre = self._workspace("gpre", (M, B, n))
-        r = self._workspace("r", (M, B, d))
-        inv_norms = self._workspace("inv_norms", (M, n))
-        loss_parts = self._workspace("loss_parts", (M, 2))  # mse_sum, l1_sum
-        fired = self._workspace("fired", (M, n))
-        gw_dec = self._workspace("gw_dec", (M, n, d))
-        gw_enc = gw_dec if self.tied else self._workspace("gw_enc", (M, n, d))
-        g_bias = self._workspace("g_bias", (M, n))
+        self.loss_parts.zero_()
+        self.g_bias.zero_()
+        self._fired_step = self.fired  # accumulated across steps for resampling
 
-        step_no = st["step"]  # [M] float
-        mu_enc = st["mu"]["encoder"]
-        nu_enc = st["nu"]["encoder"]
-        mu_bias = st["mu"]["encoder_bias"]
-        nu_bias = st["nu"]["encoder_bias"]
+        ext.row_norms(dict_w, self.norms, self.inv_norms, EPS_NORM)
+        ext.enc_fwd(x, enc, bias, self.inv_norms if self.tied else None,
+                    self.c, self.loss_parts, self.fired)
+        ext.dec_fwd(self.c, dict_w, self.inv_norms, x, self.r, self.loss_parts)
+        ext.gc(self.r, dict_w, self.inv_norms, self.c, self.l1_alpha,
+               self.gpre, self.g_bias)
+
+        gscale = 2.0 / (B * d)
         if self.tied:
-            mu_dec, nu_dec = mu_enc, nu_enc
+            ext.grad_w(self.c, self.r, self.gw, gscale, 0.0)
+            ext.grad_w(self.gpre, x, self.gw, 1.0, 1.0)
         else:
-            mu_dec, nu_dec = st["mu"]["decoder"], st["nu"]["decoder"]
+            ext.grad_w(self.c, self.r, self.gw, gscale, 0.0)
+            ext.grad_w(self.gpre, x, self.gw_enc, 1.0, 0.0)
+        self._x = x
+        return B
 
-        self.ext.sae_step_f32(
-            x, enc, dec, bias,
-            self.l1_alpha, self.bias_decay,
-            c, gpre, r, inv_norms, loss_parts, fired,
-            gw_dec, gw_enc, g_bias,
-            mu_enc, nu_enc, mu_dec, nu_dec, mu_bias, nu_bias,
-            step_no,
-            self.lr, self.beta1, self.beta2, self.eps,
-            self.tied,
-        )
+    def update_phase(self, B: int):
+        """Adam updates (projection through the renorm for the dictionary)."""
+        ens, ext = self.ens, self.ext
+        st = ens.optim_states
+        st["step"] += 1.0
+        step_no = st["step"]
+        p = ens.params
 
-        mse = loss_parts[:, 0] / (B * d)
-        l1 = loss_parts[:, 1] / B * self.l1_alpha
-        bias_norm = torch.norm(bias, 2, dim=-1)
-        l_bd = self.bias_decay * bias_norm
+        if self.tied:
+            ext.project_adam(p["encoder"], self.gw, self.norms,
+                             st["mu"]["encoder"], st["nu"]["encoder"], step_no,
+                             self.n_dict, self.lr, self.beta1, self.beta2,
+                             self.eps, EPS_NORM, True)
+        else:
+            ext.project_adam(p["decoder"], self.gw, self.norms,
+                             st["mu"]["decoder"], st["nu"]["decoder"], step_no,
+                             self.n_dict, self.lr, self.beta1, self.beta2,
+                             self.eps, EPS_NORM, True)
+            ext.project_adam(p["encoder"], self.gw_enc, self.norms,
+                             st["mu"]["encoder"], st["nu"]["encoder"], step_no,
+                             self.n_dict, self.lr, self.beta1, self.beta2,
+                             self.eps, EPS_NORM, False)
+        ext.bias_adam(p["encoder_bias"], self.g_bias, self.bias_decay,
+                      st["mu"]["encoder_bias"], st["nu"]["encoder_bias"],
+                      step_no, self.lr, self.beta1, self.beta2, self.eps)
+
+    def _loss_data(self, B: int):
+        d = self.d_act
+        mse = self.loss_parts[:, 0] / (B * d)
+        l1 = self.l1_alpha * self.loss_parts[:, 1] / B
+        bias = self.ens.params["encoder_bias"]
+        l_bd = self.bias_decay * torch.norm(bias, 2, dim=-1)
         total = mse + l1 + l_bd
-        loss_data = {
-            "loss": total,
-            "l_reconstruction": mse,
-            "l_l1": l1,
-            "l_bias_decay": l_bd,
-        }
-        aux_data = {"c": c}
-        return loss_data, aux_data
+        return {"loss": total, "l_reconstruction": mse, "l_l1": l1, "l_bias_decay": l_bd}
+
+    # -- public ---------------------------------------------------------------
+    def step(self, minibatches: torch.Tensor, expand_dims: bool = True):
+        if not expand_dims:
+            raise NotImplementedError("per-model batches not supported by the HIP step")
+        B = self.grads_phase(minibatches)
+        self.update_phase(B)
+        return self._loss_data(B), {"c": self.c}
+
+    def dp_grad_tensors(self):
+        """Tensors to all-reduce for data parallelism (average across ranks)."""
+        ts = [self.gw, self.g_bias]
+        if not self.tied:
+            ts.append(self.gw_enc)
+        return ts

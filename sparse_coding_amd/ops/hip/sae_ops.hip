@@ -1,0 +1,142 @@
+// Torch bindings + launchers for the gfx950 SAE kernels (sae_kernels.hip).
+// Built in-tree as sparse_coding_amd/ops/_sae_hip.so by ops/build.py.
+
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+#include <hip/hip_runtime.h>
+
+#include "sae_kernels.hip"
+
+#define CHECK_IN(t)                                           \
+  TORCH_CHECK(t.is_cuda(), #t " must be on GPU");             \
+  TORCH_CHECK(t.is_contiguous(), #t " must be contiguous");   \
+  TORCH_CHECK(t.scalar_type() == torch::kFloat32, #t " must be fp32");
+
+static inline hipStream_t cur_stream() {
+  return at::hip::getCurrentHIPStream().stream();
+}
+
+static inline int cdiv(long a, long b) { return (int)((a + b - 1) / b); }
+
+void row_norms(torch::Tensor W, torch::Tensor norms, torch::Tensor inv_norms, double eps) {
+  CHECK_IN(W); CHECK_IN(norms); CHECK_IN(inv_norms);
+  long rows = W.numel() / W.size(-1);
+  int d = W.size(-1);
+  dim3 grid(cdiv(rows, 4));
+  hipLaunchKernelGGL(k_row_norms, grid, dim3(NTHREADS), 0, cur_stream(),
+                     W.data_ptr<float>(), norms.data_ptr<float>(),
+                     inv_norms.data_ptr<float>(), (int)rows, d, (float)eps);
+}
+
+void enc_fwd(torch::Tensor x, torch::Tensor Wenc, torch::Tensor bias,
+             c10::optional<torch::Tensor> inv_norms, torch::Tensor c_out,
+             torch::Tensor loss_parts, torch::Tensor fired) {
+  CHECK_IN(x); CHECK_IN(Wenc); CHECK_IN(bias); CHECK_IN(c_out);
+  CHECK_IN(loss_parts); CHECK_IN(fired);
+  int M = Wenc.size(0), n = Wenc.size(1), d = Wenc.size(2);
+  int B = x.size(0);
+  TORCH_CHECK(d % 4 == 0 && n % 4 == 0, "d and n must be multiples of 4 (float4 staging)");
+  const float* inv = nullptr;
+  if (inv_norms.has_value()) {
+    CHECK_IN(inv_norms.value());
+    inv = inv_norms->data_ptr<float>();
+  }
+  dim3 grid(cdiv(n, BN), cdiv(B, BM), M);
+  hipLaunchKernelGGL(k_enc_fwd, grid, dim3(NTHREADS), 0, cur_stream(),
+                     x.data_ptr<float>(), Wenc.data_ptr<float>(),
+                     bias.data_ptr<float>(), inv, c_out.data_ptr<float>(),
+                     loss_parts.data_ptr<float>(), fired.data_ptr<float>(),
+                     B, d, n);
+}
+
+void dec_fwd(torch::Tensor c, torch::Tensor Wdec, torch::Tensor inv_norms,
+             torch::Tensor x, torch::Tensor r_out, torch::Tensor loss_parts) {
+  CHECK_IN(c); CHECK_IN(Wdec); CHECK_IN(inv_norms); CHECK_IN(x);
+  CHECK_IN(r_out); CHECK_IN(loss_parts);
+  int M = Wdec.size(0), n = Wdec.size(1), d = Wdec.size(2);
+  int B = x.size(0);
+  dim3 grid(cdiv(d, BN), cdiv(B, BM), M);
+  hipLaunchKernelGGL(k_dec_fwd, grid, dim3(NTHREADS), 0, cur_stream(),
+                     c.data_ptr<float>(), Wdec.data_ptr<float>(),
+                     inv_norms.data_ptr<float>(), x.data_ptr<float>(),
+                     r_out.data_ptr<float>(), loss_parts.data_ptr<float>(),
+                     B, d, n);
+}
+
+void gc(torch::Tensor r, torch::Tensor Wdec, torch::Tensor inv_norms,
+        torch::Tensor c, torch::Tensor l1_alpha, torch::Tensor gpre,
+        torch::Tensor g_bias) {
+  CHECK_IN(r); CHECK_IN(Wdec); CHECK_IN(inv_norms); CHECK_IN(c);
+  CHECK_IN(l1_alpha); CHECK_IN(gpre); CHECK_IN(g_bias);
+  int M = Wdec.size(0), n = Wdec.size(1), d = Wdec.size(2);
+  int B = r.size(1);
+  dim3 grid(cdiv(n, BN), cdiv(B, BM), M);
+  hipLaunchKernelGGL(k_gc, grid, dim3(NTHREADS), 0, cur_stream(),
+                     r.data_ptr<float>(), Wdec.data_ptr<float>(),
+                     inv_norms.data_ptr<float>(), c.data_ptr<float>(),
+                     l1_alpha.data_ptr<float>(), gpre.data_ptr<float>(),
+                     g_bias.data_ptr<float>(), B, d, n);
+}
+
+// gw[m] = beta * gw[m] + alpha * P[m]^T @ Q[m]; Q may be rank-shared [B, d]
+void grad_w(torch::Tensor P, torch::Tensor Q, torch::Tensor gw,
+            double alpha, double beta) {
+  CHECK_IN(P); CHECK_IN(Q); CHECK_IN(gw);
+  int M = gw.size(0), n = gw.size(1), d = gw.size(2);
+  int B;
+  long p_stride, q_stride;
+  TORCH_CHECK(P.dim() == 3, "P must be [M, B, n]");
+  B = P.size(1);
+  p_stride = (long)B * n;
+  if (Q.dim() == 3) {
+    q_stride = (long)B * d;
+  } else {
+    q_stride = 0;  // shared across models
+  }
+  dim3 grid(cdiv(d, BN), cdiv(n, BM), M);
+  hipLaunchKernelGGL(k_grad_w, grid, dim3(NTHREADS), 0, cur_stream(),
+                     P.data_ptr<float>(), p_stride, Q.data_ptr<float>(),
+                     q_stride, gw.data_ptr<float>(), (float)alpha,
+                     (float)beta, B, n, d);
+}
+
+void project_adam(torch::Tensor W, torch::Tensor gw, torch::Tensor norms,
+                  torch::Tensor mu, torch::Tensor nu, torch::Tensor step_no,
+                  long n_per_model, double lr, double b1, double b2,
+                  double eps_adam, double eps_norm, bool project) {
+  CHECK_IN(W); CHECK_IN(gw); CHECK_IN(norms); CHECK_IN(mu); CHECK_IN(nu);
+  CHECK_IN(step_no);
+  long rows = W.numel() / W.size(-1);
+  int d = W.size(-1);
+  dim3 grid(cdiv(rows, 4));
+  hipLaunchKernelGGL(k_project_adam, grid, dim3(NTHREADS), 0, cur_stream(),
+                     W.data_ptr<float>(), gw.data_ptr<float>(),
+                     norms.data_ptr<float>(), mu.data_ptr<float>(),
+                     nu.data_ptr<float>(), step_no.data_ptr<float>(),
+                     (int)rows, (int)n_per_model, d, (float)lr, (float)b1,
+                     (float)b2, (float)eps_adam, (float)eps_norm,
+                     project ? 1 : 0);
+}
+
+void bias_adam(torch::Tensor bias, torch::Tensor g_bias, torch::Tensor decay,
+               torch::Tensor mu, torch::Tensor nu, torch::Tensor step_no,
+               double lr, double b1, double b2, double eps_adam) {
+  CHECK_IN(bias); CHECK_IN(g_bias); CHECK_IN(decay); CHECK_IN(mu);
+  CHECK_IN(nu); CHECK_IN(step_no);
+  int M = bias.size(0), n = bias.size(1);
+  hipLaunchKernelGGL(k_bias_adam, dim3(M), dim3(NTHREADS), 0, cur_stream(),
+                     bias.data_ptr<float>(), g_bias.data_ptr<float>(),
+                     decay.data_ptr<float>(), mu.data_ptr<float>(),
+                     nu.data_ptr<float>(), step_no.data_ptr<float>(), n,
+                     (float)lr, (float)b1, (float)b2, (float)eps_adam);
+}
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("row_norms", &row_norms, "dictionary row norms + clamped inverses");
+  m.def("enc_fwd", &enc_fwd, "fused encoder GEMM + bias + ReLU (+L1, fired)");
+  m.def("dec_fwd", &dec_fwd, "fused decoder GEMM - x (+MSE partial)");
+  m.def("gc", &gc, "code-gradient GEMM + relu mask + l1 term (+bias grad)");
+  m.def("grad_w", &grad_w, "gw = beta*gw + alpha * P^T Q (batched over M)");
+  m.def("project_adam", &project_adam, "renorm-gradient projection + Adam");
+  m.def("bias_adam", &bias_adam, "Adam on bias with L2-norm decay");
+}

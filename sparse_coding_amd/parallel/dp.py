@@ -131,6 +131,19 @@ class DataParallelEnsembleTrainer:
                     dist.broadcast(leaf, src=0, group=self.group)
 
     def step(self, local_batch: torch.Tensor):
+        hs = getattr(self.ensemble, "_hip_step", None)
+        if hs is not None:
+            # fused path: split at the gradient boundary so the RCCL
+            # all-reduce sits between the grad GEMMs and the Adam kernels
+            B = hs.grads_phase(local_batch)
+            if self.world_size > 1:
+                works = [dist.all_reduce(t, async_op=True, group=self.group) for t in hs.dp_grad_tensors()]
+                for w, t in zip(works, hs.dp_grad_tensors()):
+                    w.wait()
+                    t.div_(self.world_size)
+            hs.update_phase(B)
+            return hs._loss_data(B), {"c": hs.c}
+
         grads, (loss_data, aux) = self.ensemble.compute_grads(local_batch)
         if self.world_size > 1:
             if self._reducer is None:
