@@ -5,23 +5,27 @@
 // backward through the in-forward decoder row renormalization, fused Adam)
 // as hand-written MFMA kernels.  fp32 is the reference's training dtype
 // (BASELINE.md); we use the exact-f32 matrix instruction
-// v_mfma_f32_32x32x2_f32 (157 TF peak, bitwise == fmaf chain).
+// v_mfma_f32_32x32x2_f32 (157 TF peak, bitwise == an fmaf chain).
 //
-// Structure: one block-tile GEMM core (128x128 output tile, BK=32, 4 waves,
-// each wave a 64x64 quadrant as 2x2 v_mfma_f32_32x32x2_f32 accumulators),
-// instantiated with different operand-staging modes and fused epilogues:
+// GEMM core structure (tuned from rocprof on MI355X):
+//  * 128x128 output tile, BK=32, 8 waves (512 threads); each wave owns a
+//    32x64 quadrant as 2 v_mfma_f32_32x32x2_f32 accumulators (32 AGPRs) --
+//    low register pressure => 4 waves/SIMD, 2 blocks/CU co-resident.
+//  * register-prefetch pipeline (guide T14/G15): the NEXT K-tile's global
+//    loads issue before the MFMA phase of the current tile (latency hides
+//    under the 64-cyc/MFMA f32 pipe), the LDS write lands after the barrier.
+//  * LDS staging is conflict-free: transposed tiles use stride BM+1 (b32
+//    writes, (k+i)%32 covers every bank once per lane group); direct tiles
+//    use stride BM with float4 (b128) writes.
+//  * XCD-aware bijective block swizzle (guide T1): consecutive remapped
+//    blocks land on one XCD and share operand panels in its private L2.
+//
+// Kernels:
 //   k_enc_fwd : c = relu(x @ Wenc^T + b)    (+ L1 partial, fired counts)
 //   k_dec_fwd : r = c @ Wdec_hat - x        (+ MSE partial)
 //   k_gc      : gpre = (c>0) .* (gs * r @ Wdec_hat^T + l1/B)  (+ bias-grad)
 //   k_grad_w  : gw = beta*gw + alpha * P^T @ Q   (K = batch contraction)
-// plus k_row_norms (dictionary row norms), k_project_adam (gradient of
-// w/max(||w||,eps) + Adam), k_bias_adam.
-//
-// LDS: transposed-stage tiles use stride BM+1 (conflict-free b32 writes and
-// reads); direct-stage tiles use stride BM with float4 (ds_write_b128)
-// writes (8-lane groups cover all 32 banks).  Grids are (tiles_n, tiles_m,
-// n_models) -- thousands of workgroups for the flagship shapes, enough to
-// fill 256 CUs across 8 XCDs.
+//   k_row_norms, k_project_adam (gradient of w/max(||w||,eps)), k_bias_adam
 
 #include <hip/hip_runtime.h>
 
@@ -30,10 +34,10 @@
 #define BN 128
 #define BK 32
 #define BMP (BM + 1)  // padded LDS stride for transposed staging
-#define NTHREADS 256
+#define NTHREADS 512
+#define NXCD 8
 
 typedef float f32x16 __attribute__((ext_vector_type(16)));
-typedef float f32x4n __attribute__((ext_vector_type(4)));
 
 // C/D fragment mapping for 32x32 MFMA (guide §3): reg r, lane l ->
 //   row = (r&3) + 8*(r>>2) + 4*(l>>5), col = l&31
@@ -41,28 +45,50 @@ __device__ __forceinline__ int acc_row(int r, int lane) {
   return (r & 3) + 8 * (r >> 2) + 4 * (lane >> 5);
 }
 
+// Bijective XCD-contiguous remap of a linear block id (guide T1; bijective
+// variant for nwg % 8 != 0).
+__device__ __forceinline__ long xcd_swizzle(long id, long nwg) {
+  long q = nwg / NXCD, r = nwg % NXCD;
+  long xcd = id % NXCD, pos = id / NXCD;
+  return (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + pos;
+}
+
+// Decompose blockIdx (x = col tile, y = row tile, z = model) after swizzling
+// the (x, y) plane per model so that co-XCD blocks share operand panels.
+__device__ __forceinline__ void tile_coords(int& tx, int& ty) {
+  long nwg = (long)gridDim.x * gridDim.y;
+  long id = (long)blockIdx.x + (long)gridDim.x * blockIdx.y;
+  long s = xcd_swizzle(id, nwg);
+  tx = (int)(s % gridDim.x);
+  ty = (int)(s / gridDim.x);
+}
+
 // ---------------------------------------------------------------------------
-// staging
+// staging: load (global -> regs) and write (regs -> LDS) split so the loads
+// overlap the previous tile's MFMA phase.
 // ---------------------------------------------------------------------------
 
-// Transposed stage: lds[k][i] = src[i0+i][k0+k] for i in [0,ROWS), k in [0,BK)
-// src is row-major [n_rows, ld].  Guarded (zero-fill outside).  Conflict-free
-// by the +1 pad: bank(lds[k][i]) = (k*BMP + i) % 32 = (k + i) % 32 and each
-// 32-lane group covers (4*kidx + s + i) = all residues once.
-template <int ROWS>
-__device__ __forceinline__ void stage_T(const float* __restrict__ src, long ld,
-                                        int i0, int k0, int n_rows, int n_k,
-                                        float* __restrict__ lds) {
+// Transposed tile: lds[k][i] = src[i0+i][k0+k] (* scale[i0+i]), i<128, k<BK.
+// 2 passes x 512 threads x float4.
+struct TStage {
+  float4 v[2];
+  float sc[2];
+};
+
+__device__ __forceinline__ void stage_T_load(const float* __restrict__ src, long ld,
+                                             int i0, int k0, int n_rows, int n_k,
+                                             const float* __restrict__ scale,
+                                             TStage& st) {
   const int t = threadIdx.x;
 #pragma unroll
-  for (int p = 0; p < ROWS * BK / (NTHREADS * 4); ++p) {
-    int i = p * (NTHREADS / 8) + t / 8;   // 32 rows per pass
+  for (int p = 0; p < 2; ++p) {
+    int i = p * 64 + t / 8;
     int kc = (t % 8) * 4;
     int gi = i0 + i;
     float4 v = make_float4(0.f, 0.f, 0.f, 0.f);
+    float sc = 1.f;
     if (gi < n_rows) {
       const float* row = src + (long)gi * ld + k0 + kc;
-      // guard k range
       if (k0 + kc + 3 < n_k) {
         v = *reinterpret_cast<const float4*>(row);
       } else {
@@ -70,26 +96,47 @@ __device__ __forceinline__ void stage_T(const float* __restrict__ src, long ld,
         for (int s = 0; s < 4; ++s)
           if (k0 + kc + s < n_k) ((float*)&v)[s] = row[s];
       }
+      if (scale) sc = scale[gi];
+    }
+    st.v[p] = v;
+    st.sc[p] = sc;
+  }
+}
+
+__device__ __forceinline__ void stage_T_write(const TStage& st, float* __restrict__ lds,
+                                              bool scaled) {
+  const int t = threadIdx.x;
+#pragma unroll
+  for (int p = 0; p < 2; ++p) {
+    int i = p * 64 + t / 8;
+    int kc = (t % 8) * 4;
+    float4 v = st.v[p];
+    if (scaled) {
+      v.x *= st.sc[p]; v.y *= st.sc[p]; v.z *= st.sc[p]; v.w *= st.sc[p];
     }
 #pragma unroll
     for (int s = 0; s < 4; ++s) lds[(kc + s) * BMP + i] = ((float*)&v)[s];
   }
 }
 
-// Direct stage: lds[k][j] = src[k0+k][j0+j] * (scale ? scale[k0+k] : 1)
-// stride BM, float4 writes (16B-aligned, conflict-free for b128).
-template <int COLS>
-__device__ __forceinline__ void stage_D(const float* __restrict__ src, long ld,
-                                        int k0, int j0, int n_k, int n_cols,
-                                        const float* __restrict__ scale,
-                                        float* __restrict__ lds) {
+// Direct tile: lds[k][j] = src[k0+k][j0+j] (* scale[k0+k]), k<BK, j<128.
+struct DStage {
+  float4 v[2];
+  float sc[2];
+};
+
+__device__ __forceinline__ void stage_D_load(const float* __restrict__ src, long ld,
+                                             int k0, int j0, int n_k, int n_cols,
+                                             const float* __restrict__ scale,
+                                             DStage& st) {
   const int t = threadIdx.x;
 #pragma unroll
-  for (int p = 0; p < BK * COLS / (NTHREADS * 4); ++p) {
-    int k = p * (NTHREADS / 32) + t / 32;  // 8 k-rows per pass
+  for (int p = 0; p < 2; ++p) {
+    int k = p * 16 + t / 32;
     int j = (t % 32) * 4;
     int gk = k0 + k;
     float4 v = make_float4(0.f, 0.f, 0.f, 0.f);
+    float sc = 1.f;
     if (gk < n_k) {
       const float* row = src + (long)gk * ld + j0 + j;
       if (j0 + j + 3 < n_cols) {
@@ -99,82 +146,57 @@ __device__ __forceinline__ void stage_D(const float* __restrict__ src, long ld,
         for (int s = 0; s < 4; ++s)
           if (j0 + j + s < n_cols) ((float*)&v)[s] = row[s];
       }
-      if (scale) {
-        float sc = scale[gk];
-        v.x *= sc; v.y *= sc; v.z *= sc; v.w *= sc;
-      }
+      if (scale) sc = scale[gk];
+    }
+    st.v[p] = v;
+    st.sc[p] = sc;
+  }
+}
+
+__device__ __forceinline__ void stage_D_write(const DStage& st, float* __restrict__ lds,
+                                              bool scaled) {
+  const int t = threadIdx.x;
+#pragma unroll
+  for (int p = 0; p < 2; ++p) {
+    int k = p * 16 + t / 32;
+    int j = (t % 32) * 4;
+    float4 v = st.v[p];
+    if (scaled) {
+      v.x *= st.sc[p]; v.y *= st.sc[p]; v.z *= st.sc[p]; v.w *= st.sc[p];
     }
     *reinterpret_cast<float4*>(&lds[k * BM + j]) = v;
   }
 }
 
-// Apply a per-row (j-indexed) scale to a transposed-staged tile's source:
-// done at read time would cost per-MFMA VALU; instead scale during staging.
-template <int ROWS>
-__device__ __forceinline__ void stage_T_scaled(const float* __restrict__ src, long ld,
-                                               int i0, int k0, int n_rows, int n_k,
-                                               const float* __restrict__ scale,
-                                               float* __restrict__ lds) {
-  const int t = threadIdx.x;
-#pragma unroll
-  for (int p = 0; p < ROWS * BK / (NTHREADS * 4); ++p) {
-    int i = p * (NTHREADS / 8) + t / 8;
-    int kc = (t % 8) * 4;
-    int gi = i0 + i;
-    float4 v = make_float4(0.f, 0.f, 0.f, 0.f);
-    if (gi < n_rows) {
-      const float* row = src + (long)gi * ld + k0 + kc;
-      if (k0 + kc + 3 < n_k) {
-        v = *reinterpret_cast<const float4*>(row);
-      } else {
-#pragma unroll
-        for (int s = 0; s < 4; ++s)
-          if (k0 + kc + s < n_k) ((float*)&v)[s] = row[s];
-      }
-      float sc = scale[gi];
-      v.x *= sc; v.y *= sc; v.z *= sc; v.w *= sc;
-    }
-#pragma unroll
-    for (int s = 0; s < 4; ++s) lds[(kc + s) * BMP + i] = ((float*)&v)[s];
-  }
-}
-
 // ---------------------------------------------------------------------------
-// the MFMA block loop
+// MFMA phase: 8 waves; wave w covers rows [(w&3)*32, +32), cols [(w>>2)*64, +64)
 // ---------------------------------------------------------------------------
-// As: [BK][BMP or BM], Bs: [BK][BMP or BM].  Each of 4 waves computes the
-// 64x64 quadrant (wr, wc); acc[ti][tj] is the (32x32) sub-tile.
 template <int ASTRIDE, int BSTRIDE>
 __device__ __forceinline__ void mfma_tile(const float* __restrict__ As,
                                           const float* __restrict__ Bs,
-                                          f32x16 acc[2][2]) {
+                                          f32x16 acc[2]) {
   const int lane = threadIdx.x & (WAVE - 1);
   const int wave = threadIdx.x / WAVE;
-  const int wr = (wave >> 1) * 64;
-  const int wc = (wave & 1) * 64;
+  const int wr = (wave & 3) * 32;
+  const int wc = (wave >> 2) * 64;
   const int l31 = lane & 31;
-  const int h = lane >> 5;  // k sub-index
+  const int h = lane >> 5;
 
 #pragma unroll
   for (int kk = 0; kk < BK; kk += 2) {
     float a0 = As[(kk + h) * ASTRIDE + wr + l31];
-    float a1 = As[(kk + h) * ASTRIDE + wr + 32 + l31];
     float b0 = Bs[(kk + h) * BSTRIDE + wc + l31];
     float b1 = Bs[(kk + h) * BSTRIDE + wc + 32 + l31];
-    acc[0][0] = __builtin_amdgcn_mfma_f32_32x32x2f32(a0, b0, acc[0][0], 0, 0, 0);
-    acc[0][1] = __builtin_amdgcn_mfma_f32_32x32x2f32(a0, b1, acc[0][1], 0, 0, 0);
-    acc[1][0] = __builtin_amdgcn_mfma_f32_32x32x2f32(a1, b0, acc[1][0], 0, 0, 0);
-    acc[1][1] = __builtin_amdgcn_mfma_f32_32x32x2f32(a1, b1, acc[1][1], 0, 0, 0);
+    acc[0] = __builtin_amdgcn_mfma_f32_32x32x2f32(a0, b0, acc[0], 0, 0, 0);
+    acc[1] = __builtin_amdgcn_mfma_f32_32x32x2f32(a0, b1, acc[1], 0, 0, 0);
   }
 }
 
-__device__ __forceinline__ void zero_acc(f32x16 acc[2][2]) {
+__device__ __forceinline__ void zero_acc(f32x16 acc[2]) {
 #pragma unroll
-  for (int i = 0; i < 2; ++i)
+  for (int j = 0; j < 2; ++j)
 #pragma unroll
-    for (int j = 0; j < 2; ++j)
-#pragma unroll
-      for (int r = 0; r < 16; ++r) acc[i][j][r] = 0.f;
+    for (int r = 0; r < 16; ++r) acc[j][r] = 0.f;
 }
 
 __device__ __forceinline__ float wave_reduce_sum(float v) {
@@ -183,8 +205,22 @@ __device__ __forceinline__ float wave_reduce_sum(float v) {
   return v;
 }
 
+// epilogue lane geometry shared by all GEMM kernels
+struct EpiGeom {
+  int lane, wave, wr, wc, l31;
+};
+__device__ __forceinline__ EpiGeom epi_geom() {
+  EpiGeom g;
+  g.lane = threadIdx.x & (WAVE - 1);
+  g.wave = threadIdx.x / WAVE;
+  g.wr = (g.wave & 3) * 32;
+  g.wc = (g.wave >> 2) * 64;
+  g.l31 = g.lane & 31;
+  return g;
+}
+
 // ---------------------------------------------------------------------------
-// k_row_norms: norms[m][i] = ||W[m][i][:]||_2 ; inv[m][i] = 1/max(norm, eps)
+// k_row_norms
 // ---------------------------------------------------------------------------
 extern "C" __global__ void k_row_norms(const float* __restrict__ W,
                                        float* __restrict__ norms,
@@ -192,7 +228,7 @@ extern "C" __global__ void k_row_norms(const float* __restrict__ W,
                                        int n_rows_total, int d, float eps) {
   const int wave = threadIdx.x / WAVE;
   const int lane = threadIdx.x & (WAVE - 1);
-  const int row = blockIdx.x * 4 + wave;
+  const int row = blockIdx.x * (NTHREADS / WAVE) + wave;
   if (row >= n_rows_total) return;
   const float* w = W + (long)row * d;
   float s = 0.f;
@@ -213,9 +249,30 @@ extern "C" __global__ void k_row_norms(const float* __restrict__ W,
 }
 
 // ---------------------------------------------------------------------------
-// k_enc_fwd: out[b, j] = relu( sum_k x[b,k] * Wenc[j,k] * (tied? inv[j]:1) + bias[j] )
-// grid: (ceil(n/BN), ceil(B/BM), M)
-// epilogue extras: l1 partial sum -> loss_parts[m*2+1]; fired[m][j] += count
+// the shared prefetch K-loop (macro: operands differ per kernel)
+// ---------------------------------------------------------------------------
+#define PREFETCH_LOOP(K_TOTAL, LOAD_A, LOAD_B, WRITE_A, WRITE_B, ASTR, BSTR)   \
+  {                                                                            \
+    int k0 = 0;                                                                \
+    LOAD_A(k0);                                                                \
+    LOAD_B(k0);                                                                \
+    WRITE_A;                                                                   \
+    WRITE_B;                                                                   \
+    __syncthreads();                                                           \
+    for (k0 = BK; k0 < (K_TOTAL); k0 += BK) {                                  \
+      LOAD_A(k0);                                                              \
+      LOAD_B(k0);                                                              \
+      mfma_tile<ASTR, BSTR>(As, Bs, acc);                                      \
+      __syncthreads();                                                         \
+      WRITE_A;                                                                 \
+      WRITE_B;                                                                 \
+      __syncthreads();                                                         \
+    }                                                                          \
+    mfma_tile<ASTR, BSTR>(As, Bs, acc);                                        \
+  }
+
+// ---------------------------------------------------------------------------
+// k_enc_fwd
 // ---------------------------------------------------------------------------
 extern "C" __global__ __launch_bounds__(NTHREADS)
 void k_enc_fwd(const float* __restrict__ x,       // [B, d]
@@ -230,31 +287,25 @@ void k_enc_fwd(const float* __restrict__ x,       // [B, d]
   __shared__ float Bs[BK * BMP];
 
   const int m = blockIdx.z;
-  const int row0 = blockIdx.y * BM;   // batch rows
-  const int col0 = blockIdx.x * BN;   // dict rows (output cols)
+  int tx, ty;
+  tile_coords(tx, ty);
+  const int row0 = ty * BM;
+  const int col0 = tx * BN;
   const float* W = Wenc + (long)m * n * d;
   const float* inv = inv_norms ? inv_norms + (long)m * n : nullptr;
+  const bool scaled = inv != nullptr;
 
-  f32x16 acc[2][2];
+  f32x16 acc[2];
   zero_acc(acc);
+  TStage sa, sb;
 
-  for (int k0 = 0; k0 < d; k0 += BK) {
-    stage_T<BM>(x, d, row0, k0, B, d, As);
-    if (inv)
-      stage_T_scaled<BN>(W, d, col0, k0, n, d, inv, Bs);
-    else
-      stage_T<BN>(W, d, col0, k0, n, d, Bs);
-    __syncthreads();
-    mfma_tile<BMP, BMP>(As, Bs, acc);
-    __syncthreads();
-  }
+#define ENC_LA(K) stage_T_load(x, d, row0, (K), B, d, nullptr, sa)
+#define ENC_LB(K) stage_T_load(W, d, col0, (K), n, d, inv, sb)
+#define ENC_WA stage_T_write(sa, As, false)
+#define ENC_WB stage_T_write(sb, Bs, scaled)
+  PREFETCH_LOOP(d, ENC_LA, ENC_LB, ENC_WA, ENC_WB, BMP, BMP)
 
-  // epilogue: bias add, relu, store, l1 partial, fired counts
-  const int lane = threadIdx.x & (WAVE - 1);
-  const int wave = threadIdx.x / WAVE;
-  const int wr = (wave >> 1) * 64;
-  const int wc = (wave & 1) * 64;
-  const int l31 = lane & 31;
+  const EpiGeom g = epi_geom();
   float* c_m = c_out + (long)m * B * n;
   const float* bias_m = bias + (long)m * n;
   float* fired_m = fired + (long)m * n;
@@ -262,40 +313,32 @@ void k_enc_fwd(const float* __restrict__ x,       // [B, d]
   float l1_sum = 0.f;
 #pragma unroll
   for (int tj = 0; tj < 2; ++tj) {
-    int col = col0 + wc + tj * 32 + l31;
+    int col = col0 + g.wc + tj * 32 + g.l31;
     bool col_ok = col < n;
     float bj = col_ok ? bias_m[col] : 0.f;
     float fired_cnt = 0.f;
 #pragma unroll
-    for (int ti = 0; ti < 2; ++ti) {
-#pragma unroll
-      for (int r = 0; r < 16; ++r) {
-        int row = row0 + wr + ti * 32 + acc_row(r, lane);
-        if (row < B && col_ok) {
-          float v = acc[ti][tj][r] + bj;
-          v = fmaxf(v, 0.f);
-          c_m[(long)row * n + col] = v;
-          l1_sum += v;
-          fired_cnt += (v > 0.f) ? 1.f : 0.f;
-        }
+    for (int r = 0; r < 16; ++r) {
+      int row = row0 + g.wr + acc_row(r, g.lane);
+      if (row < B && col_ok) {
+        float v = fmaxf(acc[tj][r] + bj, 0.f);
+        c_m[(long)row * n + col] = v;
+        l1_sum += v;
+        fired_cnt += (v > 0.f) ? 1.f : 0.f;
       }
     }
-    if (col_ok && fired_cnt > 0.f) {
-      // combine the two half-wave contributions for this column? lanes l and
-      // l+32 hold DIFFERENT rows of the SAME column: merge via shfl to halve
-      // atomics
+    if (col_ok) {
       float other = __shfl_xor(fired_cnt, 32, WAVE);
-      if (lane < 32) atomicAdd(&fired_m[col], fired_cnt + other);
+      float tot = fired_cnt + other;
+      if (g.lane < 32 && tot > 0.f) atomicAdd(&fired_m[col], tot);
     }
   }
   l1_sum = wave_reduce_sum(l1_sum);
-  if (lane == 0) atomicAdd(&loss_parts[m * 2 + 1], l1_sum);
+  if (g.lane == 0) atomicAdd(&loss_parts[m * 2 + 1], l1_sum);
 }
 
 // ---------------------------------------------------------------------------
-// k_dec_fwd: r[b, j] = sum_k c[b,k] * Wdec[k,j] * inv[k]  -  x[b,j]
-// contraction over n (k index), output [B, d].
-// grid: (ceil(d/BN), ceil(B/BM), M).  epilogue: r store + MSE partial.
+// k_dec_fwd
 // ---------------------------------------------------------------------------
 extern "C" __global__ __launch_bounds__(NTHREADS)
 void k_dec_fwd(const float* __restrict__ c,       // [M, B, n]
@@ -309,57 +352,49 @@ void k_dec_fwd(const float* __restrict__ c,       // [M, B, n]
   __shared__ float Bs[BK * BM];
 
   const int m = blockIdx.z;
-  const int row0 = blockIdx.y * BM;  // batch rows
-  const int col0 = blockIdx.x * BN;  // d cols
+  int tx, ty;
+  tile_coords(tx, ty);
+  const int row0 = ty * BM;
+  const int col0 = tx * BN;
   const float* c_m = c + (long)m * B * n;
   const float* W = Wdec + (long)m * n * d;
   const float* inv = inv_norms + (long)m * n;
 
-  f32x16 acc[2][2];
+  f32x16 acc[2];
   zero_acc(acc);
+  TStage sa;
+  DStage sb;
 
-  for (int k0 = 0; k0 < n; k0 += BK) {
-    stage_T<BM>(c_m, n, row0, k0, B, n, As);
-    stage_D<BN>(W, d, k0, col0, n, d, inv, Bs);
-    __syncthreads();
-    mfma_tile<BMP, BM>(As, Bs, acc);
-    __syncthreads();
-  }
+#define DEC_LA(K) stage_T_load(c_m, n, row0, (K), B, n, nullptr, sa)
+#define DEC_LB(K) stage_D_load(W, d, (K), col0, n, d, inv, sb)
+#define DEC_WA stage_T_write(sa, As, false)
+#define DEC_WB stage_D_write(sb, Bs, true)
+  PREFETCH_LOOP(n, DEC_LA, DEC_LB, DEC_WA, DEC_WB, BMP, BM)
 
-  const int lane = threadIdx.x & (WAVE - 1);
-  const int wave = threadIdx.x / WAVE;
-  const int wr = (wave >> 1) * 64;
-  const int wc = (wave & 1) * 64;
-  const int l31 = lane & 31;
+  const EpiGeom g = epi_geom();
   float* r_m = r_out + (long)m * B * d;
 
   float mse_sum = 0.f;
 #pragma unroll
   for (int tj = 0; tj < 2; ++tj) {
-    int col = col0 + wc + tj * 32 + l31;
+    int col = col0 + g.wc + tj * 32 + g.l31;
     bool col_ok = col < d;
 #pragma unroll
-    for (int ti = 0; ti < 2; ++ti) {
-#pragma unroll
-      for (int r = 0; r < 16; ++r) {
-        int row = row0 + wr + ti * 32 + acc_row(r, lane);
-        if (row < B && col_ok) {
-          float rv = acc[ti][tj][r] - x[(long)row * d + col];
-          r_m[(long)row * d + col] = rv;
-          mse_sum += rv * rv;
-        }
+    for (int r = 0; r < 16; ++r) {
+      int row = row0 + g.wr + acc_row(r, g.lane);
+      if (row < B && col_ok) {
+        float rv = acc[tj][r] - x[(long)row * d + col];
+        r_m[(long)row * d + col] = rv;
+        mse_sum += rv * rv;
       }
     }
   }
   mse_sum = wave_reduce_sum(mse_sum);
-  if (lane == 0) atomicAdd(&loss_parts[m * 2 + 0], mse_sum);
+  if (g.lane == 0) atomicAdd(&loss_parts[m * 2 + 0], mse_sum);
 }
 
 // ---------------------------------------------------------------------------
-// k_gc: gpre[b, j] = (c[b,j] > 0) * ( gscale * sum_k r[b,k]*Wdec[j,k]*inv[j]
-//                                     + l1_alpha[m] / B )
-// grid: (ceil(n/BN), ceil(B/BM), M).  epilogue also accumulates
-// g_bias[m][j] += sum_b gpre[b, j] over this row-tile.
+// k_gc
 // ---------------------------------------------------------------------------
 extern "C" __global__ __launch_bounds__(NTHREADS)
 void k_gc(const float* __restrict__ r,        // [M, B, d]
@@ -374,8 +409,10 @@ void k_gc(const float* __restrict__ r,        // [M, B, d]
   __shared__ float Bs[BK * BMP];
 
   const int m = blockIdx.z;
-  const int row0 = blockIdx.y * BM;
-  const int col0 = blockIdx.x * BN;
+  int tx, ty;
+  tile_coords(tx, ty);
+  const int row0 = ty * BM;
+  const int col0 = tx * BN;
   const float* r_m = r + (long)m * B * d;
   const float* W = Wdec + (long)m * n * d;
   const float* inv = inv_norms + (long)m * n;
@@ -383,58 +420,45 @@ void k_gc(const float* __restrict__ r,        // [M, B, d]
   const float gscale = 2.0f / ((float)B * (float)d);
   const float l1_term = l1_alpha[m] / (float)B;
 
-  f32x16 acc[2][2];
+  f32x16 acc[2];
   zero_acc(acc);
+  TStage sa, sb;
 
-  for (int k0 = 0; k0 < d; k0 += BK) {
-    stage_T<BM>(r_m, d, row0, k0, B, d, As);
-    stage_T_scaled<BN>(W, d, col0, k0, n, d, inv, Bs);
-    __syncthreads();
-    mfma_tile<BMP, BMP>(As, Bs, acc);
-    __syncthreads();
-  }
+#define GC_LA(K) stage_T_load(r_m, d, row0, (K), B, d, nullptr, sa)
+#define GC_LB(K) stage_T_load(W, d, col0, (K), n, d, inv, sb)
+#define GC_WA stage_T_write(sa, As, false)
+#define GC_WB stage_T_write(sb, Bs, true)
+  PREFETCH_LOOP(d, GC_LA, GC_LB, GC_WA, GC_WB, BMP, BMP)
 
-  const int lane = threadIdx.x & (WAVE - 1);
-  const int wave = threadIdx.x / WAVE;
-  const int wr = (wave >> 1) * 64;
-  const int wc = (wave & 1) * 64;
-  const int l31 = lane & 31;
+  const EpiGeom g = epi_geom();
   float* g_m = gpre_out + (long)m * B * n;
   float* gb_m = g_bias + (long)m * n;
 
 #pragma unroll
   for (int tj = 0; tj < 2; ++tj) {
-    int col = col0 + wc + tj * 32 + l31;
+    int col = col0 + g.wc + tj * 32 + g.l31;
     bool col_ok = col < n;
     float colsum = 0.f;
 #pragma unroll
-    for (int ti = 0; ti < 2; ++ti) {
-#pragma unroll
-      for (int r_ = 0; r_ < 16; ++r_) {
-        int row = row0 + wr + ti * 32 + acc_row(r_, lane);
-        if (row < B && col_ok) {
-          float cv = c_m[(long)row * n + col];
-          float g = (cv > 0.f) ? (gscale * acc[ti][tj][r_] + l1_term) : 0.f;
-          g_m[(long)row * n + col] = g;
-          colsum += g;
-        }
+    for (int r_ = 0; r_ < 16; ++r_) {
+      int row = row0 + g.wr + acc_row(r_, g.lane);
+      if (row < B && col_ok) {
+        float cv = c_m[(long)row * n + col];
+        float gv = (cv > 0.f) ? (gscale * acc[tj][r_] + l1_term) : 0.f;
+        g_m[(long)row * n + col] = gv;
+        colsum += gv;
       }
     }
     if (col_ok) {
       float other = __shfl_xor(colsum, 32, WAVE);
-      if (lane < 32) {
-        float tot = colsum + other;
-        if (tot != 0.f) atomicAdd(&gb_m[col], tot);
-      }
+      float tot = colsum + other;
+      if (g.lane < 32 && tot != 0.f) atomicAdd(&gb_m[col], tot);
     }
   }
 }
 
 // ---------------------------------------------------------------------------
-// k_grad_w: gw[i, j] = beta * gw[i, j] + alpha * sum_b P[b, i] * Q[b, j]
-// P: [M, B, n] (i over n), Q: [M, B, d] or shared [B, d] (j over d).
-// Both operands stage DIRECT (contraction index b is the row index of both).
-// grid: (ceil(d/BN), ceil(n/BM), M)
+// k_grad_w
 // ---------------------------------------------------------------------------
 extern "C" __global__ __launch_bounds__(NTHREADS)
 void k_grad_w(const float* __restrict__ P, long p_batch_stride,
@@ -446,46 +470,38 @@ void k_grad_w(const float* __restrict__ P, long p_batch_stride,
   __shared__ float Bs[BK * BM];
 
   const int m = blockIdx.z;
-  const int row0 = blockIdx.y * BM;  // n rows of gw
-  const int col0 = blockIdx.x * BN;  // d cols
+  int tx, ty;
+  tile_coords(tx, ty);
+  const int row0 = ty * BM;  // n rows
+  const int col0 = tx * BN;  // d cols
   const float* P_m = P + (long)m * p_batch_stride;
   const float* Q_m = Q + (long)m * q_batch_stride;
 
-  f32x16 acc[2][2];
+  f32x16 acc[2];
   zero_acc(acc);
+  DStage sa, sb;
 
-  for (int k0 = 0; k0 < B; k0 += BK) {
-    // As[k][i] = P[k0+k][row0+i]
-    stage_D<BM>(P_m, n, k0, row0, B, n, nullptr, As);
-    // Bs[k][j] = Q[k0+k][col0+j]
-    stage_D<BN>(Q_m, d, k0, col0, B, d, nullptr, Bs);
-    __syncthreads();
-    mfma_tile<BM, BM>(As, Bs, acc);
-    __syncthreads();
-  }
+#define GW_LA(K) stage_D_load(P_m, n, (K), row0, B, n, nullptr, sa)
+#define GW_LB(K) stage_D_load(Q_m, d, (K), col0, B, d, nullptr, sb)
+#define GW_WA stage_D_write(sa, As, false)
+#define GW_WB stage_D_write(sb, Bs, false)
+  PREFETCH_LOOP(B, GW_LA, GW_LB, GW_WA, GW_WB, BM, BM)
 
-  const int lane = threadIdx.x & (WAVE - 1);
-  const int wave = threadIdx.x / WAVE;
-  const int wr = (wave >> 1) * 64;
-  const int wc = (wave & 1) * 64;
-  const int l31 = lane & 31;
+  const EpiGeom g = epi_geom();
   float* gw_m = gw + (long)m * n * d;
 
 #pragma unroll
   for (int tj = 0; tj < 2; ++tj) {
-    int col = col0 + wc + tj * 32 + l31;
+    int col = col0 + g.wc + tj * 32 + g.l31;
     bool col_ok = col < d;
 #pragma unroll
-    for (int ti = 0; ti < 2; ++ti) {
-#pragma unroll
-      for (int r = 0; r < 16; ++r) {
-        int row = row0 + wr + ti * 32 + acc_row(r, lane);
-        if (row < n && col_ok) {
-          long idx = (long)row * d + col;
-          float v = alpha * acc[ti][tj][r];
-          if (beta != 0.f) v += beta * gw_m[idx];
-          gw_m[idx] = v;
-        }
+    for (int r = 0; r < 16; ++r) {
+      int row = row0 + g.wr + acc_row(r, g.lane);
+      if (row < n && col_ok) {
+        long idx = (long)row * d + col;
+        float v = alpha * acc[tj][r];
+        if (beta != 0.f) v += beta * gw_m[idx];
+        gw_m[idx] = v;
       }
     }
   }
@@ -493,24 +509,22 @@ void k_grad_w(const float* __restrict__ P, long p_batch_stride,
 
 // ---------------------------------------------------------------------------
 // k_project_adam: per dictionary row i of model m:
-//   if project: g = (gw - (gw . w_hat) * w_hat * [norm>eps]) / max(norm,eps)
-//               (exact gradient of w / max(||w||, eps))
-//   Adam: mu = b1 mu + (1-b1) g ; nu = b2 nu + (1-b2) g^2
-//         w -= lr * (mu/(1-b1^t)) / (sqrt(nu/(1-b2^t)) + eps_adam)
-// One wave per row; 4 rows per block.  grid: (ceil(M*n/4))
+//   if project: g = (gw - (gw . w)/norm^2 * w * [norm>eps]) / max(norm,eps)
+//   Adam with per-model bias correction.
+// One wave per row; 8 rows per block.
 // ---------------------------------------------------------------------------
 extern "C" __global__ __launch_bounds__(NTHREADS)
 void k_project_adam(float* __restrict__ W,        // [M*n, d]
                     const float* __restrict__ gw, // [M*n, d]
                     const float* __restrict__ norms,  // [M*n]
                     float* __restrict__ mu, float* __restrict__ nu,
-                    const float* __restrict__ step_no,  // [M] (post-increment)
+                    const float* __restrict__ step_no,  // [M]
                     int n_rows_total, int n_per_model, int d,
                     float lr, float b1, float b2, float eps_adam,
                     float eps_norm, int project) {
   const int wave = threadIdx.x / WAVE;
   const int lane = threadIdx.x & (WAVE - 1);
-  const int row = blockIdx.x * 4 + wave;
+  const int row = blockIdx.x * (NTHREADS / WAVE) + wave;
   if (row >= n_rows_total) return;
 
   float* w = W + (long)row * d;
@@ -531,7 +545,6 @@ void k_project_adam(float* __restrict__ W,        // [M*n, d]
     inv_s = 1.0f / s;
     do_proj = nrm > eps_norm;
     if (do_proj) {
-      // dot = (g . w) / norm^2  (so that g_proj = (g - dot * w) / s)
       float acc = 0.f;
       for (int j = lane; j < d; j += WAVE) acc += g_in[j] * w[j];
       acc = wave_reduce_sum(acc);
@@ -549,15 +562,13 @@ void k_project_adam(float* __restrict__ W,        // [M*n, d]
     float v1 = b2 * nu_r[j] + (1.0f - b2) * g * g;
     mu_r[j] = m1;
     nu_r[j] = v1;
-    float upd = lr * (m1 / bc1) / (sqrtf(v1 / bc2) + eps_adam);
-    w[j] -= upd;
+    w[j] -= lr * (m1 / bc1) / (sqrtf(v1 / bc2) + eps_adam);
   }
 }
 
 // ---------------------------------------------------------------------------
-// k_bias_adam: Adam on the [M, n] bias with optional L2-norm decay gradient
-//   g = g_bias + bias_decay[m] * b / ||b||   (term skipped when ||b|| == 0)
-// grid: (M); block 256; each block handles one model's bias vector.
+// k_bias_adam: Adam on [M, n] bias with optional L2-norm decay gradient.
+// grid (M); block NTHREADS.
 // ---------------------------------------------------------------------------
 extern "C" __global__ __launch_bounds__(NTHREADS)
 void k_bias_adam(float* __restrict__ bias,        // [M, n]
@@ -576,16 +587,20 @@ void k_bias_adam(float* __restrict__ bias,        // [M, n]
   const float bc1 = 1.0f - powf(b1, t);
   const float bc2 = 1.0f - powf(b2, t);
 
+  __shared__ float partial[NTHREADS / WAVE];
   __shared__ float norm_sq_s;
   float decay_scale = 0.f;
   if (bd != 0.f) {
     float acc = 0.f;
     for (int j = threadIdx.x; j < n; j += NTHREADS) acc += b_m[j] * b_m[j];
     acc = wave_reduce_sum(acc);
-    __shared__ float partial[4];
     if ((threadIdx.x & (WAVE - 1)) == 0) partial[threadIdx.x / WAVE] = acc;
     __syncthreads();
-    if (threadIdx.x == 0) norm_sq_s = partial[0] + partial[1] + partial[2] + partial[3];
+    if (threadIdx.x == 0) {
+      float s = 0.f;
+      for (int wv = 0; wv < NTHREADS / WAVE; ++wv) s += partial[wv];
+      norm_sq_s = s;
+    }
     __syncthreads();
     float nrm = sqrtf(norm_sq_s);
     decay_scale = (nrm > 0.f) ? bd / nrm : 0.f;

@@ -22,7 +22,7 @@ void row_norms(torch::Tensor W, torch::Tensor norms, torch::Tensor inv_norms, do
   CHECK_IN(W); CHECK_IN(norms); CHECK_IN(inv_norms);
   long rows = W.numel() / W.size(-1);
   int d = W.size(-1);
-  dim3 grid(cdiv(rows, 4));
+  dim3 grid(cdiv(rows, NTHREADS / WAVE));
   hipLaunchKernelGGL(k_row_norms, grid, dim3(NTHREADS), 0, cur_stream(),
                      W.data_ptr<float>(), norms.data_ptr<float>(),
                      inv_norms.data_ptr<float>(), (int)rows, d, (float)eps);
@@ -108,7 +108,7 @@ void project_adam(torch::Tensor W, torch::Tensor gw, torch::Tensor norms,
   CHECK_IN(step_no);
   long rows = W.numel() / W.size(-1);
   int d = W.size(-1);
-  dim3 grid(cdiv(rows, 4));
+  dim3 grid(cdiv(rows, NTHREADS / WAVE));
   hipLaunchKernelGGL(k_project_adam, grid, dim3(NTHREADS), 0, cur_stream(),
                      W.data_ptr<float>(), gw.data_ptr<float>(),
                      norms.data_ptr<float>(), mu.data_ptr<float>(),
