@@ -251,24 +251,29 @@ extern "C" __global__ void k_row_norms(const float* __restrict__ W,
 // ---------------------------------------------------------------------------
 // the shared prefetch K-loop (macro: operands differ per kernel)
 // ---------------------------------------------------------------------------
+// Ping-pong double buffer + register prefetch: ONE barrier per K-step.
+// Iteration t: issue loads of tile t+1, MFMA on buffer `cur` (tile t), write
+// tile t+1 into buffer cur^1 (nobody reads it), barrier, flip.  The barrier
+// at the end of iteration t also protects the write of tile t+2 into the
+// old `cur` (every wave has finished reading it).
 #define PREFETCH_LOOP(K_TOTAL, LOAD_A, LOAD_B, WRITE_A, WRITE_B, ASTR, BSTR)   \
   {                                                                            \
-    int k0 = 0;                                                                \
-    LOAD_A(k0);                                                                \
-    LOAD_B(k0);                                                                \
-    WRITE_A;                                                                   \
-    WRITE_B;                                                                   \
+    int cur = 0;                                                               \
+    LOAD_A(0);                                                                 \
+    LOAD_B(0);                                                                 \
+    WRITE_A(0);                                                                \
+    WRITE_B(0);                                                                \
     __syncthreads();                                                           \
-    for (k0 = BK; k0 < (K_TOTAL); k0 += BK) {                                  \
+    for (int k0 = BK; k0 < (K_TOTAL); k0 += BK) {                              \
       LOAD_A(k0);                                                              \
       LOAD_B(k0);                                                              \
-      mfma_tile<ASTR, BSTR>(As, Bs, acc);                                      \
+      mfma_tile<ASTR, BSTR>(&As[cur][0], &Bs[cur][0], acc);                    \
+      WRITE_A(cur ^ 1);                                                        \
+      WRITE_B(cur ^ 1);                                                        \
       __syncthreads();                                                         \
-      WRITE_A;                                                                 \
-      WRITE_B;                                                                 \
-      __syncthreads();                                                         \
+      cur ^= 1;                                                                \
     }                                                                          \
-    mfma_tile<ASTR, BSTR>(As, Bs, acc);                                        \
+    mfma_tile<ASTR, BSTR>(&As[cur][0], &Bs[cur][0], acc);                      \
   }
 
 // ---------------------------------------------------------------------------
@@ -282,9 +287,10 @@ void k_enc_fwd(const float* __restrict__ x,       // [B, d]
                float* __restrict__ c_out,         // [M, B, n]
                float* __restrict__ loss_parts,    // [M, 2]
                float* __restrict__ fired,         // [M, n]
-               int B, int d, int n) {
-  __shared__ float As[BK * BMP];
-  __shared__ float Bs[BK * BMP];
+               int B, int d, int n,
+               int mode) {  // 0: bias+relu (+L1/fired); 1: raw scores (TopK)
+  __shared__ float As[2][BK * BMP];
+  __shared__ float Bs[2][BK * BMP];
 
   const int m = blockIdx.z;
   int tx, ty;
@@ -301,8 +307,8 @@ void k_enc_fwd(const float* __restrict__ x,       // [B, d]
 
 #define ENC_LA(K) stage_T_load(x, d, row0, (K), B, d, nullptr, sa)
 #define ENC_LB(K) stage_T_load(W, d, col0, (K), n, d, inv, sb)
-#define ENC_WA stage_T_write(sa, As, false)
-#define ENC_WB stage_T_write(sb, Bs, scaled)
+#define ENC_WA(BUF) stage_T_write(sa, &As[BUF][0], false)
+#define ENC_WB(BUF) stage_T_write(sb, &Bs[BUF][0], scaled)
   PREFETCH_LOOP(d, ENC_LA, ENC_LB, ENC_WA, ENC_WB, BMP, BMP)
 
   const EpiGeom g = epi_geom();
@@ -315,6 +321,15 @@ void k_enc_fwd(const float* __restrict__ x,       // [B, d]
   for (int tj = 0; tj < 2; ++tj) {
     int col = col0 + g.wc + tj * 32 + g.l31;
     bool col_ok = col < n;
+    if (mode == 1) {
+      // raw scores for TopK selection: no bias, no relu, no side outputs
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        int row = row0 + g.wr + acc_row(r, g.lane);
+        if (row < B && col_ok) c_m[(long)row * n + col] = acc[tj][r];
+      }
+      continue;
+    }
     float bj = col_ok ? bias_m[col] : 0.f;
     float fired_cnt = 0.f;
 #pragma unroll
@@ -333,8 +348,10 @@ void k_enc_fwd(const float* __restrict__ x,       // [B, d]
       if (g.lane < 32 && tot > 0.f) atomicAdd(&fired_m[col], tot);
     }
   }
-  l1_sum = wave_reduce_sum(l1_sum);
-  if (g.lane == 0) atomicAdd(&loss_parts[m * 2 + 1], l1_sum);
+  if (mode == 0) {
+    l1_sum = wave_reduce_sum(l1_sum);
+    if (g.lane == 0) atomicAdd(&loss_parts[m * 2 + 1], l1_sum);
+  }
 }
 
 // ---------------------------------------------------------------------------
@@ -348,8 +365,8 @@ void k_dec_fwd(const float* __restrict__ c,       // [M, B, n]
                float* __restrict__ r_out,         // [M, B, d]
                float* __restrict__ loss_parts,    // [M, 2]
                int B, int d, int n) {
-  __shared__ float As[BK * BMP];
-  __shared__ float Bs[BK * BM];
+  __shared__ float As[2][BK * BMP];
+  __shared__ float Bs[2][BK * BM];
 
   const int m = blockIdx.z;
   int tx, ty;
@@ -367,8 +384,8 @@ void k_dec_fwd(const float* __restrict__ c,       // [M, B, n]
 
 #define DEC_LA(K) stage_T_load(c_m, n, row0, (K), B, n, nullptr, sa)
 #define DEC_LB(K) stage_D_load(W, d, (K), col0, n, d, inv, sb)
-#define DEC_WA stage_T_write(sa, As, false)
-#define DEC_WB stage_D_write(sb, Bs, true)
+#define DEC_WA(BUF) stage_T_write(sa, &As[BUF][0], false)
+#define DEC_WB(BUF) stage_D_write(sb, &Bs[BUF][0], true)
   PREFETCH_LOOP(n, DEC_LA, DEC_LB, DEC_WA, DEC_WB, BMP, BM)
 
   const EpiGeom g = epi_geom();
@@ -405,8 +422,8 @@ void k_gc(const float* __restrict__ r,        // [M, B, d]
           float* __restrict__ gpre_out,       // [M, B, n]
           float* __restrict__ g_bias,         // [M, n]
           int B, int d, int n) {
-  __shared__ float As[BK * BMP];
-  __shared__ float Bs[BK * BMP];
+  __shared__ float As[2][BK * BMP];
+  __shared__ float Bs[2][BK * BMP];
 
   const int m = blockIdx.z;
   int tx, ty;
@@ -426,8 +443,8 @@ void k_gc(const float* __restrict__ r,        // [M, B, d]
 
 #define GC_LA(K) stage_T_load(r_m, d, row0, (K), B, d, nullptr, sa)
 #define GC_LB(K) stage_T_load(W, d, col0, (K), n, d, inv, sb)
-#define GC_WA stage_T_write(sa, As, false)
-#define GC_WB stage_T_write(sb, Bs, true)
+#define GC_WA(BUF) stage_T_write(sa, &As[BUF][0], false)
+#define GC_WB(BUF) stage_T_write(sb, &Bs[BUF][0], true)
   PREFETCH_LOOP(d, GC_LA, GC_LB, GC_WA, GC_WB, BMP, BMP)
 
   const EpiGeom g = epi_geom();
@@ -466,8 +483,8 @@ void k_grad_w(const float* __restrict__ P, long p_batch_stride,
               float* __restrict__ gw,  // [M, n, d]
               float alpha, float beta,
               int B, int n, int d) {
-  __shared__ float As[BK * BM];
-  __shared__ float Bs[BK * BM];
+  __shared__ float As[2][BK * BM];
+  __shared__ float Bs[2][BK * BM];
 
   const int m = blockIdx.z;
   int tx, ty;
@@ -483,8 +500,8 @@ void k_grad_w(const float* __restrict__ P, long p_batch_stride,
 
 #define GW_LA(K) stage_D_load(P_m, n, (K), row0, B, n, nullptr, sa)
 #define GW_LB(K) stage_D_load(Q_m, d, (K), col0, B, d, nullptr, sb)
-#define GW_WA stage_D_write(sa, As, false)
-#define GW_WB stage_D_write(sb, Bs, false)
+#define GW_WA(BUF) stage_D_write(sa, &As[BUF][0], false)
+#define GW_WB(BUF) stage_D_write(sb, &Bs[BUF][0], false)
   PREFETCH_LOOP(B, GW_LA, GW_LB, GW_WA, GW_WB, BM, BM)
 
   const EpiGeom g = epi_geom();

@@ -30,7 +30,7 @@ void row_norms(torch::Tensor W, torch::Tensor norms, torch::Tensor inv_norms, do
 
 void enc_fwd(torch::Tensor x, torch::Tensor Wenc, torch::Tensor bias,
              c10::optional<torch::Tensor> inv_norms, torch::Tensor c_out,
-             torch::Tensor loss_parts, torch::Tensor fired) {
+             torch::Tensor loss_parts, torch::Tensor fired, int64_t mode) {
   CHECK_IN(x); CHECK_IN(Wenc); CHECK_IN(bias); CHECK_IN(c_out);
   CHECK_IN(loss_parts); CHECK_IN(fired);
   int M = Wenc.size(0), n = Wenc.size(1), d = Wenc.size(2);
@@ -46,7 +46,7 @@ void enc_fwd(torch::Tensor x, torch::Tensor Wenc, torch::Tensor bias,
                      x.data_ptr<float>(), Wenc.data_ptr<float>(),
                      bias.data_ptr<float>(), inv, c_out.data_ptr<float>(),
                      loss_parts.data_ptr<float>(), fired.data_ptr<float>(),
-                     B, d, n);
+                     B, d, n, (int)mode);
 }
 
 void dec_fwd(torch::Tensor c, torch::Tensor Wdec, torch::Tensor inv_norms,

@@ -56,7 +56,7 @@ def test_enc_fwd_untied(shapes):
     c = torch.empty(M, B, n, device=DEV)
     loss_parts = torch.zeros(M, 2, device=DEV)
     fired = torch.zeros(M, n, device=DEV)
-    ext.enc_fwd(x, W, bias, None, c, loss_parts, fired)
+    ext.enc_fwd(x, W, bias, None, c, loss_parts, fired, 0)
     ref = torch.clamp(torch.einsum("mnd,bd->mbn", W, x) + bias[:, None, :], min=0)
     assert _rel_err(c, ref) < 1e-5
     assert _rel_err(loss_parts[:, 1], ref.sum(dim=(1, 2))) < 1e-4
@@ -77,7 +77,7 @@ def test_enc_fwd_tied_scaled(shapes):
     c = torch.empty(M, B, n, device=DEV)
     lp = torch.zeros(M, 2, device=DEV)
     fired = torch.zeros(M, n, device=DEV)
-    ext.enc_fwd(x, W, bias, inv, c, lp, fired)
+    ext.enc_fwd(x, W, bias, inv, c, lp, fired, 0)
     What = W / torch.clamp(torch.norm(W, dim=-1, keepdim=True), 1e-8)
     ref = torch.clamp(torch.einsum("mnd,bd->mbn", What, x), min=0)
     assert _rel_err(c, ref) < 1e-5
@@ -238,3 +238,57 @@ def test_bias_decay_gradient():
         l_ref, _ = ens_ref.step_batch(x)
     assert _rel_err(ens_hip.params["encoder_bias"], ens_ref.params["encoder_bias"]) < 2e-3
     assert _rel_err(l_hip["loss"], l_ref["loss"]) < 1e-4
+
+
+def test_topk_step_matches_torch():
+    """Fused TopK step vs the reference-semantics no_stacking vmap oracle."""
+    from sparse_coding_amd.engine.ensemble import FunctionalEnsemble
+    from sparse_coding_amd.functional.optim import adam
+    from sparse_coding_amd.models.topk import TopKEncoder
+
+    torch.manual_seed(11)
+    M, B, d, n = 2, 256, 64, 128
+    ks = (4, 16)
+    models = [TopKEncoder.init(d, n, k) for k in ks]
+    ens_hip = FunctionalEnsemble(models, TopKEncoder, adam, {"lr": 1e-3}, device=DEV, backend="hip")
+    assert type(ens_hip._hip_step).__name__ == "HipTopKStep"
+    models2 = [({k: v.clone() for k, v in p.items()}, {k: v.clone() for k, v in b.items()})
+               for p, b in ens_hip.unstack()]
+    ens_ref = FunctionalEnsemble(models2, TopKEncoder, adam, {"lr": 1e-3},
+                                 device=DEV, no_stacking=True, backend="torch")
+    x = torch.randn(B, d, device=DEV)
+    for i in range(4):
+        l_hip, aux_hip = ens_hip.step_batch(x)
+        l_ref, aux_ref = ens_ref.step_batch(x)
+        assert _rel_err(l_hip["loss"], l_ref["loss"]) < 1e-4, i
+        # sparsity respected
+        for m, k in enumerate(ks):
+            assert (aux_hip["c"][m] != 0).sum(dim=-1).max() <= k
+    assert _rel_err(ens_hip.params["dict"], ens_ref.params["dict"]) < 2e-3
+
+
+def test_resampler_gpu():
+    from sparse_coding_amd.engine.ensemble import FunctionalEnsemble
+    from sparse_coding_amd.engine.resample import EnsembleResampler
+    from sparse_coding_amd.functional.optim import adam
+    from sparse_coding_amd.models.sae_signatures import FunctionalTiedSAE
+
+    torch.manual_seed(12)
+    M, B, d, n = 2, 256, 64, 256
+    models = [FunctionalTiedSAE.init(d, n, 1e-3, device=DEV) for _ in range(M)]
+    ens = FunctionalEnsemble(models, FunctionalTiedSAE, adam, {"lr": 1e-3}, device=DEV, backend="hip")
+    rs = EnsembleResampler(ens, n_track=64)
+    # kill half the features so they never fire
+    with torch.no_grad():
+        ens.params["encoder_bias"][:, : n // 2] = -1e6
+    x = torch.randn(B, d, device=DEV)
+    for _ in range(3):
+        _, aux = ens.step_batch(x)
+        rs.observe(x, aux)
+    assert (rs.fired[:, : n // 2] == 0).all()
+    before = ens.params["encoder"][:, : n // 2].clone()
+    counts = rs.resample()
+    assert (counts == 64).all()  # n_track-limited
+    after = ens.params["encoder"][:, :64]
+    assert not torch.allclose(before[:, :64], after)
+    assert (ens.optim_states["mu"]["encoder"][0, :64] == 0).all()
