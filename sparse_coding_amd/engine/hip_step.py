@@ -119,6 +119,13 @@ class HipSAEStep:
 
         # persistent workspaces, sized lazily on first batch
         self._B = None
+        # hipGraph capture of the whole step (single-GPU path): replayed
+        # after 2 eager warmup steps; disabled via SPARSE_CODING_AMD_NO_GRAPH=1
+        import os as _os
+
+        self.use_graph = _os.environ.get("SPARSE_CODING_AMD_NO_GRAPH") != "1"
+        self._graph = None
+        self._eager_steps = 0
 
     # -- workspace management -------------------------------------------------
     def _alloc(self, B: int):
@@ -137,6 +144,8 @@ class HipSAEStep:
         if not self.tied:
             self.gw_enc = f(M, n, d)
         self._B = B
+        self._graph = None
+        self._eager_steps = 0
 
     # -- phases ---------------------------------------------------------------
     def grads_phase(self, x: torch.Tensor):
@@ -171,7 +180,6 @@ class HipSAEStep:
         else:
             ext.grad_w(self.c, self.r, self.gw, gscale, 0.0)
             ext.grad_w(self.gpre, x, self.gw_enc, 1.0, 0.0)
-        self._x = x
         return B
 
     def update_phase(self, B: int):
@@ -210,11 +218,34 @@ class HipSAEStep:
         return {"loss": total, "l_reconstruction": mse, "l_l1": l1, "l_bias_decay": l_bd}
 
     # -- public ---------------------------------------------------------------
+    def _capture(self, x: torch.Tensor) -> None:
+        try:
+            self.x_static = x.clone()
+            torch.cuda.synchronize()
+            g = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(g):
+                B = self.grads_phase(self.x_static)
+                self.update_phase(B)
+            self._graph = g
+        except Exception as e:  # noqa: BLE001 - graphs are an optimization only
+            print(f"[hip_step] hipGraph capture failed ({e}); staying eager")
+            self.use_graph = False
+            self._graph = None
+
     def step(self, minibatches: torch.Tensor, expand_dims: bool = True):
         if not expand_dims:
             raise NotImplementedError("per-model batches not supported by the HIP step")
+        B = minibatches.shape[0]
+        if self.use_graph:
+            if self._B == B and self._graph is None and self._eager_steps >= 2:
+                self._capture(minibatches.contiguous())
+            if self._graph is not None:
+                self.x_static.copy_(minibatches)
+                self._graph.replay()
+                return self._loss_data(B), {"c": self.c}
         B = self.grads_phase(minibatches)
         self.update_phase(B)
+        self._eager_steps += 1
         return self._loss_data(B), {"c": self.c}
 
     def dp_grad_tensors(self):
