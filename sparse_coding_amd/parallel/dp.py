@@ -131,6 +131,10 @@ class DataParallelEnsembleTrainer:
                     dist.broadcast(leaf, src=0, group=self.group)
 
     def step(self, local_batch: torch.Tensor):
+        if self.world_size == 1:
+            # single-process: take the ensemble's own step (hipGraph-captured
+            # on the fused path)
+            return self.ensemble.step_batch(local_batch)
         hs = getattr(self.ensemble, "_hip_step", None)
         if hs is not None:
             # fused path: split at the gradient boundary so the RCCL
