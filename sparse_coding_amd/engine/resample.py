@@ -36,8 +36,10 @@ class EnsembleResampler:
         self.ens = ensemble
         self.n_track = n_track
         self.encoder_norm_ratio = encoder_norm_ratio
-        M, n, d = ensemble.params["encoder"].shape
-        dev = ensemble.params["encoder"].device
+        # TopK ensembles name their weight "dict"; SAEs use "encoder"
+        self.w_key = "encoder" if "encoder" in ensemble.params else "dict"
+        M, n, d = ensemble.params[self.w_key].shape
+        dev = ensemble.params[self.w_key].device
         self.fired = torch.zeros(M, n, device=dev)
         self.worst_losses = torch.full((M, n_track), -float("inf"), device=dev)
         self.worst_examples = torch.zeros(M, n_track, d, device=dev)
@@ -61,7 +63,7 @@ class EnsembleResampler:
             from sparse_coding_amd.models.learned_dict import normalize_rows
 
             p = self.ens.params
-            w = p["encoder"] if "decoder" not in p else p["decoder"]
+            w = p.get("decoder", p[self.w_key])
             what = normalize_rows(w)
             x_hat = torch.einsum("mbn,mnd->mbd", c, what)
             per_ex = (x_hat - batch.unsqueeze(0)).pow(2).mean(dim=-1)
@@ -84,7 +86,8 @@ class EnsembleResampler:
         ens = self.ens
         p = ens.params
         st = ens.optim_states
-        M, n, d = p["encoder"].shape
+        wk = self.w_key
+        M, n, d = p[wk].shape
         counts = torch.zeros(M, dtype=torch.long)
 
         for m in range(M):
@@ -95,11 +98,11 @@ class EnsembleResampler:
             dead = dead[:k]
             worst = self.worst_examples[m, :k]
             worst_unit = worst / torch.clamp(torch.norm(worst, dim=-1, keepdim=True), 1e-8)
-            avg_norm = torch.norm(p["encoder"][m], dim=-1).mean()
+            avg_norm = torch.norm(p[wk][m], dim=-1).mean()
 
-            p["encoder"][m, dead] = worst_unit * self.encoder_norm_ratio * avg_norm
-            st["mu"]["encoder"][m, dead] = 0
-            st["nu"]["encoder"][m, dead] = 0
+            p[wk][m, dead] = worst_unit * self.encoder_norm_ratio * avg_norm
+            st["mu"][wk][m, dead] = 0
+            st["nu"][wk][m, dead] = 0
             if "decoder" in p:
                 p["decoder"][m, dead] = worst_unit
                 st["mu"]["decoder"][m, dead] = 0
