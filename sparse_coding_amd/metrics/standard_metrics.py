@@ -348,3 +348,58 @@ def plot_capacities(dicts: List[Tuple[LearnedDict, Dict[str, Any]]], show: bool 
     ax.set_title(f"Sum of capacities vs L1 alpha - {save_name}")
     fig.savefig(save_name + ".png")
     return fig
+
+
+# ---------------------------------------------------------------------------
+# multi-device metric scans (reference :711-808: mp.Pool over explicit
+# device lists computing per-layer activity/moment statistics)
+# ---------------------------------------------------------------------------
+
+def _layer_moments_job(args):
+    dict_path, chunk_path, device, hyperparam_filter = args
+    import torch as _t
+
+    dicts = _t.load(dict_path, map_location="cpu", weights_only=False)
+    acts = _t.load(chunk_path, map_location="cpu").float()
+    out = []
+    for ld, hp in dicts:
+        if hyperparam_filter and not all(hp.get(k) == v for k, v in hyperparam_filter.items()):
+            continue
+        ld.to_device(device)
+        times_active, mean, var, skew, kurt, m4 = calc_moments_streaming(ld, acts.to(device))
+        out.append(
+            {
+                "hyperparams": hp,
+                "prop_active": float((times_active > 0).float().mean()),
+                "mean_kurtosis": float(kurt[torch.isfinite(kurt)].mean()) if torch.isfinite(kurt).any() else float("nan"),
+                "mean_skew": float(skew[torch.isfinite(skew)].mean()) if torch.isfinite(skew).any() else float("nan"),
+            }
+        )
+    return out
+
+
+def scan_layer_moments(
+    dict_paths: List[str],
+    chunk_paths: List[str],
+    devices: Optional[List[str]] = None,
+    n_procs: int = 6,
+    hyperparam_filter: Optional[Dict[str, Any]] = None,
+):
+    """Per-layer activity/skew/kurtosis scans fanned over a process pool
+    pinned to devices (reference standard_metrics.py:742-808)."""
+    import multiprocessing as mp
+
+    if devices is None:
+        devices = (
+            [f"cuda:{i}" for i in range(torch.cuda.device_count())]
+            if torch.cuda.is_available()
+            else ["cpu"]
+        )
+    jobs = [
+        (dp, cp, devices[i % len(devices)], hyperparam_filter)
+        for i, (dp, cp) in enumerate(zip(dict_paths, chunk_paths))
+    ]
+    if len(jobs) == 1 or n_procs == 1:
+        return [_layer_moments_job(j) for j in jobs]
+    with mp.get_context("spawn").Pool(min(n_procs, len(jobs))) as pool:
+        return pool.map(_layer_moments_job, jobs)

@@ -139,3 +139,32 @@ def test_fvu_sparsity_plot_script(tmp_path):
     plot_fvu_sparsity(curves, save_path=str(tmp_path / "p.png"))
     assert os.path.exists(tmp_path / "p.png")
     assert np.isnan(pareto_auc([(1.0, 0.5)]))
+
+
+def test_ablation_graph_tiny():
+    from sparse_coding_amd.metrics.ablation import build_ablation_graph
+
+    model = _tiny_model()
+    ld0 = TiedSAE(torch.randn(8, 32), torch.zeros(8))
+    ld1 = TiedSAE(torch.randn(8, 32), torch.zeros(8))
+    tokens = torch.randint(0, 128, (2, 8))
+    dicts = {(0, "residual"): ld0, (1, "residual"): ld1}
+    graph = build_ablation_graph(
+        model, dicts, tokens,
+        features_to_ablate={(0, "residual"): [0, 1]},
+        target_features={(1, "residual"): [0, 1, 2]},
+        device="cpu",
+    )
+    assert len(graph) == 6
+    # ablating an upstream residual feature must perturb downstream features
+    assert any(v > 0 for v in graph.values())
+
+
+def test_scan_layer_moments(tmp_path):
+    from sparse_coding_amd.metrics.standard_metrics import scan_layer_moments
+
+    dicts = [(TiedSAE(torch.randn(8, 8), torch.zeros(8)), {"l1_alpha": 1e-3})]
+    torch.save(dicts, tmp_path / "ld.pt")
+    torch.save(torch.randn(500, 8, dtype=torch.float16), tmp_path / "chunk.pt")
+    res = scan_layer_moments([str(tmp_path / "ld.pt")], [str(tmp_path / "chunk.pt")], devices=["cpu"], n_procs=1)
+    assert len(res) == 1 and "prop_active" in res[0][0]
