@@ -148,8 +148,14 @@ class HipSAEStep:
         self._eager_steps = 0
 
     # -- phases ---------------------------------------------------------------
-    def grads_phase(self, x: torch.Tensor):
-        """Forward + backward: fills self.gw (+gw_enc), g_bias, loss parts."""
+    def grads_phase(self, x: torch.Tensor, on_grads=None):
+        """Forward + backward: fills self.gw (+gw_enc), g_bias, loss parts.
+
+        on_grads: optional callback(list_of_tensors) fired as gradient
+        tensors become final — used by the DP trainer to launch RCCL
+        all-reduces that overlap the remaining grad GEMMs (the [M,n,d]
+        weight grads are chunked over model halves for this).
+        """
         ens, ext = self.ens, self.ext
         M, n, d = self.n_models, self.n_dict, self.d_act
         B = x.shape[0]
@@ -173,13 +179,28 @@ class HipSAEStep:
         ext.gc(self.r, dict_w, self.inv_norms, self.c, self.l1_alpha,
                self.gpre, self.g_bias)
 
+        if on_grads is not None:
+            on_grads([self.g_bias])  # final after k_gc
+
         gscale = 2.0 / (B * d)
-        if self.tied:
+        if self.tied and on_grads is not None and M > 1:
+            # chunk the weight-grad GEMMs over model halves so the first
+            # half's all-reduce rides under the second half's compute
+            half = M // 2
+            for sl in (slice(0, half), slice(half, M)):
+                ext.grad_w(self.c[sl], self.r[sl], self.gw[sl], gscale, 0.0)
+                ext.grad_w(self.gpre[sl], x, self.gw[sl], 1.0, 1.0)
+                on_grads([self.gw[sl]])
+        elif self.tied:
             ext.grad_w(self.c, self.r, self.gw, gscale, 0.0)
             ext.grad_w(self.gpre, x, self.gw, 1.0, 1.0)
         else:
             ext.grad_w(self.c, self.r, self.gw, gscale, 0.0)
+            if on_grads is not None:
+                on_grads([self.gw])
             ext.grad_w(self.gpre, x, self.gw_enc, 1.0, 0.0)
+            if on_grads is not None:
+                on_grads([self.gw_enc])
         return B
 
     def update_phase(self, B: int):

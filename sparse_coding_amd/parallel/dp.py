@@ -137,13 +137,35 @@ class DataParallelEnsembleTrainer:
             return self.ensemble.step_batch(local_batch)
         hs = getattr(self.ensemble, "_hip_step", None)
         if hs is not None:
-            # fused path: split at the gradient boundary so the RCCL
-            # all-reduce sits between the grad GEMMs and the Adam kernels
-            B = hs.grads_phase(local_batch)
+            # fused path: gradient tensors are all-reduced as they become
+            # final (g_bias after k_gc; the [M,n,d] weight grads in model
+            # halves) on a side stream, overlapping the remaining grad GEMMs
+            works = []
+            comm_stream = None
+            if torch.cuda.is_available():
+                if not hasattr(self, "_comm_stream"):
+                    self._comm_stream = torch.cuda.Stream()
+                comm_stream = self._comm_stream
+
+            def on_grads(tensors):
+                if comm_stream is not None:
+                    ev = torch.cuda.Event()
+                    ev.record()
+                    comm_stream.wait_event(ev)
+                    with torch.cuda.stream(comm_stream):
+                        for t in tensors:
+                            works.append((dist.all_reduce(t, async_op=True, group=self.group), t))
+                else:
+                    for t in tensors:
+                        works.append((dist.all_reduce(t, async_op=True, group=self.group), t))
+
+            B = hs.grads_phase(local_batch, on_grads=on_grads if self.world_size > 1 else None)
             if self.world_size > 1:
-                works = [dist.all_reduce(t, async_op=True, group=self.group) for t in hs.dp_grad_tensors()]
-                for w, t in zip(works, hs.dp_grad_tensors()):
+                for w, t in works:
                     w.wait()
+                if comm_stream is not None:
+                    torch.cuda.current_stream().wait_stream(comm_stream)
+                for _, t in works:
                     t.div_(self.world_size)
             hs.update_phase(B)
             return hs._loss_data(B), {"c": hs.c}
