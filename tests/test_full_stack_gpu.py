@@ -147,3 +147,30 @@ def test_pythia14b_topk_resample():
     # topk leaves most features unfired on a fixed batch -> resampler acts
     counts = rs.resample()
     assert (counts > 0).all()
+
+
+def test_chunked_grads_match_unchunked():
+    """grads_phase(on_grads=...) with model-half chunked grad_w GEMMs must
+    produce the same gradients as the monolithic path (the multi-GPU
+    all-reduce-overlap path)."""
+    from sparse_coding_amd.engine.ensemble import FunctionalEnsemble
+    from sparse_coding_amd.functional.optim import adam
+    from sparse_coding_amd.models.sae_signatures import FunctionalTiedSAE
+
+    torch.manual_seed(4)
+    M, B, d, n = 4, 256, 64, 128
+    models = [FunctionalTiedSAE.init(d, n, 1e-3, device=DEV) for _ in range(M)]
+    ens = FunctionalEnsemble(models, FunctionalTiedSAE, adam, {"lr": 1e-3}, device=DEV, backend="hip")
+    hs = ens._hip_step
+    hs.use_graph = False
+    x = torch.randn(B, d, device=DEV)
+
+    hs.grads_phase(x)
+    gw_mono = hs.gw.clone()
+    gb_mono = hs.g_bias.clone()
+
+    seen = []
+    hs.grads_phase(x, on_grads=lambda ts: seen.extend(t.shape for t in ts))
+    assert torch.allclose(hs.gw, gw_mono, atol=1e-6)
+    assert torch.allclose(hs.g_bias, gb_mono, atol=1e-6)
+    assert len(seen) == 3  # g_bias + two model-half gw slices
