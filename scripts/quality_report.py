@@ -54,7 +54,7 @@ def main():
         batch_size=args.batch, feature_num_nonzero=args.nonzero,
         feature_prob_decay=1.0, correlated=False, device=device,
     )
-    l1s = np.logspace(-4, -2, args.n_models)
+    l1s = np.logspace(-4.5, -3.2, args.n_models)
     models = [FunctionalTiedSAE.init(d, n_dict, float(l1), device=device) for l1 in l1s]
     ens = FunctionalEnsemble(models, FunctionalTiedSAE, adam, {"lr": 1e-3}, device=device, backend=args.backend)
     backend = type(ens._hip_step).__name__ if ens._hip_step else "torch"

@@ -141,8 +141,13 @@ class HipSAEStep:
         self.fired = torch.zeros(M, n, device=dev)
         self.g_bias = f(M, n)
         self.gw = f(M, n, d)
+        # pre-transposed operands: make every GEMM's staging direct/b128
+        self.xT = f(d, B)
+        self.rT = f(M, d, B)
+        self.WT = f(M, d, n)
         if not self.tied:
             self.gw_enc = f(M, n, d)
+            self.WencT = f(M, d, n)
         self._B = B
         self._graph = None
         self._eager_steps = 0
@@ -173,11 +178,18 @@ class HipSAEStep:
         self._fired_step = self.fired  # accumulated across steps for resampling
 
         ext.row_norms(dict_w, self.norms, self.inv_norms, EPS_NORM)
-        ext.enc_fwd(x, enc, bias, self.inv_norms if self.tied else None,
-                    self.c, self.loss_parts, self.fired, 0)
+        # pre-transpose once per step so the big GEMMs stage all operands
+        # directly (b128 LDS writes): x^T, What^T (inv-norm pre-applied)
+        ext.transpose_scale(x, self.xT, None)
+        ext.transpose_scale(dict_w, self.WT, self.inv_norms)
+        if self.tied:
+            ext.enc_fwd2(self.xT, self.WT, bias, self.c, self.loss_parts, self.fired, 0)
+        else:
+            ext.transpose_scale(enc, self.WencT, None)
+            ext.enc_fwd2(self.xT, self.WencT, bias, self.c, self.loss_parts, self.fired, 0)
         ext.dec_fwd(self.c, dict_w, self.inv_norms, x, self.r, self.loss_parts)
-        ext.gc(self.r, dict_w, self.inv_norms, self.c, self.l1_alpha,
-               self.gpre, self.g_bias)
+        ext.transpose_scale(self.r, self.rT, None)
+        ext.gc2(self.rT, self.WT, self.c, self.l1_alpha, self.gpre, self.g_bias)
 
         if on_grads is not None:
             on_grads([self.g_bias])  # final after k_gc

@@ -632,3 +632,178 @@ void k_bias_adam(float* __restrict__ bias,        // [M, n]
     b_m[j] -= lr * (m1 / bc1) / (sqrtf(v1 / bc2) + eps_adam);
   }
 }
+
+// ---------------------------------------------------------------------------
+// k_transpose_scale: dst[c][r] = src[r][c] * (scale ? scale[r] : 1), batched
+// over grid.z with explicit strides.  64x64 LDS tiles, coalesced both sides.
+// Feeds the all-direct-staged GEMM variants below (x^T, r^T, What^T).
+// ---------------------------------------------------------------------------
+extern "C" __global__ __launch_bounds__(256)
+void k_transpose_scale(const float* __restrict__ src, float* __restrict__ dst,
+                       const float* __restrict__ scale,
+                       int R, int C,
+                       long src_mstride, long dst_mstride, long scale_mstride) {
+  __shared__ float tile[64][65];
+  const int m = blockIdx.z;
+  src += (long)m * src_mstride;
+  dst += (long)m * dst_mstride;
+  if (scale) scale += (long)m * scale_mstride;
+  const int r0 = blockIdx.y * 64;
+  const int c0 = blockIdx.x * 64;
+  const int t = threadIdx.x;
+
+#pragma unroll
+  for (int p = 0; p < 16; ++p) {
+    int r = p * 4 + t / 64;
+    int c = t % 64;
+    float v = 0.f;
+    if (r0 + r < R && c0 + c < C) {
+      v = src[(long)(r0 + r) * C + c0 + c];
+      if (scale) v *= scale[r0 + r];
+    }
+    tile[r][c] = v;
+  }
+  __syncthreads();
+#pragma unroll
+  for (int p = 0; p < 16; ++p) {
+    int c = p * 4 + t / 64;  // output row (= source column)
+    int r = t % 64;          // output col (= source row)
+    if (c0 + c < C && r0 + r < R) dst[(long)(c0 + c) * R + r0 + r] = tile[r][c];
+  }
+}
+
+// ---------------------------------------------------------------------------
+// k_enc_fwd2: enc forward with PRE-TRANSPOSED operands (xT [d,B], WT [M,d,n],
+// already inv-norm-scaled for tied) — both tiles stage DIRECT (b128 writes).
+// ---------------------------------------------------------------------------
+extern "C" __global__ __launch_bounds__(NTHREADS)
+void k_enc_fwd2(const float* __restrict__ xT,      // [d, B]
+                const float* __restrict__ WT,      // [M, d, n]
+                const float* __restrict__ bias,    // [M, n]
+                float* __restrict__ c_out,         // [M, B, n]
+                float* __restrict__ loss_parts,    // [M, 2]
+                float* __restrict__ fired,         // [M, n]
+                int B, int d, int n, int mode) {
+  __shared__ float As[2][BK * BM];
+  __shared__ float Bs[2][BK * BM];
+
+  const int m = blockIdx.z;
+  int tx, ty;
+  tile_coords(tx, ty);
+  const int row0 = ty * BM;   // batch rows
+  const int col0 = tx * BN;   // dict cols
+  const float* WT_m = WT + (long)m * d * n;
+
+  f32x16 acc[2];
+  zero_acc(acc);
+  DStage sa, sb;
+
+#define ENC2_LA(K) stage_D_load(xT, B, (K), row0, d, B, nullptr, sa)
+#define ENC2_LB(K) stage_D_load(WT_m, n, (K), col0, d, n, nullptr, sb)
+#define ENC2_WA(BUF) stage_D_write(sa, &As[BUF][0], false)
+#define ENC2_WB(BUF) stage_D_write(sb, &Bs[BUF][0], false)
+  PREFETCH_LOOP(d, ENC2_LA, ENC2_LB, ENC2_WA, ENC2_WB, BM, BM)
+
+  const EpiGeom g = epi_geom();
+  float* c_m = c_out + (long)m * B * n;
+  const float* bias_m = bias + (long)m * n;
+  float* fired_m = fired + (long)m * n;
+
+  float l1_sum = 0.f;
+#pragma unroll
+  for (int tj = 0; tj < 2; ++tj) {
+    int col = col0 + g.wc + tj * 32 + g.l31;
+    bool col_ok = col < n;
+    if (mode == 1) {
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        int row = row0 + g.wr + acc_row(r, g.lane);
+        if (row < B && col_ok) c_m[(long)row * n + col] = acc[tj][r];
+      }
+      continue;
+    }
+    float bj = col_ok ? bias_m[col] : 0.f;
+    float fired_cnt = 0.f;
+#pragma unroll
+    for (int r = 0; r < 16; ++r) {
+      int row = row0 + g.wr + acc_row(r, g.lane);
+      if (row < B && col_ok) {
+        float v = fmaxf(acc[tj][r] + bj, 0.f);
+        c_m[(long)row * n + col] = v;
+        l1_sum += v;
+        fired_cnt += (v > 0.f) ? 1.f : 0.f;
+      }
+    }
+    if (col_ok) {
+      float other = __shfl_xor(fired_cnt, 32, WAVE);
+      float tot = fired_cnt + other;
+      if (g.lane < 32 && tot > 0.f) atomicAdd(&fired_m[col], tot);
+    }
+  }
+  if (mode == 0) {
+    l1_sum = wave_reduce_sum(l1_sum);
+    if (g.lane == 0) atomicAdd(&loss_parts[m * 2 + 1], l1_sum);
+  }
+}
+
+// ---------------------------------------------------------------------------
+// k_gc2: code-grad with pre-transposed rT [M,d,B] and WT (scaled) [M,d,n].
+// ---------------------------------------------------------------------------
+extern "C" __global__ __launch_bounds__(NTHREADS)
+void k_gc2(const float* __restrict__ rT,       // [M, d, B]
+           const float* __restrict__ WT,       // [M, d, n] (inv-norm scaled)
+           const float* __restrict__ c,        // [M, B, n]
+           const float* __restrict__ l1_alpha, // [M]
+           float* __restrict__ gpre_out,       // [M, B, n]
+           float* __restrict__ g_bias,         // [M, n]
+           int B, int d, int n) {
+  __shared__ float As[2][BK * BM];
+  __shared__ float Bs[2][BK * BM];
+
+  const int m = blockIdx.z;
+  int tx, ty;
+  tile_coords(tx, ty);
+  const int row0 = ty * BM;
+  const int col0 = tx * BN;
+  const float* rT_m = rT + (long)m * d * B;
+  const float* WT_m = WT + (long)m * d * n;
+  const float* c_m = c + (long)m * B * n;
+  const float gscale = 2.0f / ((float)B * (float)d);
+  const float l1_term = l1_alpha[m] / (float)B;
+
+  f32x16 acc[2];
+  zero_acc(acc);
+  DStage sa, sb;
+
+#define GC2_LA(K) stage_D_load(rT_m, B, (K), row0, d, B, nullptr, sa)
+#define GC2_LB(K) stage_D_load(WT_m, n, (K), col0, d, n, nullptr, sb)
+#define GC2_WA(BUF) stage_D_write(sa, &As[BUF][0], false)
+#define GC2_WB(BUF) stage_D_write(sb, &Bs[BUF][0], false)
+  PREFETCH_LOOP(d, GC2_LA, GC2_LB, GC2_WA, GC2_WB, BM, BM)
+
+  const EpiGeom g = epi_geom();
+  float* g_m = gpre_out + (long)m * B * n;
+  float* gb_m = g_bias + (long)m * n;
+
+#pragma unroll
+  for (int tj = 0; tj < 2; ++tj) {
+    int col = col0 + g.wc + tj * 32 + g.l31;
+    bool col_ok = col < n;
+    float colsum = 0.f;
+#pragma unroll
+    for (int r_ = 0; r_ < 16; ++r_) {
+      int row = row0 + g.wr + acc_row(r_, g.lane);
+      if (row < B && col_ok) {
+        float cv = c_m[(long)row * n + col];
+        float gv = (cv > 0.f) ? (gscale * acc[tj][r_] + l1_term) : 0.f;
+        g_m[(long)row * n + col] = gv;
+        colsum += gv;
+      }
+    }
+    if (col_ok) {
+      float other = __shfl_xor(colsum, 32, WAVE);
+      float tot = colsum + other;
+      if (g.lane < 32 && tot != 0.f) atomicAdd(&gb_m[col], tot);
+    }
+  }
+}

@@ -131,8 +131,64 @@ void bias_adam(torch::Tensor bias, torch::Tensor g_bias, torch::Tensor decay,
                      (float)lr, (float)b1, (float)b2, (float)eps_adam);
 }
 
+void transpose_scale(torch::Tensor src, torch::Tensor dst,
+                     c10::optional<torch::Tensor> scale) {
+  CHECK_IN(src); CHECK_IN(dst);
+  int R, C, M;
+  long src_ms = 0, dst_ms = 0, scale_ms = 0;
+  if (src.dim() == 3) {
+    M = src.size(0); R = src.size(1); C = src.size(2);
+    src_ms = (long)R * C; dst_ms = (long)R * C;
+  } else {
+    M = 1; R = src.size(0); C = src.size(1);
+  }
+  const float* sc = nullptr;
+  if (scale.has_value()) {
+    CHECK_IN(scale.value());
+    sc = scale->data_ptr<float>();
+    scale_ms = (src.dim() == 3) ? R : 0;
+  }
+  dim3 grid(cdiv(C, 64), cdiv(R, 64), M);
+  hipLaunchKernelGGL(k_transpose_scale, grid, dim3(256), 0, cur_stream(),
+                     src.data_ptr<float>(), dst.data_ptr<float>(), sc,
+                     R, C, src_ms, dst_ms, scale_ms);
+}
+
+void enc_fwd2(torch::Tensor xT, torch::Tensor WT, torch::Tensor bias,
+              torch::Tensor c_out, torch::Tensor loss_parts,
+              torch::Tensor fired, int64_t mode) {
+  CHECK_IN(xT); CHECK_IN(WT); CHECK_IN(bias); CHECK_IN(c_out);
+  CHECK_IN(loss_parts); CHECK_IN(fired);
+  int M = WT.size(0), d = WT.size(1), n = WT.size(2);
+  int B = xT.size(1);
+  TORCH_CHECK(B % 4 == 0 && n % 4 == 0, "B and n must be multiples of 4");
+  dim3 grid(cdiv(n, BN), cdiv(B, BM), M);
+  hipLaunchKernelGGL(k_enc_fwd2, grid, dim3(NTHREADS), 0, cur_stream(),
+                     xT.data_ptr<float>(), WT.data_ptr<float>(),
+                     bias.data_ptr<float>(), c_out.data_ptr<float>(),
+                     loss_parts.data_ptr<float>(), fired.data_ptr<float>(),
+                     B, d, n, (int)mode);
+}
+
+void gc2(torch::Tensor rT, torch::Tensor WT, torch::Tensor c,
+         torch::Tensor l1_alpha, torch::Tensor gpre, torch::Tensor g_bias) {
+  CHECK_IN(rT); CHECK_IN(WT); CHECK_IN(c); CHECK_IN(l1_alpha);
+  CHECK_IN(gpre); CHECK_IN(g_bias);
+  int M = WT.size(0), d = WT.size(1), n = WT.size(2);
+  int B = rT.size(2);
+  dim3 grid(cdiv(n, BN), cdiv(B, BM), M);
+  hipLaunchKernelGGL(k_gc2, grid, dim3(NTHREADS), 0, cur_stream(),
+                     rT.data_ptr<float>(), WT.data_ptr<float>(),
+                     c.data_ptr<float>(), l1_alpha.data_ptr<float>(),
+                     gpre.data_ptr<float>(), g_bias.data_ptr<float>(),
+                     B, d, n);
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("row_norms", &row_norms, "dictionary row norms + clamped inverses");
+  m.def("transpose_scale", &transpose_scale, "batched [R,C]->[C,R] transpose with row scale");
+  m.def("enc_fwd2", &enc_fwd2, "enc forward, pre-transposed operands (all-direct staging)");
+  m.def("gc2", &gc2, "code-grad, pre-transposed operands");
   m.def("enc_fwd", &enc_fwd, "fused encoder GEMM + bias + ReLU (+L1, fired)");
   m.def("dec_fwd", &dec_fwd, "fused decoder GEMM - x (+MSE partial)");
   m.def("gc", &gc, "code-gradient GEMM + relu mask + l1 term (+bias grad)");
