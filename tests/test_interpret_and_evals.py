@@ -168,3 +168,34 @@ def test_scan_layer_moments(tmp_path):
     torch.save(torch.randn(500, 8, dtype=torch.float16), tmp_path / "chunk.pt")
     res = scan_layer_moments([str(tmp_path / "ld.pt")], [str(tmp_path / "chunk.pt")], devices=["cpu"], n_procs=1)
     assert len(res) == 1 and "prop_active" in res[0][0]
+
+
+def test_interpret_cli(tmp_path):
+    import interpret as interp_cli
+    from sparse_coding_amd.models.learned_dict import TiedSAE
+
+    ld = TiedSAE(torch.randn(16, 512), torch.zeros(16))  # pythia-70m d_model
+    ld_path = tmp_path / "ld.pt"
+    torch.save(ld, ld_path)
+    cache = str(tmp_path / "cache.pt")
+    interp_cli.main([
+        "make_fragments", "--learned-dict", str(ld_path), "--model-name", "pythia-70m",
+        "--layer", "1", "--n-fragments", "8", "--df-n-feats", "8",
+        "--cache", cache, "--device", "cpu",
+    ])
+    assert os.path.exists(cache)
+    out = str(tmp_path / "res")
+    interp_cli.main(["interpret", "--cache", cache, "--output-folder", out, "--n-feats-explain", "3"])
+    assert len(os.listdir(out)) == 3
+
+
+def test_deep_ae():
+    from sparse_coding_amd.analysis.deep_ae import DeepShrinkageAE
+
+    torch.manual_seed(0)
+    ae = DeepShrinkageAE(16, 32, depth=2)
+    x = torch.randn(64, 16)
+    loss0, *_ = ae(x)
+    ae.train_on([x] * 200, lr=5e-3)
+    loss1, *_ = ae(x)
+    assert loss1 < loss0
