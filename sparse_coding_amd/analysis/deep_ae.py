@@ -18,7 +18,9 @@ class ShrinkLayer(nn.Module):
     def __init__(self, d_in: int, d_out: int):
         super().__init__()
         self.lin = nn.Linear(d_in, d_out)
-        self.theta = nn.Parameter(torch.full((d_out,), 0.01))
+        # softplus(theta) is the shrinkage threshold; start it near 0.01
+        # (softplus^-1(0.01) ≈ -4.6) so codes are alive at init.
+        self.theta = nn.Parameter(torch.full((d_out,), -4.6))
 
     def forward(self, x):
         z = self.lin(x)
