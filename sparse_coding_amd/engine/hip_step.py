@@ -129,6 +129,9 @@ class HipSAEStep:
 
     # -- workspace management -------------------------------------------------
     def _alloc(self, B: int):
+        from sparse_coding_amd.ops.kconfig import kernel_config
+
+        self.kc = kernel_config()
         M, n, d = self.n_models, self.n_dict, self.d_act
         dev = self.ens.params["encoder"].device
         f = lambda *shape: torch.empty(shape, device=dev, dtype=torch.float32)
@@ -141,13 +144,15 @@ class HipSAEStep:
         self.fired = torch.zeros(M, n, device=dev)
         self.g_bias = f(M, n)
         self.gw = f(M, n, d)
-        # pre-transposed operands: make every GEMM's staging direct/b128
-        self.xT = f(d, B)
-        self.rT = f(M, d, B)
-        self.WT = f(M, d, n)
+        if self.kc["staging"] == "pre":
+            # pre-transposed operands: every GEMM's staging direct/b128
+            self.xT = f(d, B)
+            self.rT = f(M, d, B)
+            self.WT = f(M, d, n)
+            if not self.tied:
+                self.WencT = f(M, d, n)
         if not self.tied:
             self.gw_enc = f(M, n, d)
-            self.WencT = f(M, d, n)
         self._B = B
         self._graph = None
         self._eager_steps = 0
@@ -177,19 +182,31 @@ class HipSAEStep:
         self.g_bias.zero_()
         self._fired_step = self.fired  # accumulated across steps for resampling
 
+        kc = self.kc
+        bk, prio = kc["bk"], kc["prio"]
+        bk_dec = kc["bk_dec"] or bk
+        bk_gw = kc["bk_grad_w"] or bk
+
         ext.row_norms(dict_w, self.norms, self.inv_norms, EPS_NORM)
-        # pre-transpose once per step so the big GEMMs stage all operands
-        # directly (b128 LDS writes): x^T, What^T (inv-norm pre-applied)
-        ext.transpose_scale(x, self.xT, None)
-        ext.transpose_scale(dict_w, self.WT, self.inv_norms)
-        if self.tied:
-            ext.enc_fwd2(self.xT, self.WT, bias, self.c, self.loss_parts, self.fired, 0)
+        if kc["staging"] == "pre":
+            # pre-transpose once per step so the big GEMMs stage all operands
+            # directly (b128 LDS writes): x^T, What^T (inv-norm pre-applied)
+            ext.transpose_scale(x, self.xT, None)
+            ext.transpose_scale(dict_w, self.WT, self.inv_norms)
+            if self.tied:
+                ext.enc_fwd2(self.xT, self.WT, bias, self.c, self.loss_parts, self.fired, 0, bk, prio)
+            else:
+                ext.transpose_scale(enc, self.WencT, None)
+                ext.enc_fwd2(self.xT, self.WencT, bias, self.c, self.loss_parts, self.fired, 0, bk, prio)
+            ext.dec_fwd(self.c, dict_w, self.inv_norms, x, self.r, self.loss_parts, bk_dec, prio)
+            ext.transpose_scale(self.r, self.rT, None)
+            ext.gc2(self.rT, self.WT, self.c, self.l1_alpha, self.gpre, self.g_bias, bk, prio)
         else:
-            ext.transpose_scale(enc, self.WencT, None)
-            ext.enc_fwd2(self.xT, self.WencT, bias, self.c, self.loss_parts, self.fired, 0)
-        ext.dec_fwd(self.c, dict_w, self.inv_norms, x, self.r, self.loss_parts)
-        ext.transpose_scale(self.r, self.rT, None)
-        ext.gc2(self.rT, self.WT, self.c, self.l1_alpha, self.gpre, self.g_bias)
+            # transpose-in-staging GEMMs (no separate transpose kernels)
+            enc_inv = self.inv_norms if self.tied else None
+            ext.enc_fwd(x, enc, bias, enc_inv, self.c, self.loss_parts, self.fired, 0, bk, prio)
+            ext.dec_fwd(self.c, dict_w, self.inv_norms, x, self.r, self.loss_parts, bk_dec, prio)
+            ext.gc(self.r, dict_w, self.inv_norms, self.c, self.l1_alpha, self.gpre, self.g_bias, bk, prio)
 
         if on_grads is not None:
             on_grads([self.g_bias])  # final after k_gc
@@ -200,17 +217,17 @@ class HipSAEStep:
             # half's all-reduce rides under the second half's compute
             half = M // 2
             for sl in (slice(0, half), slice(half, M)):
-                ext.grad_w(self.c[sl], self.r[sl], self.gw[sl], gscale, 0.0)
-                ext.grad_w(self.gpre[sl], x, self.gw[sl], 1.0, 1.0)
+                ext.grad_w(self.c[sl], self.r[sl], self.gw[sl], gscale, 0.0, bk_gw, prio)
+                ext.grad_w(self.gpre[sl], x, self.gw[sl], 1.0, 1.0, bk_gw, prio)
                 on_grads([self.gw[sl]])
         elif self.tied:
-            ext.grad_w(self.c, self.r, self.gw, gscale, 0.0)
-            ext.grad_w(self.gpre, x, self.gw, 1.0, 1.0)
+            ext.grad_w(self.c, self.r, self.gw, gscale, 0.0, bk_gw, prio)
+            ext.grad_w(self.gpre, x, self.gw, 1.0, 1.0, bk_gw, prio)
         else:
-            ext.grad_w(self.c, self.r, self.gw, gscale, 0.0)
+            ext.grad_w(self.c, self.r, self.gw, gscale, 0.0, bk_gw, prio)
             if on_grads is not None:
                 on_grads([self.gw])
-            ext.grad_w(self.gpre, x, self.gw_enc, 1.0, 0.0)
+            ext.grad_w(self.gpre, x, self.gw_enc, 1.0, 0.0, bk_gw, prio)
             if on_grads is not None:
                 on_grads([self.gw_enc])
         return B

@@ -8,31 +8,38 @@
 // v_mfma_f32_32x32x2_f32 (157 TF peak, bitwise == an fmaf chain).
 //
 // GEMM core structure (tuned from rocprof on MI355X):
-//  * 128x128 output tile, BK=32, 8 waves (512 threads); each wave owns a
-//    32x64 quadrant as 2 v_mfma_f32_32x32x2_f32 accumulators (32 AGPRs) --
-//    low register pressure => 4 waves/SIMD, 2 blocks/CU co-resident.
-//  * register-prefetch pipeline (guide T14/G15): the NEXT K-tile's global
-//    loads issue before the MFMA phase of the current tile (latency hides
-//    under the 64-cyc/MFMA f32 pipe), the LDS write lands after the barrier.
+//  * 128x128 output tile, 8 waves (512 threads); each wave owns a 32x64
+//    quadrant as 2 v_mfma_f32_32x32x2_f32 accumulators (32 AGPRs).
+//  * K-tile depth TBK is a template parameter:
+//      TBK=32, __launch_bounds__(512,4): 64-66 KB LDS, 2 blocks/CU.
+//      TBK=16, __launch_bounds__(512,6): 32-33 KB LDS, <=80 VGPRs,
+//        3 blocks/CU — more co-resident blocks to hide barrier skew
+//        (SQ_WAIT_ANY was 25-33% of wave cycles at 2 blocks/CU).
+//    Both are compiled; the launcher picks at runtime (ops.kconfig).
+//  * ping-pong LDS double buffer + register prefetch (guide T14/G15): ONE
+//    barrier per K-step; the NEXT K-tile's global loads issue before the
+//    MFMA phase of the current tile.
 //  * LDS staging is conflict-free: transposed tiles use stride BM+1 (b32
-//    writes, (k+i)%32 covers every bank once per lane group); direct tiles
-//    use stride BM with float4 (b128) writes.
+//    writes); direct tiles use stride BM with float4 (b128) writes.
+//  * optional s_setprio(1) on the second-dispatched wave half (runtime
+//    `prio` flag; MI355X_MICROARCH "Two waves per SIMD" item 4).
 //  * XCD-aware bijective block swizzle (guide T1): consecutive remapped
 //    blocks land on one XCD and share operand panels in its private L2.
 //
-// Kernels:
-//   k_enc_fwd : c = relu(x @ Wenc^T + b)    (+ L1 partial, fired counts)
-//   k_dec_fwd : r = c @ Wdec_hat - x        (+ MSE partial)
-//   k_gc      : gpre = (c>0) .* (gs * r @ Wdec_hat^T + l1/B)  (+ bias-grad)
-//   k_grad_w  : gw = beta*gw + alpha * P^T @ Q   (K = batch contraction)
-//   k_row_norms, k_project_adam (gradient of w/max(||w||,eps)), k_bias_adam
+// Kernels (each a template over TBK, MINW):
+//   k_enc_fwd_t  : c = relu(x @ Wenc^T + b)    (+ L1 partial, fired counts)
+//   k_enc_fwd2_t : same with pre-transposed operands (all-direct staging)
+//   k_dec_fwd_t  : r = c @ Wdec_hat - x        (+ MSE partial)
+//   k_gc_t/k_gc2_t: gpre = (c>0) .* (gs * r @ Wdec_hat^T + l1/B) (+ bias-grad)
+//   k_grad_w_t   : gw = beta*gw + alpha * P^T @ Q   (K = batch contraction)
+//   k_row_norms, k_project_adam (gradient of w/max(||w||,eps)), k_bias_adam,
+//   k_transpose_scale
 
 #include <hip/hip_runtime.h>
 
 #define WAVE 64
 #define BM 128
 #define BN 128
-#define BK 32
 #define BMP (BM + 1)  // padded LDS stride for transposed staging
 #define NTHREADS 512
 #define NXCD 8
@@ -63,27 +70,37 @@ __device__ __forceinline__ void tile_coords(int& tx, int& ty) {
   ty = (int)(s / gridDim.x);
 }
 
+// Optional static priority for the second-dispatched wave half: the younger
+// waves lose VALU arbitration on every segment; one setprio before the main
+// loop, no per-segment flips.
+__device__ __forceinline__ void maybe_prio(int prio) {
+  if (prio && threadIdx.x >= NTHREADS / 2) __builtin_amdgcn_s_setprio(1);
+}
+
 // ---------------------------------------------------------------------------
 // staging: load (global -> regs) and write (regs -> LDS) split so the loads
-// overlap the previous tile's MFMA phase.
+// overlap the previous tile's MFMA phase.  TBK/16 float4 passes per thread.
 // ---------------------------------------------------------------------------
 
-// Transposed tile: lds[k][i] = src[i0+i][k0+k] (* scale[i0+i]), i<128, k<BK.
-// 2 passes x 512 threads x float4.
+// Transposed tile: lds[k][i] = src[i0+i][k0+k] (* scale[i0+i]), i<128, k<TBK.
+template <int TBK>
 struct TStage {
-  float4 v[2];
-  float sc[2];
+  float4 v[TBK / 16];
+  float sc[TBK / 16];
 };
 
+template <int TBK>
 __device__ __forceinline__ void stage_T_load(const float* __restrict__ src, long ld,
                                              int i0, int k0, int n_rows, int n_k,
                                              const float* __restrict__ scale,
-                                             TStage& st) {
+                                             TStage<TBK>& st) {
   const int t = threadIdx.x;
+  constexpr int TPR = TBK / 4;                 // threads covering one row's K
+  constexpr int RPP = NTHREADS / TPR;          // rows per pass
 #pragma unroll
-  for (int p = 0; p < 2; ++p) {
-    int i = p * 64 + t / 8;
-    int kc = (t % 8) * 4;
+  for (int p = 0; p < TBK / 16; ++p) {
+    int i = p * RPP + t / TPR;
+    int kc = (t % TPR) * 4;
     int gi = i0 + i;
     float4 v = make_float4(0.f, 0.f, 0.f, 0.f);
     float sc = 1.f;
@@ -103,13 +120,16 @@ __device__ __forceinline__ void stage_T_load(const float* __restrict__ src, long
   }
 }
 
-__device__ __forceinline__ void stage_T_write(const TStage& st, float* __restrict__ lds,
+template <int TBK>
+__device__ __forceinline__ void stage_T_write(const TStage<TBK>& st, float* __restrict__ lds,
                                               bool scaled) {
   const int t = threadIdx.x;
+  constexpr int TPR = TBK / 4;
+  constexpr int RPP = NTHREADS / TPR;
 #pragma unroll
-  for (int p = 0; p < 2; ++p) {
-    int i = p * 64 + t / 8;
-    int kc = (t % 8) * 4;
+  for (int p = 0; p < TBK / 16; ++p) {
+    int i = p * RPP + t / TPR;
+    int kc = (t % TPR) * 4;
     float4 v = st.v[p];
     if (scaled) {
       v.x *= st.sc[p]; v.y *= st.sc[p]; v.z *= st.sc[p]; v.w *= st.sc[p];
@@ -119,19 +139,21 @@ __device__ __forceinline__ void stage_T_write(const TStage& st, float* __restric
   }
 }
 
-// Direct tile: lds[k][j] = src[k0+k][j0+j] (* scale[k0+k]), k<BK, j<128.
+// Direct tile: lds[k][j] = src[k0+k][j0+j] (* scale[k0+k]), k<TBK, j<128.
+template <int TBK>
 struct DStage {
-  float4 v[2];
-  float sc[2];
+  float4 v[TBK / 16];
+  float sc[TBK / 16];
 };
 
+template <int TBK>
 __device__ __forceinline__ void stage_D_load(const float* __restrict__ src, long ld,
                                              int k0, int j0, int n_k, int n_cols,
                                              const float* __restrict__ scale,
-                                             DStage& st) {
+                                             DStage<TBK>& st) {
   const int t = threadIdx.x;
 #pragma unroll
-  for (int p = 0; p < 2; ++p) {
+  for (int p = 0; p < TBK / 16; ++p) {
     int k = p * 16 + t / 32;
     int j = (t % 32) * 4;
     int gk = k0 + k;
@@ -153,11 +175,12 @@ __device__ __forceinline__ void stage_D_load(const float* __restrict__ src, long
   }
 }
 
-__device__ __forceinline__ void stage_D_write(const DStage& st, float* __restrict__ lds,
+template <int TBK>
+__device__ __forceinline__ void stage_D_write(const DStage<TBK>& st, float* __restrict__ lds,
                                               bool scaled) {
   const int t = threadIdx.x;
 #pragma unroll
-  for (int p = 0; p < 2; ++p) {
+  for (int p = 0; p < TBK / 16; ++p) {
     int k = p * 16 + t / 32;
     int j = (t % 32) * 4;
     float4 v = st.v[p];
@@ -171,7 +194,7 @@ __device__ __forceinline__ void stage_D_write(const DStage& st, float* __restric
 // ---------------------------------------------------------------------------
 // MFMA phase: 8 waves; wave w covers rows [(w&3)*32, +32), cols [(w>>2)*64, +64)
 // ---------------------------------------------------------------------------
-template <int ASTRIDE, int BSTRIDE>
+template <int TBK, int ASTRIDE, int BSTRIDE>
 __device__ __forceinline__ void mfma_tile(const float* __restrict__ As,
                                           const float* __restrict__ Bs,
                                           f32x16 acc[2]) {
@@ -183,7 +206,7 @@ __device__ __forceinline__ void mfma_tile(const float* __restrict__ As,
   const int h = lane >> 5;
 
 #pragma unroll
-  for (int kk = 0; kk < BK; kk += 2) {
+  for (int kk = 0; kk < TBK; kk += 2) {
     float a0 = As[(kk + h) * ASTRIDE + wr + l31];
     float b0 = Bs[(kk + h) * BSTRIDE + wc + l31];
     float b1 = Bs[(kk + h) * BSTRIDE + wc + 32 + l31];
@@ -256,7 +279,7 @@ extern "C" __global__ void k_row_norms(const float* __restrict__ W,
 // tile t+1 into buffer cur^1 (nobody reads it), barrier, flip.  The barrier
 // at the end of iteration t also protects the write of tile t+2 into the
 // old `cur` (every wave has finished reading it).
-#define PREFETCH_LOOP(K_TOTAL, LOAD_A, LOAD_B, WRITE_A, WRITE_B, ASTR, BSTR)   \
+#define PREFETCH_LOOP(TBK, K_TOTAL, LOAD_A, LOAD_B, WRITE_A, WRITE_B, ASTR, BSTR) \
   {                                                                            \
     int cur = 0;                                                               \
     LOAD_A(0);                                                                 \
@@ -264,34 +287,37 @@ extern "C" __global__ void k_row_norms(const float* __restrict__ W,
     WRITE_A(0);                                                                \
     WRITE_B(0);                                                                \
     __syncthreads();                                                           \
-    for (int k0 = BK; k0 < (K_TOTAL); k0 += BK) {                              \
+    for (int k0 = (TBK); k0 < (K_TOTAL); k0 += (TBK)) {                        \
       LOAD_A(k0);                                                              \
       LOAD_B(k0);                                                              \
-      mfma_tile<ASTR, BSTR>(&As[cur][0], &Bs[cur][0], acc);                    \
+      mfma_tile<TBK, ASTR, BSTR>(&As[cur][0], &Bs[cur][0], acc);               \
       WRITE_A(cur ^ 1);                                                        \
       WRITE_B(cur ^ 1);                                                        \
       __syncthreads();                                                         \
       cur ^= 1;                                                                \
     }                                                                          \
-    mfma_tile<ASTR, BSTR>(&As[cur][0], &Bs[cur][0], acc);                      \
+    mfma_tile<TBK, ASTR, BSTR>(&As[cur][0], &Bs[cur][0], acc);                 \
   }
 
 // ---------------------------------------------------------------------------
-// k_enc_fwd
+// k_enc_fwd_t
 // ---------------------------------------------------------------------------
-extern "C" __global__ __launch_bounds__(NTHREADS)
-void k_enc_fwd(const float* __restrict__ x,       // [B, d]
-               const float* __restrict__ Wenc,    // [M, n, d]
-               const float* __restrict__ bias,    // [M, n]
-               const float* __restrict__ inv_norms, // [M, n] or nullptr
-               float* __restrict__ c_out,         // [M, B, n]
-               float* __restrict__ loss_parts,    // [M, 2]
-               float* __restrict__ fired,         // [M, n]
-               int B, int d, int n,
-               int mode) {  // 0: bias+relu (+L1/fired); 1: raw scores (TopK)
-  __shared__ float As[2][BK * BMP];
-  __shared__ float Bs[2][BK * BMP];
+template <int TBK, int MINW>
+__global__ __launch_bounds__(NTHREADS, MINW)
+void k_enc_fwd_t(const float* __restrict__ x,       // [B, d]
+                 const float* __restrict__ Wenc,    // [M, n, d]
+                 const float* __restrict__ bias,    // [M, n]
+                 const float* __restrict__ inv_norms, // [M, n] or nullptr
+                 float* __restrict__ c_out,         // [M, B, n]
+                 float* __restrict__ loss_parts,    // [M, 2]
+                 float* __restrict__ fired,         // [M, n]
+                 int B, int d, int n,
+                 int mode,  // 0: bias+relu (+L1/fired); 1: raw scores (TopK)
+                 int prio) {
+  __shared__ float As[2][TBK * BMP];
+  __shared__ float Bs[2][TBK * BMP];
 
+  maybe_prio(prio);
   const int m = blockIdx.z;
   int tx, ty;
   tile_coords(tx, ty);
@@ -303,13 +329,17 @@ void k_enc_fwd(const float* __restrict__ x,       // [B, d]
 
   f32x16 acc[2];
   zero_acc(acc);
-  TStage sa, sb;
+  TStage<TBK> sa, sb;
 
-#define ENC_LA(K) stage_T_load(x, d, row0, (K), B, d, nullptr, sa)
-#define ENC_LB(K) stage_T_load(W, d, col0, (K), n, d, inv, sb)
-#define ENC_WA(BUF) stage_T_write(sa, &As[BUF][0], false)
-#define ENC_WB(BUF) stage_T_write(sb, &Bs[BUF][0], scaled)
-  PREFETCH_LOOP(d, ENC_LA, ENC_LB, ENC_WA, ENC_WB, BMP, BMP)
+#define ENC_LA(K) stage_T_load<TBK>(x, d, row0, (K), B, d, nullptr, sa)
+#define ENC_LB(K) stage_T_load<TBK>(W, d, col0, (K), n, d, inv, sb)
+#define ENC_WA(BUF) stage_T_write<TBK>(sa, &As[BUF][0], false)
+#define ENC_WB(BUF) stage_T_write<TBK>(sb, &Bs[BUF][0], scaled)
+  PREFETCH_LOOP(TBK, d, ENC_LA, ENC_LB, ENC_WA, ENC_WB, BMP, BMP)
+#undef ENC_LA
+#undef ENC_LB
+#undef ENC_WA
+#undef ENC_WB
 
   const EpiGeom g = epi_geom();
   float* c_m = c_out + (long)m * B * n;
@@ -355,19 +385,21 @@ void k_enc_fwd(const float* __restrict__ x,       // [B, d]
 }
 
 // ---------------------------------------------------------------------------
-// k_dec_fwd
+// k_dec_fwd_t
 // ---------------------------------------------------------------------------
-extern "C" __global__ __launch_bounds__(NTHREADS)
-void k_dec_fwd(const float* __restrict__ c,       // [M, B, n]
-               const float* __restrict__ Wdec,    // [M, n, d]
-               const float* __restrict__ inv_norms, // [M, n]
-               const float* __restrict__ x,       // [B, d]
-               float* __restrict__ r_out,         // [M, B, d]
-               float* __restrict__ loss_parts,    // [M, 2]
-               int B, int d, int n) {
-  __shared__ float As[2][BK * BMP];
-  __shared__ float Bs[2][BK * BM];
+template <int TBK, int MINW>
+__global__ __launch_bounds__(NTHREADS, MINW)
+void k_dec_fwd_t(const float* __restrict__ c,       // [M, B, n]
+                 const float* __restrict__ Wdec,    // [M, n, d]
+                 const float* __restrict__ inv_norms, // [M, n]
+                 const float* __restrict__ x,       // [B, d]
+                 float* __restrict__ r_out,         // [M, B, d]
+                 float* __restrict__ loss_parts,    // [M, 2]
+                 int B, int d, int n, int prio) {
+  __shared__ float As[2][TBK * BMP];
+  __shared__ float Bs[2][TBK * BM];
 
+  maybe_prio(prio);
   const int m = blockIdx.z;
   int tx, ty;
   tile_coords(tx, ty);
@@ -379,14 +411,18 @@ void k_dec_fwd(const float* __restrict__ c,       // [M, B, n]
 
   f32x16 acc[2];
   zero_acc(acc);
-  TStage sa;
-  DStage sb;
+  TStage<TBK> sa;
+  DStage<TBK> sb;
 
-#define DEC_LA(K) stage_T_load(c_m, n, row0, (K), B, n, nullptr, sa)
-#define DEC_LB(K) stage_D_load(W, d, (K), col0, n, d, inv, sb)
-#define DEC_WA(BUF) stage_T_write(sa, &As[BUF][0], false)
-#define DEC_WB(BUF) stage_D_write(sb, &Bs[BUF][0], true)
-  PREFETCH_LOOP(n, DEC_LA, DEC_LB, DEC_WA, DEC_WB, BMP, BM)
+#define DEC_LA(K) stage_T_load<TBK>(c_m, n, row0, (K), B, n, nullptr, sa)
+#define DEC_LB(K) stage_D_load<TBK>(W, d, (K), col0, n, d, inv, sb)
+#define DEC_WA(BUF) stage_T_write<TBK>(sa, &As[BUF][0], false)
+#define DEC_WB(BUF) stage_D_write<TBK>(sb, &Bs[BUF][0], true)
+  PREFETCH_LOOP(TBK, n, DEC_LA, DEC_LB, DEC_WA, DEC_WB, BMP, BM)
+#undef DEC_LA
+#undef DEC_LB
+#undef DEC_WA
+#undef DEC_WB
 
   const EpiGeom g = epi_geom();
   float* r_m = r_out + (long)m * B * d;
@@ -411,20 +447,22 @@ void k_dec_fwd(const float* __restrict__ c,       // [M, B, n]
 }
 
 // ---------------------------------------------------------------------------
-// k_gc
+// k_gc_t
 // ---------------------------------------------------------------------------
-extern "C" __global__ __launch_bounds__(NTHREADS)
-void k_gc(const float* __restrict__ r,        // [M, B, d]
-          const float* __restrict__ Wdec,     // [M, n, d]
-          const float* __restrict__ inv_norms,// [M, n]
-          const float* __restrict__ c,        // [M, B, n]
-          const float* __restrict__ l1_alpha, // [M]
-          float* __restrict__ gpre_out,       // [M, B, n]
-          float* __restrict__ g_bias,         // [M, n]
-          int B, int d, int n) {
-  __shared__ float As[2][BK * BMP];
-  __shared__ float Bs[2][BK * BMP];
+template <int TBK, int MINW>
+__global__ __launch_bounds__(NTHREADS, MINW)
+void k_gc_t(const float* __restrict__ r,        // [M, B, d]
+            const float* __restrict__ Wdec,     // [M, n, d]
+            const float* __restrict__ inv_norms,// [M, n]
+            const float* __restrict__ c,        // [M, B, n]
+            const float* __restrict__ l1_alpha, // [M]
+            float* __restrict__ gpre_out,       // [M, B, n]
+            float* __restrict__ g_bias,         // [M, n]
+            int B, int d, int n, int prio) {
+  __shared__ float As[2][TBK * BMP];
+  __shared__ float Bs[2][TBK * BMP];
 
+  maybe_prio(prio);
   const int m = blockIdx.z;
   int tx, ty;
   tile_coords(tx, ty);
@@ -439,13 +477,17 @@ void k_gc(const float* __restrict__ r,        // [M, B, d]
 
   f32x16 acc[2];
   zero_acc(acc);
-  TStage sa, sb;
+  TStage<TBK> sa, sb;
 
-#define GC_LA(K) stage_T_load(r_m, d, row0, (K), B, d, nullptr, sa)
-#define GC_LB(K) stage_T_load(W, d, col0, (K), n, d, inv, sb)
-#define GC_WA(BUF) stage_T_write(sa, &As[BUF][0], false)
-#define GC_WB(BUF) stage_T_write(sb, &Bs[BUF][0], true)
-  PREFETCH_LOOP(d, GC_LA, GC_LB, GC_WA, GC_WB, BMP, BMP)
+#define GC_LA(K) stage_T_load<TBK>(r_m, d, row0, (K), B, d, nullptr, sa)
+#define GC_LB(K) stage_T_load<TBK>(W, d, col0, (K), n, d, inv, sb)
+#define GC_WA(BUF) stage_T_write<TBK>(sa, &As[BUF][0], false)
+#define GC_WB(BUF) stage_T_write<TBK>(sb, &Bs[BUF][0], true)
+  PREFETCH_LOOP(TBK, d, GC_LA, GC_LB, GC_WA, GC_WB, BMP, BMP)
+#undef GC_LA
+#undef GC_LB
+#undef GC_WA
+#undef GC_WB
 
   const EpiGeom g = epi_geom();
   float* g_m = gpre_out + (long)m * B * n;
@@ -475,17 +517,19 @@ void k_gc(const float* __restrict__ r,        // [M, B, d]
 }
 
 // ---------------------------------------------------------------------------
-// k_grad_w
+// k_grad_w_t
 // ---------------------------------------------------------------------------
-extern "C" __global__ __launch_bounds__(NTHREADS)
-void k_grad_w(const float* __restrict__ P, long p_batch_stride,
-              const float* __restrict__ Q, long q_batch_stride,
-              float* __restrict__ gw,  // [M, n, d]
-              float alpha, float beta,
-              int B, int n, int d) {
-  __shared__ float As[2][BK * BM];
-  __shared__ float Bs[2][BK * BM];
+template <int TBK, int MINW>
+__global__ __launch_bounds__(NTHREADS, MINW)
+void k_grad_w_t(const float* __restrict__ P, long p_batch_stride,
+                const float* __restrict__ Q, long q_batch_stride,
+                float* __restrict__ gw,  // [M, n, d]
+                float alpha, float beta,
+                int B, int n, int d, int prio) {
+  __shared__ float As[2][TBK * BM];
+  __shared__ float Bs[2][TBK * BM];
 
+  maybe_prio(prio);
   const int m = blockIdx.z;
   int tx, ty;
   tile_coords(tx, ty);
@@ -496,13 +540,17 @@ void k_grad_w(const float* __restrict__ P, long p_batch_stride,
 
   f32x16 acc[2];
   zero_acc(acc);
-  DStage sa, sb;
+  DStage<TBK> sa, sb;
 
-#define GW_LA(K) stage_D_load(P_m, n, (K), row0, B, n, nullptr, sa)
-#define GW_LB(K) stage_D_load(Q_m, d, (K), col0, B, d, nullptr, sb)
-#define GW_WA(BUF) stage_D_write(sa, &As[BUF][0], false)
-#define GW_WB(BUF) stage_D_write(sb, &Bs[BUF][0], false)
-  PREFETCH_LOOP(B, GW_LA, GW_LB, GW_WA, GW_WB, BM, BM)
+#define GW_LA(K) stage_D_load<TBK>(P_m, n, (K), row0, B, n, nullptr, sa)
+#define GW_LB(K) stage_D_load<TBK>(Q_m, d, (K), col0, B, d, nullptr, sb)
+#define GW_WA(BUF) stage_D_write<TBK>(sa, &As[BUF][0], false)
+#define GW_WB(BUF) stage_D_write<TBK>(sb, &Bs[BUF][0], false)
+  PREFETCH_LOOP(TBK, B, GW_LA, GW_LB, GW_WA, GW_WB, BM, BM)
+#undef GW_LA
+#undef GW_LB
+#undef GW_WA
+#undef GW_WB
 
   const EpiGeom g = epi_geom();
   float* gw_m = gw + (long)m * n * d;
@@ -673,20 +721,22 @@ void k_transpose_scale(const float* __restrict__ src, float* __restrict__ dst,
 }
 
 // ---------------------------------------------------------------------------
-// k_enc_fwd2: enc forward with PRE-TRANSPOSED operands (xT [d,B], WT [M,d,n],
+// k_enc_fwd2_t: enc forward with PRE-TRANSPOSED operands (xT [d,B], WT [M,d,n],
 // already inv-norm-scaled for tied) — both tiles stage DIRECT (b128 writes).
 // ---------------------------------------------------------------------------
-extern "C" __global__ __launch_bounds__(NTHREADS)
-void k_enc_fwd2(const float* __restrict__ xT,      // [d, B]
-                const float* __restrict__ WT,      // [M, d, n]
-                const float* __restrict__ bias,    // [M, n]
-                float* __restrict__ c_out,         // [M, B, n]
-                float* __restrict__ loss_parts,    // [M, 2]
-                float* __restrict__ fired,         // [M, n]
-                int B, int d, int n, int mode) {
-  __shared__ float As[2][BK * BM];
-  __shared__ float Bs[2][BK * BM];
+template <int TBK, int MINW>
+__global__ __launch_bounds__(NTHREADS, MINW)
+void k_enc_fwd2_t(const float* __restrict__ xT,      // [d, B]
+                  const float* __restrict__ WT,      // [M, d, n]
+                  const float* __restrict__ bias,    // [M, n]
+                  float* __restrict__ c_out,         // [M, B, n]
+                  float* __restrict__ loss_parts,    // [M, 2]
+                  float* __restrict__ fired,         // [M, n]
+                  int B, int d, int n, int mode, int prio) {
+  __shared__ float As[2][TBK * BM];
+  __shared__ float Bs[2][TBK * BM];
 
+  maybe_prio(prio);
   const int m = blockIdx.z;
   int tx, ty;
   tile_coords(tx, ty);
@@ -696,13 +746,17 @@ void k_enc_fwd2(const float* __restrict__ xT,      // [d, B]
 
   f32x16 acc[2];
   zero_acc(acc);
-  DStage sa, sb;
+  DStage<TBK> sa, sb;
 
-#define ENC2_LA(K) stage_D_load(xT, B, (K), row0, d, B, nullptr, sa)
-#define ENC2_LB(K) stage_D_load(WT_m, n, (K), col0, d, n, nullptr, sb)
-#define ENC2_WA(BUF) stage_D_write(sa, &As[BUF][0], false)
-#define ENC2_WB(BUF) stage_D_write(sb, &Bs[BUF][0], false)
-  PREFETCH_LOOP(d, ENC2_LA, ENC2_LB, ENC2_WA, ENC2_WB, BM, BM)
+#define ENC2_LA(K) stage_D_load<TBK>(xT, B, (K), row0, d, B, nullptr, sa)
+#define ENC2_LB(K) stage_D_load<TBK>(WT_m, n, (K), col0, d, n, nullptr, sb)
+#define ENC2_WA(BUF) stage_D_write<TBK>(sa, &As[BUF][0], false)
+#define ENC2_WB(BUF) stage_D_write<TBK>(sb, &Bs[BUF][0], false)
+  PREFETCH_LOOP(TBK, d, ENC2_LA, ENC2_LB, ENC2_WA, ENC2_WB, BM, BM)
+#undef ENC2_LA
+#undef ENC2_LB
+#undef ENC2_WA
+#undef ENC2_WB
 
   const EpiGeom g = epi_geom();
   float* c_m = c_out + (long)m * B * n;
@@ -747,19 +801,21 @@ void k_enc_fwd2(const float* __restrict__ xT,      // [d, B]
 }
 
 // ---------------------------------------------------------------------------
-// k_gc2: code-grad with pre-transposed rT [M,d,B] and WT (scaled) [M,d,n].
+// k_gc2_t: code-grad with pre-transposed rT [M,d,B] and WT (scaled) [M,d,n].
 // ---------------------------------------------------------------------------
-extern "C" __global__ __launch_bounds__(NTHREADS)
-void k_gc2(const float* __restrict__ rT,       // [M, d, B]
-           const float* __restrict__ WT,       // [M, d, n] (inv-norm scaled)
-           const float* __restrict__ c,        // [M, B, n]
-           const float* __restrict__ l1_alpha, // [M]
-           float* __restrict__ gpre_out,       // [M, B, n]
-           float* __restrict__ g_bias,         // [M, n]
-           int B, int d, int n) {
-  __shared__ float As[2][BK * BM];
-  __shared__ float Bs[2][BK * BM];
+template <int TBK, int MINW>
+__global__ __launch_bounds__(NTHREADS, MINW)
+void k_gc2_t(const float* __restrict__ rT,       // [M, d, B]
+             const float* __restrict__ WT,       // [M, d, n] (inv-norm scaled)
+             const float* __restrict__ c,        // [M, B, n]
+             const float* __restrict__ l1_alpha, // [M]
+             float* __restrict__ gpre_out,       // [M, B, n]
+             float* __restrict__ g_bias,         // [M, n]
+             int B, int d, int n, int prio) {
+  __shared__ float As[2][TBK * BM];
+  __shared__ float Bs[2][TBK * BM];
 
+  maybe_prio(prio);
   const int m = blockIdx.z;
   int tx, ty;
   tile_coords(tx, ty);
@@ -773,13 +829,17 @@ void k_gc2(const float* __restrict__ rT,       // [M, d, B]
 
   f32x16 acc[2];
   zero_acc(acc);
-  DStage sa, sb;
+  DStage<TBK> sa, sb;
 
-#define GC2_LA(K) stage_D_load(rT_m, B, (K), row0, d, B, nullptr, sa)
-#define GC2_LB(K) stage_D_load(WT_m, n, (K), col0, d, n, nullptr, sb)
-#define GC2_WA(BUF) stage_D_write(sa, &As[BUF][0], false)
-#define GC2_WB(BUF) stage_D_write(sb, &Bs[BUF][0], false)
-  PREFETCH_LOOP(d, GC2_LA, GC2_LB, GC2_WA, GC2_WB, BM, BM)
+#define GC2_LA(K) stage_D_load<TBK>(rT_m, B, (K), row0, d, B, nullptr, sa)
+#define GC2_LB(K) stage_D_load<TBK>(WT_m, n, (K), col0, d, n, nullptr, sb)
+#define GC2_WA(BUF) stage_D_write<TBK>(sa, &As[BUF][0], false)
+#define GC2_WB(BUF) stage_D_write<TBK>(sb, &Bs[BUF][0], false)
+  PREFETCH_LOOP(TBK, d, GC2_LA, GC2_LB, GC2_WA, GC2_WB, BM, BM)
+#undef GC2_LA
+#undef GC2_LB
+#undef GC2_WA
+#undef GC2_WB
 
   const EpiGeom g = epi_geom();
   float* g_m = gpre_out + (long)m * B * n;

@@ -30,7 +30,8 @@ void row_norms(torch::Tensor W, torch::Tensor norms, torch::Tensor inv_norms, do
 
 void enc_fwd(torch::Tensor x, torch::Tensor Wenc, torch::Tensor bias,
              c10::optional<torch::Tensor> inv_norms, torch::Tensor c_out,
-             torch::Tensor loss_parts, torch::Tensor fired, int64_t mode) {
+             torch::Tensor loss_parts, torch::Tensor fired, int64_t mode,
+             int64_t bk, bool prio) {
   CHECK_IN(x); CHECK_IN(Wenc); CHECK_IN(bias); CHECK_IN(c_out);
   CHECK_IN(loss_parts); CHECK_IN(fired);
   int M = Wenc.size(0), n = Wenc.size(1), d = Wenc.size(2);
@@ -42,45 +43,67 @@ void enc_fwd(torch::Tensor x, torch::Tensor Wenc, torch::Tensor bias,
     inv = inv_norms->data_ptr<float>();
   }
   dim3 grid(cdiv(n, BN), cdiv(B, BM), M);
-  hipLaunchKernelGGL(k_enc_fwd, grid, dim3(NTHREADS), 0, cur_stream(),
-                     x.data_ptr<float>(), Wenc.data_ptr<float>(),
-                     bias.data_ptr<float>(), inv, c_out.data_ptr<float>(),
-                     loss_parts.data_ptr<float>(), fired.data_ptr<float>(),
-                     B, d, n, (int)mode);
+  if (bk == 16)
+    hipLaunchKernelGGL((k_enc_fwd_t<16, 6>), grid, dim3(NTHREADS), 0, cur_stream(),
+                       x.data_ptr<float>(), Wenc.data_ptr<float>(),
+                       bias.data_ptr<float>(), inv, c_out.data_ptr<float>(),
+                       loss_parts.data_ptr<float>(), fired.data_ptr<float>(),
+                       B, d, n, (int)mode, prio ? 1 : 0);
+  else
+    hipLaunchKernelGGL((k_enc_fwd_t<32, 4>), grid, dim3(NTHREADS), 0, cur_stream(),
+                       x.data_ptr<float>(), Wenc.data_ptr<float>(),
+                       bias.data_ptr<float>(), inv, c_out.data_ptr<float>(),
+                       loss_parts.data_ptr<float>(), fired.data_ptr<float>(),
+                       B, d, n, (int)mode, prio ? 1 : 0);
 }
 
 void dec_fwd(torch::Tensor c, torch::Tensor Wdec, torch::Tensor inv_norms,
-             torch::Tensor x, torch::Tensor r_out, torch::Tensor loss_parts) {
+             torch::Tensor x, torch::Tensor r_out, torch::Tensor loss_parts,
+             int64_t bk, bool prio) {
   CHECK_IN(c); CHECK_IN(Wdec); CHECK_IN(inv_norms); CHECK_IN(x);
   CHECK_IN(r_out); CHECK_IN(loss_parts);
   int M = Wdec.size(0), n = Wdec.size(1), d = Wdec.size(2);
   int B = x.size(0);
   dim3 grid(cdiv(d, BN), cdiv(B, BM), M);
-  hipLaunchKernelGGL(k_dec_fwd, grid, dim3(NTHREADS), 0, cur_stream(),
-                     c.data_ptr<float>(), Wdec.data_ptr<float>(),
-                     inv_norms.data_ptr<float>(), x.data_ptr<float>(),
-                     r_out.data_ptr<float>(), loss_parts.data_ptr<float>(),
-                     B, d, n);
+  if (bk == 16)
+    hipLaunchKernelGGL((k_dec_fwd_t<16, 6>), grid, dim3(NTHREADS), 0, cur_stream(),
+                       c.data_ptr<float>(), Wdec.data_ptr<float>(),
+                       inv_norms.data_ptr<float>(), x.data_ptr<float>(),
+                       r_out.data_ptr<float>(), loss_parts.data_ptr<float>(),
+                       B, d, n, prio ? 1 : 0);
+  else
+    hipLaunchKernelGGL((k_dec_fwd_t<32, 4>), grid, dim3(NTHREADS), 0, cur_stream(),
+                       c.data_ptr<float>(), Wdec.data_ptr<float>(),
+                       inv_norms.data_ptr<float>(), x.data_ptr<float>(),
+                       r_out.data_ptr<float>(), loss_parts.data_ptr<float>(),
+                       B, d, n, prio ? 1 : 0);
 }
 
 void gc(torch::Tensor r, torch::Tensor Wdec, torch::Tensor inv_norms,
         torch::Tensor c, torch::Tensor l1_alpha, torch::Tensor gpre,
-        torch::Tensor g_bias) {
+        torch::Tensor g_bias, int64_t bk, bool prio) {
   CHECK_IN(r); CHECK_IN(Wdec); CHECK_IN(inv_norms); CHECK_IN(c);
   CHECK_IN(l1_alpha); CHECK_IN(gpre); CHECK_IN(g_bias);
   int M = Wdec.size(0), n = Wdec.size(1), d = Wdec.size(2);
   int B = r.size(1);
   dim3 grid(cdiv(n, BN), cdiv(B, BM), M);
-  hipLaunchKernelGGL(k_gc, grid, dim3(NTHREADS), 0, cur_stream(),
-                     r.data_ptr<float>(), Wdec.data_ptr<float>(),
-                     inv_norms.data_ptr<float>(), c.data_ptr<float>(),
-                     l1_alpha.data_ptr<float>(), gpre.data_ptr<float>(),
-                     g_bias.data_ptr<float>(), B, d, n);
+  if (bk == 16)
+    hipLaunchKernelGGL((k_gc_t<16, 6>), grid, dim3(NTHREADS), 0, cur_stream(),
+                       r.data_ptr<float>(), Wdec.data_ptr<float>(),
+                       inv_norms.data_ptr<float>(), c.data_ptr<float>(),
+                       l1_alpha.data_ptr<float>(), gpre.data_ptr<float>(),
+                       g_bias.data_ptr<float>(), B, d, n, prio ? 1 : 0);
+  else
+    hipLaunchKernelGGL((k_gc_t<32, 4>), grid, dim3(NTHREADS), 0, cur_stream(),
+                       r.data_ptr<float>(), Wdec.data_ptr<float>(),
+                       inv_norms.data_ptr<float>(), c.data_ptr<float>(),
+                       l1_alpha.data_ptr<float>(), gpre.data_ptr<float>(),
+                       g_bias.data_ptr<float>(), B, d, n, prio ? 1 : 0);
 }
 
 // gw[m] = beta * gw[m] + alpha * P[m]^T @ Q[m]; Q may be rank-shared [B, d]
 void grad_w(torch::Tensor P, torch::Tensor Q, torch::Tensor gw,
-            double alpha, double beta) {
+            double alpha, double beta, int64_t bk, bool prio) {
   CHECK_IN(P); CHECK_IN(Q); CHECK_IN(gw);
   int M = gw.size(0), n = gw.size(1), d = gw.size(2);
   int B;
@@ -94,10 +117,16 @@ void grad_w(torch::Tensor P, torch::Tensor Q, torch::Tensor gw,
     q_stride = 0;  // shared across models
   }
   dim3 grid(cdiv(d, BN), cdiv(n, BM), M);
-  hipLaunchKernelGGL(k_grad_w, grid, dim3(NTHREADS), 0, cur_stream(),
-                     P.data_ptr<float>(), p_stride, Q.data_ptr<float>(),
-                     q_stride, gw.data_ptr<float>(), (float)alpha,
-                     (float)beta, B, n, d);
+  if (bk == 16)
+    hipLaunchKernelGGL((k_grad_w_t<16, 6>), grid, dim3(NTHREADS), 0, cur_stream(),
+                       P.data_ptr<float>(), p_stride, Q.data_ptr<float>(),
+                       q_stride, gw.data_ptr<float>(), (float)alpha,
+                       (float)beta, B, n, d, prio ? 1 : 0);
+  else
+    hipLaunchKernelGGL((k_grad_w_t<32, 4>), grid, dim3(NTHREADS), 0, cur_stream(),
+                       P.data_ptr<float>(), p_stride, Q.data_ptr<float>(),
+                       q_stride, gw.data_ptr<float>(), (float)alpha,
+                       (float)beta, B, n, d, prio ? 1 : 0);
 }
 
 void project_adam(torch::Tensor W, torch::Tensor gw, torch::Tensor norms,
@@ -156,43 +185,75 @@ void transpose_scale(torch::Tensor src, torch::Tensor dst,
 
 void enc_fwd2(torch::Tensor xT, torch::Tensor WT, torch::Tensor bias,
               torch::Tensor c_out, torch::Tensor loss_parts,
-              torch::Tensor fired, int64_t mode) {
+              torch::Tensor fired, int64_t mode, int64_t bk, bool prio) {
   CHECK_IN(xT); CHECK_IN(WT); CHECK_IN(bias); CHECK_IN(c_out);
   CHECK_IN(loss_parts); CHECK_IN(fired);
   int M = WT.size(0), d = WT.size(1), n = WT.size(2);
   int B = xT.size(1);
   TORCH_CHECK(B % 4 == 0 && n % 4 == 0, "B and n must be multiples of 4");
   dim3 grid(cdiv(n, BN), cdiv(B, BM), M);
-  hipLaunchKernelGGL(k_enc_fwd2, grid, dim3(NTHREADS), 0, cur_stream(),
-                     xT.data_ptr<float>(), WT.data_ptr<float>(),
-                     bias.data_ptr<float>(), c_out.data_ptr<float>(),
-                     loss_parts.data_ptr<float>(), fired.data_ptr<float>(),
-                     B, d, n, (int)mode);
+  if (bk == 16)
+    hipLaunchKernelGGL((k_enc_fwd2_t<16, 6>), grid, dim3(NTHREADS), 0, cur_stream(),
+                       xT.data_ptr<float>(), WT.data_ptr<float>(),
+                       bias.data_ptr<float>(), c_out.data_ptr<float>(),
+                       loss_parts.data_ptr<float>(), fired.data_ptr<float>(),
+                       B, d, n, (int)mode, prio ? 1 : 0);
+  else
+    hipLaunchKernelGGL((k_enc_fwd2_t<32, 4>), grid, dim3(NTHREADS), 0, cur_stream(),
+                       xT.data_ptr<float>(), WT.data_ptr<float>(),
+                       bias.data_ptr<float>(), c_out.data_ptr<float>(),
+                       loss_parts.data_ptr<float>(), fired.data_ptr<float>(),
+                       B, d, n, (int)mode, prio ? 1 : 0);
 }
 
 void gc2(torch::Tensor rT, torch::Tensor WT, torch::Tensor c,
-         torch::Tensor l1_alpha, torch::Tensor gpre, torch::Tensor g_bias) {
+         torch::Tensor l1_alpha, torch::Tensor gpre, torch::Tensor g_bias,
+         int64_t bk, bool prio) {
   CHECK_IN(rT); CHECK_IN(WT); CHECK_IN(c); CHECK_IN(l1_alpha);
   CHECK_IN(gpre); CHECK_IN(g_bias);
   int M = WT.size(0), d = WT.size(1), n = WT.size(2);
   int B = rT.size(2);
   dim3 grid(cdiv(n, BN), cdiv(B, BM), M);
-  hipLaunchKernelGGL(k_gc2, grid, dim3(NTHREADS), 0, cur_stream(),
-                     rT.data_ptr<float>(), WT.data_ptr<float>(),
-                     c.data_ptr<float>(), l1_alpha.data_ptr<float>(),
-                     gpre.data_ptr<float>(), g_bias.data_ptr<float>(),
-                     B, d, n);
+  if (bk == 16)
+    hipLaunchKernelGGL((k_gc2_t<16, 6>), grid, dim3(NTHREADS), 0, cur_stream(),
+                       rT.data_ptr<float>(), WT.data_ptr<float>(),
+                       c.data_ptr<float>(), l1_alpha.data_ptr<float>(),
+                       gpre.data_ptr<float>(), g_bias.data_ptr<float>(),
+                       B, d, n, prio ? 1 : 0);
+  else
+    hipLaunchKernelGGL((k_gc2_t<32, 4>), grid, dim3(NTHREADS), 0, cur_stream(),
+                       rT.data_ptr<float>(), WT.data_ptr<float>(),
+                       c.data_ptr<float>(), l1_alpha.data_ptr<float>(),
+                       gpre.data_ptr<float>(), g_bias.data_ptr<float>(),
+                       B, d, n, prio ? 1 : 0);
 }
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("row_norms", &row_norms, "dictionary row norms + clamped inverses");
   m.def("transpose_scale", &transpose_scale, "batched [R,C]->[C,R] transpose with row scale");
-  m.def("enc_fwd2", &enc_fwd2, "enc forward, pre-transposed operands (all-direct staging)");
-  m.def("gc2", &gc2, "code-grad, pre-transposed operands");
-  m.def("enc_fwd", &enc_fwd, "fused encoder GEMM + bias + ReLU (+L1, fired)");
-  m.def("dec_fwd", &dec_fwd, "fused decoder GEMM - x (+MSE partial)");
-  m.def("gc", &gc, "code-gradient GEMM + relu mask + l1 term (+bias grad)");
-  m.def("grad_w", &grad_w, "gw = beta*gw + alpha * P^T Q (batched over M)");
+  m.def("enc_fwd2", &enc_fwd2, "enc forward, pre-transposed operands (all-direct staging)",
+        py::arg("xT"), py::arg("WT"), py::arg("bias"), py::arg("c_out"),
+        py::arg("loss_parts"), py::arg("fired"), py::arg("mode"),
+        py::arg("bk") = 32, py::arg("prio") = false);
+  m.def("gc2", &gc2, "code-grad, pre-transposed operands",
+        py::arg("rT"), py::arg("WT"), py::arg("c"), py::arg("l1_alpha"),
+        py::arg("gpre"), py::arg("g_bias"),
+        py::arg("bk") = 32, py::arg("prio") = false);
+  m.def("enc_fwd", &enc_fwd, "fused encoder GEMM + bias + ReLU (+L1, fired)",
+        py::arg("x"), py::arg("Wenc"), py::arg("bias"), py::arg("inv_norms"),
+        py::arg("c_out"), py::arg("loss_parts"), py::arg("fired"), py::arg("mode"),
+        py::arg("bk") = 32, py::arg("prio") = false);
+  m.def("dec_fwd", &dec_fwd, "fused decoder GEMM - x (+MSE partial)",
+        py::arg("c"), py::arg("Wdec"), py::arg("inv_norms"), py::arg("x"),
+        py::arg("r_out"), py::arg("loss_parts"),
+        py::arg("bk") = 32, py::arg("prio") = false);
+  m.def("gc", &gc, "code-gradient GEMM + relu mask + l1 term (+bias grad)",
+        py::arg("r"), py::arg("Wdec"), py::arg("inv_norms"), py::arg("c"),
+        py::arg("l1_alpha"), py::arg("gpre"), py::arg("g_bias"),
+        py::arg("bk") = 32, py::arg("prio") = false);
+  m.def("grad_w", &grad_w, "gw = beta*gw + alpha * P^T Q (batched over M)",
+        py::arg("P"), py::arg("Q"), py::arg("gw"), py::arg("alpha"), py::arg("beta"),
+        py::arg("bk") = 32, py::arg("prio") = false);
   m.def("project_adam", &project_adam, "renorm-gradient projection + Adam");
   m.def("bias_adam", &bias_adam, "Adam on bias with L2-norm decay");
 }

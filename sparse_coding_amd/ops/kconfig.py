@@ -1,0 +1,60 @@
+"""Runtime kernel-variant configuration for the fused SAE step.
+
+The GEMM kernels are compiled in two tile depths (see sae_kernels.hip):
+
+  bk=32  128x128 tile, 64-66 KB LDS, 2 blocks/CU  (default)
+  bk=16  128x128 tile, 32-33 KB LDS, 3 blocks/CU  (more co-resident blocks
+         to hide barrier skew; <=80 VGPRs via __launch_bounds__(512, 6))
+
+plus a `prio` flag (s_setprio(1) on the second-dispatched wave half) and a
+`staging` choice for the encoder/code-grad GEMMs:
+
+  staging="pre"  pre-transpose x/r/What once per step (k_transpose_scale),
+                 then all GEMM operands stage direct (float4 LDS writes)
+  staging="t"    transpose inside the GEMM staging (stride-BM+1 LDS, b32
+                 writes), no separate transpose kernels
+
+Defaults come from env (SC_AMD_BK / SC_AMD_PRIO / SC_AMD_STAGING) and can be
+overridden per-process via set_kernel_config() — scripts/ktune.py sweeps the
+combinations on a GPU box and the measured winner is baked into DEFAULTS.
+"""
+
+from __future__ import annotations
+
+import os
+from typing import Literal
+
+# Measured winners (MI355X, flagship bench shape; see profiles/README.md).
+DEFAULTS = {
+    "bk": 32,
+    "prio": False,
+    "staging": "t",
+    # per-kernel overrides, None -> use "bk"
+    "bk_grad_w": None,
+    "bk_dec": None,
+}
+
+_cfg = dict(DEFAULTS)
+
+if os.environ.get("SC_AMD_BK"):
+    _cfg["bk"] = int(os.environ["SC_AMD_BK"])
+if os.environ.get("SC_AMD_PRIO"):
+    _cfg["prio"] = os.environ["SC_AMD_PRIO"] == "1"
+if os.environ.get("SC_AMD_STAGING"):
+    _cfg["staging"] = os.environ["SC_AMD_STAGING"]
+if os.environ.get("SC_AMD_BK_GRAD_W"):
+    _cfg["bk_grad_w"] = int(os.environ["SC_AMD_BK_GRAD_W"])
+if os.environ.get("SC_AMD_BK_DEC"):
+    _cfg["bk_dec"] = int(os.environ["SC_AMD_BK_DEC"])
+
+
+def kernel_config() -> dict:
+    return dict(_cfg)
+
+
+def set_kernel_config(**kwargs) -> None:
+    for k, v in kwargs.items():
+        if k not in _cfg:
+            raise KeyError(f"unknown kernel-config key {k!r}; valid: {sorted(_cfg)}")
+        _cfg[k] = v
+    assert _cfg["bk"] in (16, 32) and _cfg["staging"] in ("pre", "t")
