@@ -14,8 +14,15 @@ import itertools
 import json
 import time
 
+import os
+import sys
+
 import numpy as np
 import torch
+
+_ROOT = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+if _ROOT not in sys.path:
+    sys.path.insert(0, _ROOT)
 
 
 def bench_combo(args, staging, bk, prio, bk_gw=None):

@@ -31,6 +31,21 @@ def shapes():
     return dict(M=3, B=192, d=96, n=160)
 
 
+# every test in this module runs under each compiled kernel variant:
+# (staging, tile depth TBK, younger-half s_setprio) — see ops/kconfig.py
+@pytest.fixture(autouse=True, params=[("t", 32, False), ("t", 16, True),
+                                      ("pre", 32, True), ("pre", 16, False)],
+                ids=lambda p: f"{p[0]}-bk{p[1]}-p{int(p[2])}")
+def kcfg(request):
+    from sparse_coding_amd.ops.kconfig import kernel_config, set_kernel_config
+
+    old = kernel_config()
+    staging, bk, prio = request.param
+    set_kernel_config(staging=staging, bk=bk, prio=prio)
+    yield {"bk": bk, "prio": prio, "staging": staging}
+    set_kernel_config(**old)
+
+
 def test_row_norms(shapes):
     ext = _ext()
     M, n, d = shapes["M"], shapes["n"], shapes["d"]
@@ -46,7 +61,7 @@ def test_row_norms(shapes):
     assert _rel_err(inv, ref_inv) < 1e-6
 
 
-def test_enc_fwd_untied(shapes):
+def test_enc_fwd_untied(shapes, kcfg):
     ext = _ext()
     M, B, d, n = shapes["M"], shapes["B"], shapes["d"], shapes["n"]
     torch.manual_seed(1)
@@ -56,7 +71,7 @@ def test_enc_fwd_untied(shapes):
     c = torch.empty(M, B, n, device=DEV)
     loss_parts = torch.zeros(M, 2, device=DEV)
     fired = torch.zeros(M, n, device=DEV)
-    ext.enc_fwd(x, W, bias, None, c, loss_parts, fired, 0)
+    ext.enc_fwd(x, W, bias, None, c, loss_parts, fired, 0, kcfg["bk"], kcfg["prio"])
     ref = torch.clamp(torch.einsum("mnd,bd->mbn", W, x) + bias[:, None, :], min=0)
     assert _rel_err(c, ref) < 1e-5
     assert _rel_err(loss_parts[:, 1], ref.sum(dim=(1, 2))) < 1e-4
@@ -64,7 +79,7 @@ def test_enc_fwd_untied(shapes):
     assert torch.equal(fired, ref_fired)
 
 
-def test_enc_fwd_tied_scaled(shapes):
+def test_enc_fwd_tied_scaled(shapes, kcfg):
     ext = _ext()
     M, B, d, n = shapes["M"], shapes["B"], shapes["d"], shapes["n"]
     torch.manual_seed(2)
@@ -77,13 +92,13 @@ def test_enc_fwd_tied_scaled(shapes):
     c = torch.empty(M, B, n, device=DEV)
     lp = torch.zeros(M, 2, device=DEV)
     fired = torch.zeros(M, n, device=DEV)
-    ext.enc_fwd(x, W, bias, inv, c, lp, fired, 0)
+    ext.enc_fwd(x, W, bias, inv, c, lp, fired, 0, kcfg["bk"], kcfg["prio"])
     What = W / torch.clamp(torch.norm(W, dim=-1, keepdim=True), 1e-8)
     ref = torch.clamp(torch.einsum("mnd,bd->mbn", What, x), min=0)
     assert _rel_err(c, ref) < 1e-5
 
 
-def test_dec_fwd(shapes):
+def test_dec_fwd(shapes, kcfg):
     ext = _ext()
     M, B, d, n = shapes["M"], shapes["B"], shapes["d"], shapes["n"]
     torch.manual_seed(3)
@@ -95,14 +110,14 @@ def test_dec_fwd(shapes):
     ext.row_norms(W, norms, inv, 1e-8)
     r = torch.empty(M, B, d, device=DEV)
     lp = torch.zeros(M, 2, device=DEV)
-    ext.dec_fwd(c, W, inv, x, r, lp)
+    ext.dec_fwd(c, W, inv, x, r, lp, kcfg["bk"], kcfg["prio"])
     What = W / torch.clamp(torch.norm(W, dim=-1, keepdim=True), 1e-8)
     ref_r = torch.einsum("mnd,mbn->mbd", What, c) - x
     assert _rel_err(r, ref_r) < 1e-5
     assert _rel_err(lp[:, 0], ref_r.pow(2).sum(dim=(1, 2))) < 1e-4
 
 
-def test_gc(shapes):
+def test_gc(shapes, kcfg):
     ext = _ext()
     M, B, d, n = shapes["M"], shapes["B"], shapes["d"], shapes["n"]
     torch.manual_seed(4)
@@ -115,7 +130,7 @@ def test_gc(shapes):
     ext.row_norms(W, norms, inv, 1e-8)
     gpre = torch.empty(M, B, n, device=DEV)
     g_bias = torch.zeros(M, n, device=DEV)
-    ext.gc(r, W, inv, c, l1, gpre, g_bias)
+    ext.gc(r, W, inv, c, l1, gpre, g_bias, kcfg["bk"], kcfg["prio"])
     What = W / torch.clamp(torch.norm(W, dim=-1, keepdim=True), 1e-8)
     gscale = 2.0 / (B * d)
     g = gscale * torch.einsum("mnd,mbd->mbn", What, r) + l1[:, None, None] / B
@@ -124,19 +139,19 @@ def test_gc(shapes):
     assert _rel_err(g_bias, ref.sum(dim=1)) < 1e-4
 
 
-def test_grad_w(shapes):
+def test_grad_w(shapes, kcfg):
     ext = _ext()
     M, B, d, n = shapes["M"], shapes["B"], shapes["d"], shapes["n"]
     torch.manual_seed(5)
     P = torch.randn(M, B, n, device=DEV)
     Q = torch.randn(M, B, d, device=DEV)
     gw = torch.zeros(M, n, d, device=DEV)
-    ext.grad_w(P, Q, gw, 0.5, 0.0)
+    ext.grad_w(P, Q, gw, 0.5, 0.0, kcfg["bk"], kcfg["prio"])
     ref = 0.5 * torch.einsum("mbn,mbd->mnd", P, Q)
     assert _rel_err(gw, ref) < 1e-5
     # beta accumulate + shared Q
     x = torch.randn(B, d, device=DEV)
-    ext.grad_w(P, x, gw, 1.0, 1.0)
+    ext.grad_w(P, x, gw, 1.0, 1.0, kcfg["bk"], kcfg["prio"])
     ref = ref + torch.einsum("mbn,bd->mnd", P, x)
     assert _rel_err(gw, ref) < 1e-5
 
