@@ -26,7 +26,9 @@ from typing import Literal
 
 # Measured winners (MI355X, flagship bench shape; see profiles/README.md).
 DEFAULTS = {
-    "bk": 32,
+    # ktune sweep r01 (profiles/README.md): t/bk16 639k acts/s vs t/bk32
+    # 628k, pre/bk16 632k, pre/bk32 629k; prio neutral.  bk16 = 4 blocks/CU.
+    "bk": 16,
     "prio": False,
     "staging": "t",
     # per-kernel overrides, None -> use "bk"
