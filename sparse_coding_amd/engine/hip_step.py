@@ -71,6 +71,10 @@ def maybe_make_step(ensemble, required: bool = False) -> Optional["HipSAEStep"]:
         ext = _ops.get_extension(required=True)
         return HipTopKStep(ensemble, ext)
 
+    if sig is sigs.FunctionalThresholdingSAE:
+        ext = _ops.get_extension(required=True)
+        return HipThresholdStep(ensemble, ext)
+
     tied = sig is sigs.FunctionalTiedSAE
     untied = sig is sigs.FunctionalSAE
     if not (tied or untied):
@@ -304,6 +308,91 @@ class HipSAEStep:
         if not self.tied:
             ts.append(self.gw_enc)
         return ts
+
+
+class HipThresholdStep(HipSAEStep):
+    """Fused step for FunctionalThresholdingSAE.
+
+    Tied dictionary with a learned soft-threshold gate instead of bias+ReLU
+    (reference sae_ensemble.py:232-289): the encoder GEMM runs with a gate
+    epilogue (k_enc_fwd_t mode 2) that keeps u = (c+gain)/max(a^2,eps) for
+    the backward, and k_gc_thresh_t pushes dL/dcode through g'(u) while
+    accumulating the per-feature gain/scale gradients as column sums.
+    Weight grads + renorm-projected Adam are the tied-SAE kernels.
+    """
+
+    def __init__(self, ensemble, ext):
+        super().__init__(ensemble, ext, tied=True)
+
+    def _alloc(self, B: int):
+        super()._alloc(B)
+        M, n, d = self.n_models, self.n_dict, self.d_act
+        dev = self.ens.params["encoder"].device
+        self.u = torch.empty(M, B, n, device=dev)
+        self.g_gain = torch.zeros(M, n, device=dev)
+        self.g_scale = torch.zeros(M, n, device=dev)
+        self.dummy_bias = torch.zeros(M, n, device=dev)
+        self.zero_decay = torch.zeros_like(self.bias_decay)
+        # the gate epilogue lives only in the transpose-in-staging kernels
+        self.kc["staging"] = "t"
+
+    def grads_phase(self, x: torch.Tensor, on_grads=None):
+        ens, ext = self.ens, self.ext
+        B = x.shape[0]
+        if self._B != B:
+            self._alloc(B)
+        x = x.contiguous()
+        p = ens.params
+        enc = p["encoder"]
+        a = p["activation_scale"]
+        gain = p["activation_gain"]
+        bk, prio = self.kc["bk"], self.kc["prio"]
+        bk_gw = self.kc["bk_grad_w"] or bk
+
+        self.loss_parts.zero_()
+        self.g_gain.zero_()
+        self.g_scale.zero_()
+
+        ext.row_norms(enc, self.norms, self.inv_norms, EPS_NORM)
+        ext.enc_fwd(x, enc, self.dummy_bias, self.inv_norms, self.c,
+                    self.loss_parts, self.fired, 2, bk, prio, a, gain, self.u)
+        ext.dec_fwd(self.c, enc, self.inv_norms, x, self.r, self.loss_parts,
+                    self.kc["bk_dec"] or bk, prio)
+        ext.gc_thresh(self.r, enc, self.inv_norms, self.c, self.u, a,
+                      self.l1_alpha, self.gpre, self.g_gain, self.g_scale, bk, prio)
+
+        gscale = 2.0 / (B * self.d_act)
+        ext.grad_w(self.c, self.r, self.gw, gscale, 0.0, bk_gw, prio)
+        ext.grad_w(self.gpre, x, self.gw, 1.0, 1.0, bk_gw, prio)
+        if on_grads is not None:
+            on_grads([self.gw, self.g_gain, self.g_scale])
+        return B
+
+    def update_phase(self, B: int):
+        ens, ext = self.ens, self.ext
+        st = ens.optim_states
+        st["step"] += 1.0
+        step_no = st["step"]
+        p = ens.params
+        zero_decay = self.zero_decay
+        ext.project_adam(p["encoder"], self.gw, self.norms,
+                         st["mu"]["encoder"], st["nu"]["encoder"], step_no,
+                         self.n_dict, self.lr, self.beta1, self.beta2,
+                         self.eps, EPS_NORM, True)
+        ext.bias_adam(p["activation_scale"], self.g_scale, zero_decay,
+                      st["mu"]["activation_scale"], st["nu"]["activation_scale"],
+                      step_no, self.lr, self.beta1, self.beta2, self.eps)
+        ext.bias_adam(p["activation_gain"], self.g_gain, zero_decay,
+                      st["mu"]["activation_gain"], st["nu"]["activation_gain"],
+                      step_no, self.lr, self.beta1, self.beta2, self.eps)
+
+    def _loss_data(self, B: int):
+        mse = self.loss_parts[:, 0] / (B * self.d_act)
+        l1 = self.l1_alpha * self.loss_parts[:, 1] / B
+        return {"loss": mse + l1, "l_reconstruction": mse, "l_l1": l1}
+
+    def dp_grad_tensors(self):
+        return [self.gw, self.g_gain, self.g_scale]
 
 
 class HipTopKStep:

@@ -299,6 +299,21 @@ extern "C" __global__ void k_row_norms(const float* __restrict__ W,
     mfma_tile<TBK, ASTR, BSTR>(&As[cur][0], &Bs[cur][0], acc);                 \
   }
 
+// The soft-threshold gate of the thresholding SAE (reference
+// sae_ensemble.py:256-259): u = (c + gain) / max(a^2, eps),
+// g(u) = relu6(60(u-0.9))/6 + relu(u-1), code = g(u) * a^2 (RAW a^2 — the
+// reference multiplies by the unclamped square).
+#define GATE_EPS 1e-8f
+__device__ __forceinline__ float gate_g(float u) {
+  return fminf(fmaxf(60.0f * (u - 0.9f), 0.0f), 6.0f) * (1.0f / 6.0f) +
+         fmaxf(u - 1.0f, 0.0f);
+}
+__device__ __forceinline__ float gate_gp(float u) {  // dg/du (0 at the kinks)
+  float gp = (u > 0.9f && u < 1.0f) ? 10.0f : 0.0f;
+  if (u > 1.0f) gp += 1.0f;
+  return gp;
+}
+
 // ---------------------------------------------------------------------------
 // k_enc_fwd_t
 // ---------------------------------------------------------------------------
@@ -312,8 +327,12 @@ void k_enc_fwd_t(const float* __restrict__ x,       // [B, d]
                  float* __restrict__ loss_parts,    // [M, 2]
                  float* __restrict__ fired,         // [M, n]
                  int B, int d, int n,
-                 int mode,  // 0: bias+relu (+L1/fired); 1: raw scores (TopK)
-                 int prio) {
+                 int mode,  // 0: bias+relu (+L1/fired); 1: raw scores (TopK);
+                            // 2: threshold gate via act_scale/act_gain/u_out
+                 int prio,
+                 const float* __restrict__ act_scale,  // [M, n] (mode 2)
+                 const float* __restrict__ act_gain,   // [M, n] (mode 2)
+                 float* __restrict__ u_out) {          // [M, B, n] (mode 2)
   __shared__ float As[2][TBK * BMP];
   __shared__ float Bs[2][TBK * BMP];
 
@@ -360,6 +379,34 @@ void k_enc_fwd_t(const float* __restrict__ x,       // [B, d]
       }
       continue;
     }
+    if (mode == 2) {
+      // thresholding SAE: code = g((c+gain)/max(a^2,eps)) * a^2; u kept for
+      // the backward's gate derivative (k_gc_thresh_t)
+      float a = col_ok ? act_scale[(long)m * n + col] : 1.f;
+      float gn = col_ok ? act_gain[(long)m * n + col] : 0.f;
+      float s_raw = a * a;
+      float s_inv = 1.0f / fmaxf(s_raw, GATE_EPS);
+      float* u_m = u_out + (long)m * B * n;
+      float fired_cnt = 0.f;
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        int row = row0 + g.wr + acc_row(r, g.lane);
+        if (row < B && col_ok) {
+          float u = (acc[tj][r] + gn) * s_inv;
+          float code = gate_g(u) * s_raw;
+          c_m[(long)row * n + col] = code;
+          u_m[(long)row * n + col] = u;
+          l1_sum += code;
+          fired_cnt += (code > 0.f) ? 1.f : 0.f;
+        }
+      }
+      if (col_ok) {
+        float other = __shfl_xor(fired_cnt, 32, WAVE);
+        float tot = fired_cnt + other;
+        if (g.lane < 32 && tot > 0.f) atomicAdd(&fired_m[col], tot);
+      }
+      continue;
+    }
     float bj = col_ok ? bias_m[col] : 0.f;
     float fired_cnt = 0.f;
 #pragma unroll
@@ -378,7 +425,7 @@ void k_enc_fwd_t(const float* __restrict__ x,       // [B, d]
       if (g.lane < 32 && tot > 0.f) atomicAdd(&fired_m[col], tot);
     }
   }
-  if (mode == 0) {
+  if (mode == 0 || mode == 2) {
     l1_sum = wave_reduce_sum(l1_sum);
     if (g.lane == 0) atomicAdd(&loss_parts[m * 2 + 1], l1_sum);
   }
@@ -512,6 +559,99 @@ void k_gc_t(const float* __restrict__ r,        // [M, B, d]
       float other = __shfl_xor(colsum, 32, WAVE);
       float tot = colsum + other;
       if (g.lane < 32 && tot != 0.f) atomicAdd(&gb_m[col], tot);
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
+// k_gc_thresh_t: backward of the thresholding SAE's gate (K15).
+// Same GEMM as k_gc_t (acc = r @ What^T); epilogue turns dL/dcode into
+//   gpre   = dL/dc      = gv * g'(u) * a^2/max(a^2,eps)   (feeds grad_w)
+//   g_gain = sum_b dL/dgain = same weight as gpre (column sum)
+//   g_scale= sum_b dL/da = gv * 2a * (g(u) - [a^2>eps] g'(u) u a^2/max(..))
+// with gv = gscale*acc + l1/B * [code>0].
+// ---------------------------------------------------------------------------
+template <int TBK, int MINW>
+__global__ __launch_bounds__(NTHREADS, MINW)
+void k_gc_thresh_t(const float* __restrict__ r,        // [M, B, d]
+                   const float* __restrict__ Wdec,     // [M, n, d]
+                   const float* __restrict__ inv_norms,// [M, n]
+                   const float* __restrict__ c,        // [M, B, n] (codes)
+                   const float* __restrict__ u,        // [M, B, n]
+                   const float* __restrict__ act_scale,// [M, n]
+                   const float* __restrict__ l1_alpha, // [M]
+                   float* __restrict__ gpre_out,       // [M, B, n]
+                   float* __restrict__ g_gain,         // [M, n]
+                   float* __restrict__ g_scale,        // [M, n]
+                   int B, int d, int n, int prio) {
+  __shared__ float As[2][TBK * BMP];
+  __shared__ float Bs[2][TBK * BMP];
+
+  maybe_prio(prio);
+  const int m = blockIdx.z;
+  int tx, ty;
+  tile_coords(tx, ty);
+  const int row0 = ty * BM;
+  const int col0 = tx * BN;
+  const float* r_m = r + (long)m * B * d;
+  const float* W = Wdec + (long)m * n * d;
+  const float* inv = inv_norms + (long)m * n;
+  const float* c_m = c + (long)m * B * n;
+  const float* u_m = u + (long)m * B * n;
+  const float gscale = 2.0f / ((float)B * (float)d);
+  const float l1_term = l1_alpha[m] / (float)B;
+
+  f32x16 acc[2];
+  zero_acc(acc);
+  TStage<TBK> sa, sb;
+
+#define GCT_LA(K) stage_T_load<TBK>(r_m, d, row0, (K), B, d, nullptr, sa)
+#define GCT_LB(K) stage_T_load<TBK>(W, d, col0, (K), n, d, inv, sb)
+#define GCT_WA(BUF) stage_T_write<TBK>(sa, &As[BUF][0], false)
+#define GCT_WB(BUF) stage_T_write<TBK>(sb, &Bs[BUF][0], true)
+  PREFETCH_LOOP(TBK, d, GCT_LA, GCT_LB, GCT_WA, GCT_WB, BMP, BMP)
+#undef GCT_LA
+#undef GCT_LB
+#undef GCT_WA
+#undef GCT_WB
+
+  const EpiGeom g = epi_geom();
+  float* g_m = gpre_out + (long)m * B * n;
+  float* gg_m = g_gain + (long)m * n;
+  float* gs_m = g_scale + (long)m * n;
+
+#pragma unroll
+  for (int tj = 0; tj < 2; ++tj) {
+    int col = col0 + g.wc + tj * 32 + g.l31;
+    bool col_ok = col < n;
+    float a = col_ok ? act_scale[(long)m * n + col] : 1.f;
+    float s_raw = a * a;
+    float s = fmaxf(s_raw, GATE_EPS);
+    float clamp_act = (s_raw > GATE_EPS) ? 1.f : 0.f;  // clamp passes grad?
+    float f = s_raw / s;  // a^2 / clamp(a^2): ==1 away from the clamp
+    float gain_colsum = 0.f, scale_colsum = 0.f;
+#pragma unroll
+    for (int r_ = 0; r_ < 16; ++r_) {
+      int row = row0 + g.wr + acc_row(r_, g.lane);
+      if (row < B && col_ok) {
+        long idx = (long)row * n + col;
+        float code = c_m[idx];
+        float uv = u_m[idx];
+        float gp = gate_gp(uv);
+        float gv = gscale * acc[tj][r_] + ((code > 0.f) ? l1_term : 0.f);
+        float gc_ = gv * gp * f;
+        g_m[idx] = gc_;
+        gain_colsum += gc_;
+        scale_colsum += gv * 2.0f * a * (gate_g(uv) - clamp_act * gp * uv * f);
+      }
+    }
+    if (col_ok) {
+      float og = __shfl_xor(gain_colsum, 32, WAVE);
+      float os = __shfl_xor(scale_colsum, 32, WAVE);
+      if (g.lane < 32) {
+        if (gain_colsum + og != 0.f) atomicAdd(&gg_m[col], gain_colsum + og);
+        if (scale_colsum + os != 0.f) atomicAdd(&gs_m[col], scale_colsum + os);
+      }
     }
   }
 }

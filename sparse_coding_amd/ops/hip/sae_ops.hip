@@ -31,7 +31,10 @@ void row_norms(torch::Tensor W, torch::Tensor norms, torch::Tensor inv_norms, do
 void enc_fwd(torch::Tensor x, torch::Tensor Wenc, torch::Tensor bias,
              c10::optional<torch::Tensor> inv_norms, torch::Tensor c_out,
              torch::Tensor loss_parts, torch::Tensor fired, int64_t mode,
-             int64_t bk, bool prio) {
+             int64_t bk, bool prio,
+             c10::optional<torch::Tensor> act_scale,
+             c10::optional<torch::Tensor> act_gain,
+             c10::optional<torch::Tensor> u_out) {
   CHECK_IN(x); CHECK_IN(Wenc); CHECK_IN(bias); CHECK_IN(c_out);
   CHECK_IN(loss_parts); CHECK_IN(fired);
   int M = Wenc.size(0), n = Wenc.size(1), d = Wenc.size(2);
@@ -42,19 +45,29 @@ void enc_fwd(torch::Tensor x, torch::Tensor Wenc, torch::Tensor bias,
     CHECK_IN(inv_norms.value());
     inv = inv_norms->data_ptr<float>();
   }
+  const float* a_p = nullptr;
+  const float* gn_p = nullptr;
+  float* u_p = nullptr;
+  if (mode == 2) {
+    TORCH_CHECK(act_scale && act_gain && u_out, "mode 2 needs act_scale/act_gain/u_out");
+    CHECK_IN(act_scale.value()); CHECK_IN(act_gain.value()); CHECK_IN(u_out.value());
+    a_p = act_scale->data_ptr<float>();
+    gn_p = act_gain->data_ptr<float>();
+    u_p = u_out->data_ptr<float>();
+  }
   dim3 grid(cdiv(n, BN), cdiv(B, BM), M);
   if (bk == 16)
     hipLaunchKernelGGL((k_enc_fwd_t<16, 6>), grid, dim3(NTHREADS), 0, cur_stream(),
                        x.data_ptr<float>(), Wenc.data_ptr<float>(),
                        bias.data_ptr<float>(), inv, c_out.data_ptr<float>(),
                        loss_parts.data_ptr<float>(), fired.data_ptr<float>(),
-                       B, d, n, (int)mode, prio ? 1 : 0);
+                       B, d, n, (int)mode, prio ? 1 : 0, a_p, gn_p, u_p);
   else
     hipLaunchKernelGGL((k_enc_fwd_t<32, 4>), grid, dim3(NTHREADS), 0, cur_stream(),
                        x.data_ptr<float>(), Wenc.data_ptr<float>(),
                        bias.data_ptr<float>(), inv, c_out.data_ptr<float>(),
                        loss_parts.data_ptr<float>(), fired.data_ptr<float>(),
-                       B, d, n, (int)mode, prio ? 1 : 0);
+                       B, d, n, (int)mode, prio ? 1 : 0, a_p, gn_p, u_p);
 }
 
 void dec_fwd(torch::Tensor c, torch::Tensor Wdec, torch::Tensor inv_norms,
@@ -99,6 +112,35 @@ void gc(torch::Tensor r, torch::Tensor Wdec, torch::Tensor inv_norms,
                        inv_norms.data_ptr<float>(), c.data_ptr<float>(),
                        l1_alpha.data_ptr<float>(), gpre.data_ptr<float>(),
                        g_bias.data_ptr<float>(), B, d, n, prio ? 1 : 0);
+}
+
+void gc_thresh(torch::Tensor r, torch::Tensor Wdec, torch::Tensor inv_norms,
+               torch::Tensor c, torch::Tensor u, torch::Tensor act_scale,
+               torch::Tensor l1_alpha, torch::Tensor gpre,
+               torch::Tensor g_gain, torch::Tensor g_scale,
+               int64_t bk, bool prio) {
+  CHECK_IN(r); CHECK_IN(Wdec); CHECK_IN(inv_norms); CHECK_IN(c); CHECK_IN(u);
+  CHECK_IN(act_scale); CHECK_IN(l1_alpha); CHECK_IN(gpre);
+  CHECK_IN(g_gain); CHECK_IN(g_scale);
+  int M = Wdec.size(0), n = Wdec.size(1), d = Wdec.size(2);
+  int B = r.size(1);
+  dim3 grid(cdiv(n, BN), cdiv(B, BM), M);
+  if (bk == 16)
+    hipLaunchKernelGGL((k_gc_thresh_t<16, 6>), grid, dim3(NTHREADS), 0, cur_stream(),
+                       r.data_ptr<float>(), Wdec.data_ptr<float>(),
+                       inv_norms.data_ptr<float>(), c.data_ptr<float>(),
+                       u.data_ptr<float>(), act_scale.data_ptr<float>(),
+                       l1_alpha.data_ptr<float>(), gpre.data_ptr<float>(),
+                       g_gain.data_ptr<float>(), g_scale.data_ptr<float>(),
+                       B, d, n, prio ? 1 : 0);
+  else
+    hipLaunchKernelGGL((k_gc_thresh_t<32, 4>), grid, dim3(NTHREADS), 0, cur_stream(),
+                       r.data_ptr<float>(), Wdec.data_ptr<float>(),
+                       inv_norms.data_ptr<float>(), c.data_ptr<float>(),
+                       u.data_ptr<float>(), act_scale.data_ptr<float>(),
+                       l1_alpha.data_ptr<float>(), gpre.data_ptr<float>(),
+                       g_gain.data_ptr<float>(), g_scale.data_ptr<float>(),
+                       B, d, n, prio ? 1 : 0);
 }
 
 // gw[m] = beta * gw[m] + alpha * P[m]^T @ Q[m]; Q may be rank-shared [B, d]
@@ -239,9 +281,16 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("rT"), py::arg("WT"), py::arg("c"), py::arg("l1_alpha"),
         py::arg("gpre"), py::arg("g_bias"),
         py::arg("bk") = 32, py::arg("prio") = false);
-  m.def("enc_fwd", &enc_fwd, "fused encoder GEMM + bias + ReLU (+L1, fired)",
+  m.def("enc_fwd", &enc_fwd, "fused encoder GEMM + bias + ReLU/TopK/gate (+L1, fired)",
         py::arg("x"), py::arg("Wenc"), py::arg("bias"), py::arg("inv_norms"),
         py::arg("c_out"), py::arg("loss_parts"), py::arg("fired"), py::arg("mode"),
+        py::arg("bk") = 32, py::arg("prio") = false,
+        py::arg("act_scale") = py::none(), py::arg("act_gain") = py::none(),
+        py::arg("u_out") = py::none());
+  m.def("gc_thresh", &gc_thresh, "code-grad through the threshold gate (+gain/scale grads)",
+        py::arg("r"), py::arg("Wdec"), py::arg("inv_norms"), py::arg("c"),
+        py::arg("u"), py::arg("act_scale"), py::arg("l1_alpha"), py::arg("gpre"),
+        py::arg("g_gain"), py::arg("g_scale"),
         py::arg("bk") = 32, py::arg("prio") = false);
   m.def("dec_fwd", &dec_fwd, "fused decoder GEMM - x (+MSE partial)",
         py::arg("c"), py::arg("Wdec"), py::arg("inv_norms"), py::arg("x"),

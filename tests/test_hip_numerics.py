@@ -255,6 +255,35 @@ def test_bias_decay_gradient():
     assert _rel_err(l_hip["loss"], l_ref["loss"]) < 1e-4
 
 
+def test_thresholding_step_matches_torch():
+    """Fused threshold-gate step (k_enc_fwd mode 2 + k_gc_thresh) vs the vmap
+    oracle: gate forward, g'(u) backward, and the gain/scale column-sum
+    grads all have to line up."""
+    from sparse_coding_amd.engine.ensemble import FunctionalEnsemble
+    from sparse_coding_amd.functional.optim import adam
+    from sparse_coding_amd.models.sae_signatures import FunctionalThresholdingSAE
+
+    torch.manual_seed(10)
+    M, B, d, n = 2, 256, 64, 128
+    models = [FunctionalThresholdingSAE.init(d, n, l1, device=DEV) for l1 in (1e-3, 3e-3)]
+    ens_hip = FunctionalEnsemble(models, FunctionalThresholdingSAE, adam, {"lr": 1e-3},
+                                 device=DEV, backend="hip")
+    assert type(ens_hip._hip_step).__name__ == "HipThresholdStep"
+    models2 = [({k: v.clone() for k, v in p.items()}, {k: v.clone() for k, v in b.items()})
+               for p, b in ens_hip.unstack()]
+    ens_ref = FunctionalEnsemble(models2, FunctionalThresholdingSAE, adam, {"lr": 1e-3},
+                                 device=DEV, backend="torch")
+    # scale up inputs so the gate's u regularly crosses the 0.9/1.0 knees
+    x = 2.0 * torch.randn(B, d, device=DEV)
+    for i in range(5):
+        l_hip, aux_hip = ens_hip.step_batch(x)
+        l_ref, aux_ref = ens_ref.step_batch(x)
+        assert _rel_err(l_hip["loss"], l_ref["loss"]) < 1e-4, i
+        assert _rel_err(aux_hip["c"], aux_ref["c"]) < 1e-3, i
+    for k in ens_ref.params:
+        assert _rel_err(ens_hip.params[k], ens_ref.params[k]) < 2e-3, k
+
+
 def test_topk_step_matches_torch():
     """Fused TopK step vs the reference-semantics no_stacking vmap oracle."""
     from sparse_coding_amd.engine.ensemble import FunctionalEnsemble
