@@ -75,6 +75,13 @@ def maybe_make_step(ensemble, required: bool = False) -> Optional["HipSAEStep"]:
         ext = _ops.get_extension(required=True)
         return HipThresholdStep(ensemble, ext)
 
+    if sig is sigs.FunctionalMaskedTiedSAE:
+        ext = _ops.get_extension(required=True)
+        return HipSAEStep(ensemble, ext, tied=True, masked=True)
+    if sig is sigs.FunctionalMaskedSAE:
+        ext = _ops.get_extension(required=True)
+        return HipSAEStep(ensemble, ext, tied=False, masked=True)
+
     tied = sig is sigs.FunctionalTiedSAE
     untied = sig is sigs.FunctionalSAE
     if not (tied or untied):
@@ -93,10 +100,11 @@ def maybe_make_step(ensemble, required: bool = False) -> Optional["HipSAEStep"]:
 class HipSAEStep:
     """Workspaces + kernel-sequence launcher for one ensemble."""
 
-    def __init__(self, ensemble, ext, tied: bool):
+    def __init__(self, ensemble, ext, tied: bool, masked: bool = False):
         self.ens = ensemble
         self.ext = ext
         self.tied = tied
+        self.masked = masked
         self._ws = {}
 
         p = ensemble.params
@@ -108,9 +116,17 @@ class HipSAEStep:
         b = ensemble.buffers
         self.l1_alpha = b["l1_alpha"].detach().to(dev, torch.float32).reshape(self.n_models).contiguous()
         bd = b.get("bias_decay")
-        if bd is None:
+        # the masked signatures carry a bias_decay buffer but never apply it
+        # (reference sae_ensemble.py:347-373,425-444) — force zero
+        if bd is None or masked:
             bd = torch.zeros(self.n_models, device=dev)
         self.bias_decay = bd.detach().to(dev, torch.float32).reshape(self.n_models).contiguous()
+        self.dict_sizes = None
+        if masked:
+            # K9: coefficient columns >= dict_size[m] are masked to zero in
+            # the encoder epilogue; every downstream grad is then zero via
+            # the c>0 gating, so the rest of the pipeline is unchanged
+            self.dict_sizes = b["dict_size"].detach().reshape(self.n_models).to(dev, torch.int32).contiguous()
 
         opt = ensemble.optimizer_kwargs
         self.lr = float(opt.get("lr", 1e-3))
@@ -198,17 +214,20 @@ class HipSAEStep:
             ext.transpose_scale(x, self.xT, None)
             ext.transpose_scale(dict_w, self.WT, self.inv_norms)
             if self.tied:
-                ext.enc_fwd2(self.xT, self.WT, bias, self.c, self.loss_parts, self.fired, 0, bk, prio)
+                ext.enc_fwd2(self.xT, self.WT, bias, self.c, self.loss_parts, self.fired, 0, bk, prio,
+                             dict_sizes=self.dict_sizes)
             else:
                 ext.transpose_scale(enc, self.WencT, None)
-                ext.enc_fwd2(self.xT, self.WencT, bias, self.c, self.loss_parts, self.fired, 0, bk, prio)
+                ext.enc_fwd2(self.xT, self.WencT, bias, self.c, self.loss_parts, self.fired, 0, bk, prio,
+                             dict_sizes=self.dict_sizes)
             ext.dec_fwd(self.c, dict_w, self.inv_norms, x, self.r, self.loss_parts, bk_dec, prio)
             ext.transpose_scale(self.r, self.rT, None)
             ext.gc2(self.rT, self.WT, self.c, self.l1_alpha, self.gpre, self.g_bias, bk, prio)
         else:
             # transpose-in-staging GEMMs (no separate transpose kernels)
             enc_inv = self.inv_norms if self.tied else None
-            ext.enc_fwd(x, enc, bias, enc_inv, self.c, self.loss_parts, self.fired, 0, bk, prio)
+            ext.enc_fwd(x, enc, bias, enc_inv, self.c, self.loss_parts, self.fired, 0, bk, prio,
+                        dict_sizes=self.dict_sizes)
             ext.dec_fwd(self.c, dict_w, self.inv_norms, x, self.r, self.loss_parts, bk_dec, prio)
             ext.gc(self.r, dict_w, self.inv_norms, self.c, self.l1_alpha, self.gpre, self.g_bias, bk, prio)
 
@@ -266,6 +285,8 @@ class HipSAEStep:
         d = self.d_act
         mse = self.loss_parts[:, 0] / (B * d)
         l1 = self.l1_alpha * self.loss_parts[:, 1] / B
+        if self.masked:  # masked losses carry no bias-decay term
+            return {"loss": mse + l1, "l_reconstruction": mse, "l_l1": l1}
         bias = self.ens.params["encoder_bias"]
         l_bd = self.bias_decay * torch.norm(bias, 2, dim=-1)
         total = mse + l1 + l_bd

@@ -332,7 +332,9 @@ void k_enc_fwd_t(const float* __restrict__ x,       // [B, d]
                  int prio,
                  const float* __restrict__ act_scale,  // [M, n] (mode 2)
                  const float* __restrict__ act_gain,   // [M, n] (mode 2)
-                 float* __restrict__ u_out) {          // [M, B, n] (mode 2)
+                 float* __restrict__ u_out,            // [M, B, n] (mode 2)
+                 const int* __restrict__ dict_sizes) { // [M] or nullptr:
+                                       // masked sigs zero cols >= dict_sizes[m]
   __shared__ float As[2][TBK * BMP];
   __shared__ float Bs[2][TBK * BMP];
 
@@ -408,12 +410,14 @@ void k_enc_fwd_t(const float* __restrict__ x,       // [B, d]
       continue;
     }
     float bj = col_ok ? bias_m[col] : 0.f;
+    // coefficient mask (reference K9): columns >= dict_sizes[m] are dead
+    bool live = col_ok && (!dict_sizes || col < dict_sizes[m]);
     float fired_cnt = 0.f;
 #pragma unroll
     for (int r = 0; r < 16; ++r) {
       int row = row0 + g.wr + acc_row(r, g.lane);
       if (row < B && col_ok) {
-        float v = fmaxf(acc[tj][r] + bj, 0.f);
+        float v = live ? fmaxf(acc[tj][r] + bj, 0.f) : 0.f;
         c_m[(long)row * n + col] = v;
         l1_sum += v;
         fired_cnt += (v > 0.f) ? 1.f : 0.f;
@@ -872,7 +876,8 @@ void k_enc_fwd2_t(const float* __restrict__ xT,      // [d, B]
                   float* __restrict__ c_out,         // [M, B, n]
                   float* __restrict__ loss_parts,    // [M, 2]
                   float* __restrict__ fired,         // [M, n]
-                  int B, int d, int n, int mode, int prio) {
+                  int B, int d, int n, int mode, int prio,
+                  const int* __restrict__ dict_sizes) {
   __shared__ float As[2][TBK * BM];
   __shared__ float Bs[2][TBK * BM];
 
@@ -917,12 +922,13 @@ void k_enc_fwd2_t(const float* __restrict__ xT,      // [d, B]
       continue;
     }
     float bj = col_ok ? bias_m[col] : 0.f;
+    bool live = col_ok && (!dict_sizes || col < dict_sizes[m]);
     float fired_cnt = 0.f;
 #pragma unroll
     for (int r = 0; r < 16; ++r) {
       int row = row0 + g.wr + acc_row(r, g.lane);
       if (row < B && col_ok) {
-        float v = fmaxf(acc[tj][r] + bj, 0.f);
+        float v = live ? fmaxf(acc[tj][r] + bj, 0.f) : 0.f;
         c_m[(long)row * n + col] = v;
         l1_sum += v;
         fired_cnt += (v > 0.f) ? 1.f : 0.f;

@@ -34,7 +34,8 @@ void enc_fwd(torch::Tensor x, torch::Tensor Wenc, torch::Tensor bias,
              int64_t bk, bool prio,
              c10::optional<torch::Tensor> act_scale,
              c10::optional<torch::Tensor> act_gain,
-             c10::optional<torch::Tensor> u_out) {
+             c10::optional<torch::Tensor> u_out,
+             c10::optional<torch::Tensor> dict_sizes) {
   CHECK_IN(x); CHECK_IN(Wenc); CHECK_IN(bias); CHECK_IN(c_out);
   CHECK_IN(loss_parts); CHECK_IN(fired);
   int M = Wenc.size(0), n = Wenc.size(1), d = Wenc.size(2);
@@ -55,19 +56,25 @@ void enc_fwd(torch::Tensor x, torch::Tensor Wenc, torch::Tensor bias,
     gn_p = act_gain->data_ptr<float>();
     u_p = u_out->data_ptr<float>();
   }
+  const int* ds_p = nullptr;
+  if (dict_sizes.has_value()) {
+    TORCH_CHECK(dict_sizes->is_cuda() && dict_sizes->is_contiguous() &&
+                dict_sizes->scalar_type() == torch::kInt32, "dict_sizes must be int32 GPU");
+    ds_p = dict_sizes->data_ptr<int>();
+  }
   dim3 grid(cdiv(n, BN), cdiv(B, BM), M);
   if (bk == 16)
     hipLaunchKernelGGL((k_enc_fwd_t<16, 6>), grid, dim3(NTHREADS), 0, cur_stream(),
                        x.data_ptr<float>(), Wenc.data_ptr<float>(),
                        bias.data_ptr<float>(), inv, c_out.data_ptr<float>(),
                        loss_parts.data_ptr<float>(), fired.data_ptr<float>(),
-                       B, d, n, (int)mode, prio ? 1 : 0, a_p, gn_p, u_p);
+                       B, d, n, (int)mode, prio ? 1 : 0, a_p, gn_p, u_p, ds_p);
   else
     hipLaunchKernelGGL((k_enc_fwd_t<32, 4>), grid, dim3(NTHREADS), 0, cur_stream(),
                        x.data_ptr<float>(), Wenc.data_ptr<float>(),
                        bias.data_ptr<float>(), inv, c_out.data_ptr<float>(),
                        loss_parts.data_ptr<float>(), fired.data_ptr<float>(),
-                       B, d, n, (int)mode, prio ? 1 : 0, a_p, gn_p, u_p);
+                       B, d, n, (int)mode, prio ? 1 : 0, a_p, gn_p, u_p, ds_p);
 }
 
 void dec_fwd(torch::Tensor c, torch::Tensor Wdec, torch::Tensor inv_norms,
@@ -227,25 +234,32 @@ void transpose_scale(torch::Tensor src, torch::Tensor dst,
 
 void enc_fwd2(torch::Tensor xT, torch::Tensor WT, torch::Tensor bias,
               torch::Tensor c_out, torch::Tensor loss_parts,
-              torch::Tensor fired, int64_t mode, int64_t bk, bool prio) {
+              torch::Tensor fired, int64_t mode, int64_t bk, bool prio,
+              c10::optional<torch::Tensor> dict_sizes) {
   CHECK_IN(xT); CHECK_IN(WT); CHECK_IN(bias); CHECK_IN(c_out);
   CHECK_IN(loss_parts); CHECK_IN(fired);
   int M = WT.size(0), d = WT.size(1), n = WT.size(2);
   int B = xT.size(1);
   TORCH_CHECK(B % 4 == 0 && n % 4 == 0, "B and n must be multiples of 4");
+  const int* ds_p = nullptr;
+  if (dict_sizes.has_value()) {
+    TORCH_CHECK(dict_sizes->is_cuda() && dict_sizes->is_contiguous() &&
+                dict_sizes->scalar_type() == torch::kInt32, "dict_sizes must be int32 GPU");
+    ds_p = dict_sizes->data_ptr<int>();
+  }
   dim3 grid(cdiv(n, BN), cdiv(B, BM), M);
   if (bk == 16)
     hipLaunchKernelGGL((k_enc_fwd2_t<16, 6>), grid, dim3(NTHREADS), 0, cur_stream(),
                        xT.data_ptr<float>(), WT.data_ptr<float>(),
                        bias.data_ptr<float>(), c_out.data_ptr<float>(),
                        loss_parts.data_ptr<float>(), fired.data_ptr<float>(),
-                       B, d, n, (int)mode, prio ? 1 : 0);
+                       B, d, n, (int)mode, prio ? 1 : 0, ds_p);
   else
     hipLaunchKernelGGL((k_enc_fwd2_t<32, 4>), grid, dim3(NTHREADS), 0, cur_stream(),
                        xT.data_ptr<float>(), WT.data_ptr<float>(),
                        bias.data_ptr<float>(), c_out.data_ptr<float>(),
                        loss_parts.data_ptr<float>(), fired.data_ptr<float>(),
-                       B, d, n, (int)mode, prio ? 1 : 0);
+                       B, d, n, (int)mode, prio ? 1 : 0, ds_p);
 }
 
 void gc2(torch::Tensor rT, torch::Tensor WT, torch::Tensor c,
@@ -276,7 +290,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("enc_fwd2", &enc_fwd2, "enc forward, pre-transposed operands (all-direct staging)",
         py::arg("xT"), py::arg("WT"), py::arg("bias"), py::arg("c_out"),
         py::arg("loss_parts"), py::arg("fired"), py::arg("mode"),
-        py::arg("bk") = 32, py::arg("prio") = false);
+        py::arg("bk") = 32, py::arg("prio") = false,
+        py::arg("dict_sizes") = py::none());
   m.def("gc2", &gc2, "code-grad, pre-transposed operands",
         py::arg("rT"), py::arg("WT"), py::arg("c"), py::arg("l1_alpha"),
         py::arg("gpre"), py::arg("g_bias"),
@@ -286,7 +301,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("c_out"), py::arg("loss_parts"), py::arg("fired"), py::arg("mode"),
         py::arg("bk") = 32, py::arg("prio") = false,
         py::arg("act_scale") = py::none(), py::arg("act_gain") = py::none(),
-        py::arg("u_out") = py::none());
+        py::arg("u_out") = py::none(), py::arg("dict_sizes") = py::none());
   m.def("gc_thresh", &gc_thresh, "code-grad through the threshold gate (+gain/scale grads)",
         py::arg("r"), py::arg("Wdec"), py::arg("inv_norms"), py::arg("c"),
         py::arg("u"), py::arg("act_scale"), py::arg("l1_alpha"), py::arg("gpre"),

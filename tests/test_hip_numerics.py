@@ -284,6 +284,37 @@ def test_thresholding_step_matches_torch():
         assert _rel_err(ens_hip.params[k], ens_ref.params[k]) < 2e-3, k
 
 
+def test_masked_step_matches_torch():
+    """Fused masked steps (K9): different dict sizes stacked to one width;
+    coefficient columns >= dict_size[m] stay zero and get no grads."""
+    from sparse_coding_amd.engine.ensemble import FunctionalEnsemble
+    from sparse_coding_amd.functional.optim import adam
+    from sparse_coding_amd.models.sae_signatures import (
+        FunctionalMaskedSAE,
+        FunctionalMaskedTiedSAE,
+    )
+
+    for sig in (FunctionalMaskedTiedSAE, FunctionalMaskedSAE):
+        torch.manual_seed(13)
+        B, d, n_stack = 256, 64, 192
+        dict_sizes = (64, 192)
+        models = [sig.init(d, nd, n_stack, 1e-3, device=DEV) for nd in dict_sizes]
+        ens_hip = FunctionalEnsemble(models, sig, adam, {"lr": 1e-3}, device=DEV, backend="hip")
+        assert type(ens_hip._hip_step).__name__ == "HipSAEStep" and ens_hip._hip_step.masked
+        models2 = [({k: v.clone() for k, v in p.items()}, {k: v.clone() for k, v in b.items()})
+                   for p, b in ens_hip.unstack()]
+        ens_ref = FunctionalEnsemble(models2, sig, adam, {"lr": 1e-3}, device=DEV, backend="torch")
+        x = torch.randn(B, d, device=DEV)
+        for i in range(4):
+            l_hip, aux_hip = ens_hip.step_batch(x)
+            l_ref, aux_ref = ens_ref.step_batch(x)
+            assert _rel_err(l_hip["loss"], l_ref["loss"]) < 1e-4, (sig.__name__, i)
+            # masked coefficients exactly zero
+            assert (aux_hip["c"][0, :, dict_sizes[0]:] == 0).all()
+        for k in ens_ref.params:
+            assert _rel_err(ens_hip.params[k], ens_ref.params[k]) < 2e-3, (sig.__name__, k)
+
+
 def test_topk_step_matches_torch():
     """Fused TopK step vs the reference-semantics no_stacking vmap oracle."""
     from sparse_coding_amd.engine.ensemble import FunctionalEnsemble
