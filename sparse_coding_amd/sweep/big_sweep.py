@@ -8,7 +8,10 @@ init_synthetic_dataset :269-295, checkpoint layout ``_{i}/learned_dicts.pt``
 Differences (MI355X build): wandb is optional (RunLogger JSONL always on);
 the host LM is an HF transformers model (activation_dataset); dispatch
 across GPUs goes through sweep/cluster_runs (process-per-ensemble) or the
-RCCL DP trainer for a single big ensemble.
+RCCL DP trainer for a single big ensemble; and training is RESUMABLE — the
+reference never saves optimizer state, here every checkpoint
+also writes ``resume_state.pt`` (params + Adam moments + chunk cursor,
+atomic rename) and ``cfg.resume=True`` continues from it.
 """
 
 from __future__ import annotations
@@ -71,16 +74,26 @@ def make_hyperparam_name(setting):
 
 
 def ensemble_train_loop(ensemble, cfg, args, ensemble_name, sampler, dataset, progress_counter):
-    """Inner loop run by each dispatched worker (reference :159-199)."""
+    """Inner loop run by each dispatched worker (reference :159-199).
+
+    Adds a throughput counter the reference lacks: activations/sec per
+    ensemble, logged at chunk end (SURVEY.md §5 "rebuild adds its own
+    throughput counters").
+    """
+    import time as _time
+
     torch.set_grad_enabled(False)
     torch.manual_seed(0)
     np.random.seed(0)
 
     logger = getattr(cfg, "logger", None)
+    t0 = _time.perf_counter()
+    n_acts = 0
 
     for i, batch_idxs in enumerate(sampler):
         batch = dataset[batch_idxs].to(args["device"])
         losses, aux = ensemble.step_batch(batch)
+        n_acts += batch.shape[0]
 
         if logger is not None and i % getattr(cfg, "log_every", 10) == 0:
             num_nonzero = aux["c"].count_nonzero(dim=-1).float().mean(dim=-1)
@@ -98,6 +111,13 @@ def ensemble_train_loop(ensemble, cfg, args, ensemble_name, sampler, dataset, pr
             logger.log(log)
 
         progress_counter.value = i
+
+    if torch.cuda.is_available():
+        torch.cuda.synchronize(args["device"])
+    dt = _time.perf_counter() - t0
+    if logger is not None and dt > 0:
+        logger.log({f"{ensemble_name}_acts_per_sec": n_acts / dt,
+                    f"{ensemble_name}_chunk_wall_s": dt})
 
 
 def unstacked_to_learned_dicts(ensemble, args, ensemble_hyperparams, buffer_hyperparams):
@@ -260,8 +280,40 @@ def sweep(ensemble_init_func, cfg):
     if getattr(cfg, "n_repetitions", None) is not None:
         chunk_order = np.tile(chunk_order, cfg.n_repetitions)
 
+    # -- resume (this framework; the reference has no resume path) ----------
+    resume_path = os.path.join(cfg.output_folder, "resume_state.pt")
+    start_chunk = 0
+    if getattr(cfg, "resume", False) and os.path.exists(resume_path):
+        saved = torch.load(resume_path, map_location="cpu", weights_only=False)
+        if len(saved["chunk_order"]) == len(chunk_order):
+            chunk_order = np.asarray(saved["chunk_order"])
+        else:
+            # run extended/shortened (e.g. n_repetitions raised to continue
+            # training): keep the fresh order, whose prefix matches because
+            # sweep() reseeds np before drawing it
+            print("[resume] chunk_order length changed; keeping the new schedule")
+        start_chunk = min(saved["chunk_pos"], len(chunk_order))
+        for (ensemble, _, name), st in zip(ensembles, saved["ensemble_states"]):
+            dev = ensemble.device
+            ensemble.params = _to_device_tree(st["params"], dev)
+            ensemble.optim_states = _to_device_tree(st["optim_states"], dev)
+            ensemble.buffers = _to_device_tree(st["buffers"], dev)
+            ensemble._hip_step = None
+            ensemble.init_functions()
+        print(f"Resumed from {resume_path} at chunk {start_chunk}/{len(chunk_order)}")
+
     means = None
+    if getattr(cfg, "center_activations", False) and start_chunk > 0:
+        means = torch.load(os.path.join(cfg.output_folder, "means.pt"), map_location="cpu")
+
+    learned_dicts = []
+    if start_chunk > 0:
+        # a fully-resumed run (nothing left to train) still returns dicts
+        for ensemble, arg, _ in ensembles:
+            learned_dicts.extend(unstacked_to_learned_dicts(ensemble, arg, ensemble_hyperparams, buffer_hyperparams))
     for i, chunk_idx in enumerate(chunk_order):
+        if i < start_chunk:
+            continue
         print(f"Chunk {i + 1}/{len(chunk_order)}")
         chunk = torch.load(os.path.join(cfg.dataset_folder, f"{chunk_idx}.pt")).to(device="cpu", dtype=torch.float32)
         if getattr(cfg, "center_activations", False):
@@ -288,6 +340,33 @@ def sweep(ensemble_init_func, cfg):
                 cfg_dict = cfg.as_dict() if hasattr(cfg, "as_dict") else dict(cfg)
                 cfg_dict.pop("logger", None)
                 yaml.dump({k: v for k, v in cfg_dict.items() if isinstance(v, (int, float, str, bool, list, type(None)))}, f)
+            _save_resume_state(resume_path, ensembles, chunk_order, i + 1)
 
     cfg.logger.close()
     return learned_dicts
+
+
+def _to_device_tree(tree, device):
+    from sparse_coding_amd.utils.tree import tree_map
+
+    return tree_map(lambda t: t.to(device) if hasattr(t, "to") else t, tree)
+
+
+def _save_resume_state(path, ensembles, chunk_order, chunk_pos):
+    """Full training state (params + optimizer moments + chunk cursor),
+    written atomically so a crash mid-save never corrupts the last good
+    state.  The reference never persists optimizer state."""
+    states = []
+    for ensemble, _, name in ensembles:
+        st = ensemble.state_dict()
+        states.append({
+            "name": name,
+            "params": _to_device_tree(st["params"], "cpu"),
+            "buffers": _to_device_tree(st["buffers"], "cpu"),
+            "optim_states": _to_device_tree(st["optim_states"], "cpu"),
+        })
+    tmp = path + ".tmp"
+    torch.save({"chunk_order": np.asarray(chunk_order).tolist(),
+                "chunk_pos": int(chunk_pos),
+                "ensemble_states": states}, tmp)
+    os.replace(tmp, path)

@@ -23,6 +23,13 @@ from torch.utils.data import BatchSampler, RandomSampler
 
 from sparse_coding_amd.engine.ensemble import FunctionalEnsemble
 
+# Always fork-free: a fork while the parent's OpenMP/BLAS pools hold locks
+# deadlocks the child inside its first torch op (observed on CPU runs where
+# a vmap-heavy step preceded dispatch).  The reference forces spawn globally
+# (big_sweep.py:302); here the context is explicit so library users are safe
+# no matter what the global start method is.
+_ctx = mp.get_context("spawn")
+
 
 def job_wrapper(job, ensemble_state_dict, cfg, args, tag, dataset, done_flag, progress_counter):
     torch.manual_seed(0)
@@ -50,11 +57,11 @@ def dispatch_job_on_chunk(ensembles: List[Tuple[Any, dict, str]], cfg, chunk: to
 
     for ensemble, args, name in ensembles:
         ensemble.to_shared_memory()
-        done = mp.Value("i", 0)
-        counter = mp.Value("i", 0)
+        done = _ctx.Value("i", 0)
+        counter = _ctx.Value("i", 0)
         batch_size = args.get("batch_size", getattr(cfg, "batch_size", 256))
         n_batches.append((chunk.shape[0] + batch_size - 1) // batch_size)
-        proc = mp.Process(
+        proc = _ctx.Process(
             target=job_wrapper,
             args=(job, ensemble.state_dict(), cfg, args, name, chunk, done, counter),
         )
@@ -91,10 +98,10 @@ def dispatch_lite(cfg, chunk: torch.Tensor, ensemble, name: str, job: Callable):
         chunk.pin_memory()
     chunk.share_memory_()
     ensemble.to_shared_memory()
-    done = mp.Value("i", 0)
-    counter = mp.Value("i", 0)
+    done = _ctx.Value("i", 0)
+    counter = _ctx.Value("i", 0)
     args = {"batch_size": getattr(cfg, "batch_size", 256), "device": ensemble.device}
-    proc = mp.Process(
+    proc = _ctx.Process(
         target=job_wrapper,
         args=(job, ensemble.state_dict(), cfg, args, name, chunk, done, counter),
     )
