@@ -174,3 +174,70 @@ def test_chunked_grads_match_unchunked():
     assert torch.allclose(hs.gw, gw_mono, atol=1e-6)
     assert torch.allclose(hs.g_bias, gb_mono, atol=1e-6)
     assert len(seen) == 3  # g_bias + two model-half gw slices
+
+
+def test_dp_phase_split_matches_step():
+    """The DP trainer's grads_phase(on_grads)/update_phase split — including
+    the model-half chunked grad_w path taken only when a callback is
+    installed — must produce the same update as the fused step()."""
+    from sparse_coding_amd.engine.ensemble import FunctionalEnsemble
+    from sparse_coding_amd.functional.optim import adam
+    from sparse_coding_amd.models.sae_signatures import FunctionalTiedSAE
+
+    torch.manual_seed(21)
+    M, B, d, n = 4, 512, 128, 512
+    models = [FunctionalTiedSAE.init(d, n, 1e-3, device=DEV) for _ in range(M)]
+    ens_a = FunctionalEnsemble(models, FunctionalTiedSAE, adam, {"lr": 1e-3}, device=DEV, backend="hip")
+    models2 = [({k: v.clone() for k, v in p.items()}, {k: v.clone() for k, v in b.items()})
+               for p, b in ens_a.unstack()]
+    ens_b = FunctionalEnsemble(models2, FunctionalTiedSAE, adam, {"lr": 1e-3}, device=DEV, backend="hip")
+
+    x = torch.randn(B, d, device=DEV)
+    seen = []
+    for _ in range(3):
+        ens_a.step_batch(x)
+        hs = ens_b._hip_step
+        nb = hs.grads_phase(x, on_grads=lambda ts: seen.extend(t.shape for t in ts))
+        hs.update_phase(nb)
+    torch.cuda.synchronize()
+    # callback fired for g_bias + two model-half gw chunks per step
+    assert (M, n) in seen and (M // 2, n, d) in seen
+    for k in ens_a.params:
+        err = (ens_a.params[k] - ens_b.params[k]).abs().max().item()
+        assert err < 1e-6, (k, err)
+
+
+def test_cluster_dispatch_gpu(tmp_path):
+    """P1 (process-per-ensemble, shared chunk) with CUDA-resident ensembles:
+    two children attach to shared GPU tensors and train on one chunk."""
+    import numpy as np
+
+    from sparse_coding_amd.engine.ensemble import FunctionalEnsemble
+    from sparse_coding_amd.functional.optim import adam
+    from sparse_coding_amd.models.sae_signatures import FunctionalTiedSAE
+    from sparse_coding_amd.sweep.cluster_runs import dispatch_job_on_chunk
+    from sparse_coding_amd.sweep.big_sweep import ensemble_train_loop
+
+    torch.manual_seed(22)
+    d, n = 32, 64
+    ensembles = []
+    for gi, l1s in enumerate([(1e-4, 1e-3), (3e-3,)]):
+        models = [FunctionalTiedSAE.init(d, n, l1, device=DEV) for l1 in l1s]
+        ens = FunctionalEnsemble(models, FunctionalTiedSAE, adam, {"lr": 1e-3}, device=DEV, backend="hip")
+        ensembles.append((ens, {"batch_size": 128, "device": DEV, "dict_size": n, "l1_alpha": list(l1s)}, f"ens{gi}"))
+
+    before = [e[0].params["encoder"].clone() for e in ensembles]
+    chunk = torch.randn(1024, d)
+
+    class Cfg:
+        batch_size = 128
+        show_progress = False
+        logger = None
+        ensemble_hyperparams = []
+        buffer_hyperparams = ["l1_alpha"]
+        log_every = 100
+
+    dispatch_job_on_chunk(ensembles, Cfg(), chunk, ensemble_train_loop)
+    for (ens, _, _), enc0 in zip(ensembles, before):
+        assert torch.isfinite(ens.params["encoder"]).all()
+        assert not torch.allclose(ens.params["encoder"], enc0)
