@@ -409,6 +409,30 @@ void k_enc_fwd_t(const float* __restrict__ x,       // [B, d]
       }
       continue;
     }
+    if (mode == 3) {
+      // reverse SAE (sae_ensemble.py:447-503): code = pre * [pre + b > 0]
+      // (the bias is removed from active features before decoding, so the
+      // code can be negative and the L1 partial needs |code|)
+      float bj = col_ok ? bias_m[col] : 0.f;
+      float fired_cnt = 0.f;
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        int row = row0 + g.wr + acc_row(r, g.lane);
+        if (row < B && col_ok) {
+          float pre = acc[tj][r];
+          float v = (pre + bj > 0.f) ? pre : 0.f;
+          c_m[(long)row * n + col] = v;
+          l1_sum += fabsf(v);
+          fired_cnt += (v != 0.f) ? 1.f : 0.f;
+        }
+      }
+      if (col_ok) {
+        float other = __shfl_xor(fired_cnt, 32, WAVE);
+        float tot = fired_cnt + other;
+        if (g.lane < 32 && tot > 0.f) atomicAdd(&fired_m[col], tot);
+      }
+      continue;
+    }
     float bj = col_ok ? bias_m[col] : 0.f;
     // coefficient mask (reference K9): columns >= dict_sizes[m] are dead
     bool live = col_ok && (!dict_sizes || col < dict_sizes[m]);
@@ -429,7 +453,7 @@ void k_enc_fwd_t(const float* __restrict__ x,       // [B, d]
       if (g.lane < 32 && tot > 0.f) atomicAdd(&fired_m[col], tot);
     }
   }
-  if (mode == 0 || mode == 2) {
+  if (mode == 0 || mode == 2 || mode == 3) {
     l1_sum = wave_reduce_sum(l1_sum);
     if (g.lane == 0) atomicAdd(&loss_parts[m * 2 + 1], l1_sum);
   }
@@ -509,7 +533,7 @@ void k_gc_t(const float* __restrict__ r,        // [M, B, d]
             const float* __restrict__ l1_alpha, // [M]
             float* __restrict__ gpre_out,       // [M, B, n]
             float* __restrict__ g_bias,         // [M, n]
-            int B, int d, int n, int prio) {
+            int B, int d, int n, int prio, int gc_mode) {
   __shared__ float As[2][TBK * BMP];
   __shared__ float Bs[2][TBK * BMP];
 
@@ -554,12 +578,20 @@ void k_gc_t(const float* __restrict__ r,        // [M, B, d]
       int row = row0 + g.wr + acc_row(r_, g.lane);
       if (row < B && col_ok) {
         float cv = c_m[(long)row * n + col];
-        float gv = (cv > 0.f) ? (gscale * acc[tj][r_] + l1_term) : 0.f;
+        float gv;
+        if (gc_mode == 1) {
+          // reverse SAE: active set is cv != 0, l1 grad is sign(cv), and
+          // the bias receives no gradient from the code path
+          float sgn = (cv > 0.f) ? 1.f : ((cv < 0.f) ? -1.f : 0.f);
+          gv = (cv != 0.f) ? (gscale * acc[tj][r_] + l1_term * sgn) : 0.f;
+        } else {
+          gv = (cv > 0.f) ? (gscale * acc[tj][r_] + l1_term) : 0.f;
+        }
         g_m[(long)row * n + col] = gv;
         colsum += gv;
       }
     }
-    if (col_ok) {
+    if (col_ok && gc_mode == 0) {
       float other = __shfl_xor(colsum, 32, WAVE);
       float tot = colsum + other;
       if (g.lane < 32 && tot != 0.f) atomicAdd(&gb_m[col], tot);
@@ -957,7 +989,7 @@ void k_gc2_t(const float* __restrict__ rT,       // [M, d, B]
              const float* __restrict__ l1_alpha, // [M]
              float* __restrict__ gpre_out,       // [M, B, n]
              float* __restrict__ g_bias,         // [M, n]
-             int B, int d, int n, int prio) {
+             int B, int d, int n, int prio, int gc_mode) {
   __shared__ float As[2][TBK * BM];
   __shared__ float Bs[2][TBK * BM];
 
@@ -1001,12 +1033,18 @@ void k_gc2_t(const float* __restrict__ rT,       // [M, d, B]
       int row = row0 + g.wr + acc_row(r_, g.lane);
       if (row < B && col_ok) {
         float cv = c_m[(long)row * n + col];
-        float gv = (cv > 0.f) ? (gscale * acc[tj][r_] + l1_term) : 0.f;
+        float gv;
+        if (gc_mode == 1) {
+          float sgn = (cv > 0.f) ? 1.f : ((cv < 0.f) ? -1.f : 0.f);
+          gv = (cv != 0.f) ? (gscale * acc[tj][r_] + l1_term * sgn) : 0.f;
+        } else {
+          gv = (cv > 0.f) ? (gscale * acc[tj][r_] + l1_term) : 0.f;
+        }
         g_m[(long)row * n + col] = gv;
         colsum += gv;
       }
     }
-    if (col_ok) {
+    if (col_ok && gc_mode == 0) {
       float other = __shfl_xor(colsum, 32, WAVE);
       float tot = colsum + other;
       if (g.lane < 32 && tot != 0.f) atomicAdd(&gb_m[col], tot);

@@ -229,15 +229,13 @@ def test_cluster_dispatch_gpu(tmp_path):
     before = [e[0].params["encoder"].clone() for e in ensembles]
     chunk = torch.randn(1024, d)
 
-    class Cfg:
-        batch_size = 128
-        show_progress = False
-        logger = None
-        ensemble_hyperparams = []
-        buffer_hyperparams = ["l1_alpha"]
-        log_every = 100
+    # spawn-context children pickle cfg: must not be a test-local class
+    from types import SimpleNamespace
 
-    dispatch_job_on_chunk(ensembles, Cfg(), chunk, ensemble_train_loop)
+    cfg = SimpleNamespace(batch_size=128, show_progress=False, logger=None,
+                          ensemble_hyperparams=[], buffer_hyperparams=["l1_alpha"],
+                          log_every=100)
+    dispatch_job_on_chunk(ensembles, cfg, chunk, ensemble_train_loop)
     for (ens, _, _), enc0 in zip(ensembles, before):
         assert torch.isfinite(ens.params["encoder"]).all()
         assert not torch.allclose(ens.params["encoder"], enc0)
