@@ -75,6 +75,10 @@ def maybe_make_step(ensemble, required: bool = False) -> Optional["HipSAEStep"]:
         ext = _ops.get_extension(required=True)
         return HipThresholdStep(ensemble, ext)
 
+    if sig is sigs.FunctionalReverseSAE:
+        ext = _ops.get_extension(required=True)
+        return HipSAEStep(ensemble, ext, tied=True, reverse=True)
+
     if sig is sigs.FunctionalMaskedTiedSAE:
         ext = _ops.get_extension(required=True)
         return HipSAEStep(ensemble, ext, tied=True, masked=True)
@@ -100,11 +104,15 @@ def maybe_make_step(ensemble, required: bool = False) -> Optional["HipSAEStep"]:
 class HipSAEStep:
     """Workspaces + kernel-sequence launcher for one ensemble."""
 
-    def __init__(self, ensemble, ext, tied: bool, masked: bool = False):
+    def __init__(self, ensemble, ext, tied: bool, masked: bool = False, reverse: bool = False):
         self.ens = ensemble
         self.ext = ext
         self.tied = tied
         self.masked = masked
+        # reverse SAE (sae_ensemble.py:447-503): code = pre * [pre+b > 0],
+        # |code| in the L1, sign-aware backward, NO bias gradient from the
+        # code path (only the L2-decay term moves the bias)
+        self.reverse = reverse
         self._ws = {}
 
         p = ensemble.params
@@ -164,6 +172,8 @@ class HipSAEStep:
         self.fired = torch.zeros(M, n, device=dev)
         self.g_bias = f(M, n)
         self.gw = f(M, n, d)
+        if self.reverse:
+            self.kc["staging"] = "t"
         if self.kc["staging"] == "pre":
             # pre-transposed operands: every GEMM's staging direct/b128
             self.xT = f(d, B)
@@ -226,10 +236,12 @@ class HipSAEStep:
         else:
             # transpose-in-staging GEMMs (no separate transpose kernels)
             enc_inv = self.inv_norms if self.tied else None
-            ext.enc_fwd(x, enc, bias, enc_inv, self.c, self.loss_parts, self.fired, 0, bk, prio,
+            mode = 3 if self.reverse else 0
+            ext.enc_fwd(x, enc, bias, enc_inv, self.c, self.loss_parts, self.fired, mode, bk, prio,
                         dict_sizes=self.dict_sizes)
             ext.dec_fwd(self.c, dict_w, self.inv_norms, x, self.r, self.loss_parts, bk_dec, prio)
-            ext.gc(self.r, dict_w, self.inv_norms, self.c, self.l1_alpha, self.gpre, self.g_bias, bk, prio)
+            ext.gc(self.r, dict_w, self.inv_norms, self.c, self.l1_alpha, self.gpre, self.g_bias, bk, prio,
+                   gc_mode=1 if self.reverse else 0)
 
         if on_grads is not None:
             on_grads([self.g_bias])  # final after k_gc

@@ -101,7 +101,7 @@ void dec_fwd(torch::Tensor c, torch::Tensor Wdec, torch::Tensor inv_norms,
 
 void gc(torch::Tensor r, torch::Tensor Wdec, torch::Tensor inv_norms,
         torch::Tensor c, torch::Tensor l1_alpha, torch::Tensor gpre,
-        torch::Tensor g_bias, int64_t bk, bool prio) {
+        torch::Tensor g_bias, int64_t bk, bool prio, int64_t gc_mode) {
   CHECK_IN(r); CHECK_IN(Wdec); CHECK_IN(inv_norms); CHECK_IN(c);
   CHECK_IN(l1_alpha); CHECK_IN(gpre); CHECK_IN(g_bias);
   int M = Wdec.size(0), n = Wdec.size(1), d = Wdec.size(2);
@@ -112,13 +112,13 @@ void gc(torch::Tensor r, torch::Tensor Wdec, torch::Tensor inv_norms,
                        r.data_ptr<float>(), Wdec.data_ptr<float>(),
                        inv_norms.data_ptr<float>(), c.data_ptr<float>(),
                        l1_alpha.data_ptr<float>(), gpre.data_ptr<float>(),
-                       g_bias.data_ptr<float>(), B, d, n, prio ? 1 : 0);
+                       g_bias.data_ptr<float>(), B, d, n, prio ? 1 : 0, (int)gc_mode);
   else
     hipLaunchKernelGGL((k_gc_t<32, 4>), grid, dim3(NTHREADS), 0, cur_stream(),
                        r.data_ptr<float>(), Wdec.data_ptr<float>(),
                        inv_norms.data_ptr<float>(), c.data_ptr<float>(),
                        l1_alpha.data_ptr<float>(), gpre.data_ptr<float>(),
-                       g_bias.data_ptr<float>(), B, d, n, prio ? 1 : 0);
+                       g_bias.data_ptr<float>(), B, d, n, prio ? 1 : 0, (int)gc_mode);
 }
 
 void gc_thresh(torch::Tensor r, torch::Tensor Wdec, torch::Tensor inv_norms,
@@ -264,7 +264,7 @@ void enc_fwd2(torch::Tensor xT, torch::Tensor WT, torch::Tensor bias,
 
 void gc2(torch::Tensor rT, torch::Tensor WT, torch::Tensor c,
          torch::Tensor l1_alpha, torch::Tensor gpre, torch::Tensor g_bias,
-         int64_t bk, bool prio) {
+         int64_t bk, bool prio, int64_t gc_mode) {
   CHECK_IN(rT); CHECK_IN(WT); CHECK_IN(c); CHECK_IN(l1_alpha);
   CHECK_IN(gpre); CHECK_IN(g_bias);
   int M = WT.size(0), d = WT.size(1), n = WT.size(2);
@@ -275,13 +275,13 @@ void gc2(torch::Tensor rT, torch::Tensor WT, torch::Tensor c,
                        rT.data_ptr<float>(), WT.data_ptr<float>(),
                        c.data_ptr<float>(), l1_alpha.data_ptr<float>(),
                        gpre.data_ptr<float>(), g_bias.data_ptr<float>(),
-                       B, d, n, prio ? 1 : 0);
+                       B, d, n, prio ? 1 : 0, (int)gc_mode);
   else
     hipLaunchKernelGGL((k_gc2_t<32, 4>), grid, dim3(NTHREADS), 0, cur_stream(),
                        rT.data_ptr<float>(), WT.data_ptr<float>(),
                        c.data_ptr<float>(), l1_alpha.data_ptr<float>(),
                        gpre.data_ptr<float>(), g_bias.data_ptr<float>(),
-                       B, d, n, prio ? 1 : 0);
+                       B, d, n, prio ? 1 : 0, (int)gc_mode);
 }
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -295,7 +295,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gc2", &gc2, "code-grad, pre-transposed operands",
         py::arg("rT"), py::arg("WT"), py::arg("c"), py::arg("l1_alpha"),
         py::arg("gpre"), py::arg("g_bias"),
-        py::arg("bk") = 32, py::arg("prio") = false);
+        py::arg("bk") = 32, py::arg("prio") = false, py::arg("gc_mode") = 0);
   m.def("enc_fwd", &enc_fwd, "fused encoder GEMM + bias + ReLU/TopK/gate (+L1, fired)",
         py::arg("x"), py::arg("Wenc"), py::arg("bias"), py::arg("inv_norms"),
         py::arg("c_out"), py::arg("loss_parts"), py::arg("fired"), py::arg("mode"),
@@ -311,10 +311,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("c"), py::arg("Wdec"), py::arg("inv_norms"), py::arg("x"),
         py::arg("r_out"), py::arg("loss_parts"),
         py::arg("bk") = 32, py::arg("prio") = false);
-  m.def("gc", &gc, "code-gradient GEMM + relu mask + l1 term (+bias grad)",
+  m.def("gc", &gc, "code-gradient GEMM + relu/reverse mask + l1 term (+bias grad)",
         py::arg("r"), py::arg("Wdec"), py::arg("inv_norms"), py::arg("c"),
         py::arg("l1_alpha"), py::arg("gpre"), py::arg("g_bias"),
-        py::arg("bk") = 32, py::arg("prio") = false);
+        py::arg("bk") = 32, py::arg("prio") = false, py::arg("gc_mode") = 0);
   m.def("grad_w", &grad_w, "gw = beta*gw + alpha * P^T Q (batched over M)",
         py::arg("P"), py::arg("Q"), py::arg("gw"), py::arg("alpha"), py::arg("beta"),
         py::arg("bk") = 32, py::arg("prio") = false);

@@ -284,6 +284,37 @@ def test_thresholding_step_matches_torch():
         assert _rel_err(ens_hip.params[k], ens_ref.params[k]) < 2e-3, k
 
 
+def test_reverse_step_matches_torch():
+    """Fused reverse-SAE step (enc mode 3 + gc_mode 1): bias-subtracted
+    codes (possibly negative), |code| L1, and NO bias grad from the code
+    path — vs the vmap oracle."""
+    from sparse_coding_amd.engine.ensemble import FunctionalEnsemble
+    from sparse_coding_amd.functional.optim import adam
+    from sparse_coding_amd.models.sae_signatures import FunctionalReverseSAE
+
+    torch.manual_seed(14)
+    M, B, d, n = 2, 256, 64, 128
+    models = [FunctionalReverseSAE.init(d, n, l1, bias_decay=0.01, device=DEV) for l1 in (1e-3, 3e-3)]
+    ens_hip = FunctionalEnsemble(models, FunctionalReverseSAE, adam, {"lr": 1e-3}, device=DEV, backend="hip")
+    assert getattr(ens_hip._hip_step, "reverse", False)
+    models2 = [({k: v.clone() for k, v in p.items()}, {k: v.clone() for k, v in b.items()})
+               for p, b in ens_hip.unstack()]
+    ens_ref = FunctionalEnsemble(models2, FunctionalReverseSAE, adam, {"lr": 1e-3}, device=DEV, backend="torch")
+    # nonzero bias so the bias-subtract path actually differs from tied
+    with torch.no_grad():
+        ens_hip.params["encoder_bias"].normal_(0, 0.2)
+        ens_ref.params["encoder_bias"].copy_(ens_hip.params["encoder_bias"])
+    x = torch.randn(B, d, device=DEV)
+    for i in range(4):
+        l_hip, aux_hip = ens_hip.step_batch(x)
+        l_ref, aux_ref = ens_ref.step_batch(x)
+        assert _rel_err(l_hip["loss"], l_ref["loss"]) < 1e-4, i
+        assert _rel_err(aux_hip["c"], aux_ref["c"]) < 1e-3, i
+        assert (aux_hip["c"] < 0).any()  # reverse codes go negative
+    for k in ens_ref.params:
+        assert _rel_err(ens_hip.params[k], ens_ref.params[k]) < 2e-3, k
+
+
 def test_masked_step_matches_torch():
     """Fused masked steps (K9): different dict sizes stacked to one width;
     coefficient columns >= dict_size[m] stay zero and get no grads."""
