@@ -79,6 +79,16 @@ def maybe_make_step(ensemble, required: bool = False) -> Optional["HipSAEStep"]:
         ext = _ops.get_extension(required=True)
         return HipSAEStep(ensemble, ext, tied=True, reverse=True)
 
+    if sig is sigs.FunctionalTiedCenteredSAE:
+        ext = _ops.get_extension(required=True)
+        return HipCenteredStep(ensemble, ext)
+
+    from sparse_coding_amd.models.positive import FunctionalPositiveTiedSAE
+
+    if sig is FunctionalPositiveTiedSAE:
+        ext = _ops.get_extension(required=True)
+        return HipPositiveStep(ensemble, ext)
+
     if sig is sigs.FunctionalMaskedTiedSAE:
         ext = _ops.get_extension(required=True)
         return HipSAEStep(ensemble, ext, tied=True, masked=True)
@@ -341,6 +351,155 @@ class HipSAEStep:
         if not self.tied:
             ts.append(self.gw_enc)
         return ts
+
+
+class HipCenteredStep(HipSAEStep):
+    """Fused step for FunctionalTiedCenteredSAE (sae_ensemble.py:164-230):
+    the tied pipeline runs on the per-model centered input x' = x - t[m]
+    (enc/dec kernels take an [M,B,d] x with a model stride), and the
+    learnable center's gradient falls out of workspaces that already exist:
+        dL/dt = gscale * sum_b r_b  -  g_bias @ What
+    (g_bias is the column sum of gpre from k_gc)."""
+
+    def __init__(self, ensemble, ext):
+        super().__init__(ensemble, ext, tied=True)
+
+    def _alloc(self, B: int):
+        super()._alloc(B)
+        M, n, d = self.n_models, self.n_dict, self.d_act
+        dev = self.ens.params["encoder"].device
+        self.xc = torch.empty(M, B, d, device=dev)
+        self.g_center = torch.empty(M, d, device=dev)
+        self.zero_decay = torch.zeros_like(self.bias_decay)
+        self.kc["staging"] = "t"
+
+    def grads_phase(self, x: torch.Tensor, on_grads=None):
+        ens, ext = self.ens, self.ext
+        B = x.shape[0]
+        if self._B != B:
+            self._alloc(B)
+        x = x.contiguous()
+        p = ens.params
+        enc = p["encoder"]
+        bias = p["encoder_bias"]
+        center = p["center"]
+        bk, prio = self.kc["bk"], self.kc["prio"]
+        bk_gw = self.kc["bk_grad_w"] or bk
+
+        self.loss_parts.zero_()
+        self.g_bias.zero_()
+
+        torch.sub(x.unsqueeze(0), center.unsqueeze(1), out=self.xc)
+        ext.row_norms(enc, self.norms, self.inv_norms, EPS_NORM)
+        ext.enc_fwd(self.xc, enc, bias, self.inv_norms, self.c,
+                    self.loss_parts, self.fired, 0, bk, prio)
+        ext.dec_fwd(self.c, enc, self.inv_norms, self.xc, self.r,
+                    self.loss_parts, self.kc["bk_dec"] or bk, prio)
+        ext.gc(self.r, enc, self.inv_norms, self.c, self.l1_alpha,
+               self.gpre, self.g_bias, bk, prio)
+
+        gscale = 2.0 / (B * self.d_act)
+        ext.grad_w(self.c, self.r, self.gw, gscale, 0.0, bk_gw, prio)
+        ext.grad_w(self.gpre, self.xc, self.gw, 1.0, 1.0, bk_gw, prio)
+        # center gradient from existing reductions (docstring derivation)
+        torch.sum(self.r, dim=1, out=self.g_center)
+        self.g_center.mul_(gscale)
+        self.g_center.sub_(torch.einsum("mn,mnd->md", self.g_bias * self.inv_norms, enc))
+        if on_grads is not None:
+            on_grads([self.gw, self.g_bias, self.g_center])
+        return B
+
+    def update_phase(self, B: int):
+        ens, ext = self.ens, self.ext
+        st = ens.optim_states
+        st["step"] += 1.0
+        step_no = st["step"]
+        p = ens.params
+        ext.project_adam(p["encoder"], self.gw, self.norms,
+                         st["mu"]["encoder"], st["nu"]["encoder"], step_no,
+                         self.n_dict, self.lr, self.beta1, self.beta2,
+                         self.eps, EPS_NORM, True)
+        ext.bias_adam(p["encoder_bias"], self.g_bias, self.zero_decay,
+                      st["mu"]["encoder_bias"], st["nu"]["encoder_bias"],
+                      step_no, self.lr, self.beta1, self.beta2, self.eps)
+        ext.bias_adam(p["center"], self.g_center, self.zero_decay,
+                      st["mu"]["center"], st["nu"]["center"],
+                      step_no, self.lr, self.beta1, self.beta2, self.eps)
+
+    def _loss_data(self, B: int):
+        mse = self.loss_parts[:, 0] / (B * self.d_act)
+        l1 = self.l1_alpha * self.loss_parts[:, 1] / B
+        return {"loss": mse + l1, "l_reconstruction": mse, "l_l1": l1}
+
+    def dp_grad_tensors(self):
+        return [self.gw, self.g_bias, self.g_center]
+
+
+class HipPositiveStep(HipSAEStep):
+    """Fused step for FunctionalPositiveTiedSAE (mlp_tests.py:68-125):
+    the tied pipeline runs on Wc = clamp(W, 0) and the shifted input
+    x' = x + 0.18; the projected gradient is masked by the clamp derivative
+    before Adam updates the raw (signed) encoder."""
+
+    def __init__(self, ensemble, ext):
+        super().__init__(ensemble, ext, tied=True)
+
+    def _alloc(self, B: int):
+        super()._alloc(B)
+        M, n, d = self.n_models, self.n_dict, self.d_act
+        dev = self.ens.params["encoder"].device
+        self.Wc = torch.empty(M, n, d, device=dev)
+        self.x_shift = torch.empty(B, d, device=dev)
+        self.kc["staging"] = "t"
+
+    def grads_phase(self, x: torch.Tensor, on_grads=None):
+        from sparse_coding_amd.models.positive import INPUT_SHIFT
+
+        ens, ext = self.ens, self.ext
+        B = x.shape[0]
+        if self._B != B:
+            self._alloc(B)
+        x = x.contiguous()
+        p = ens.params
+        enc = p["encoder"]
+        bias = p["encoder_bias"]
+        bk, prio = self.kc["bk"], self.kc["prio"]
+        bk_gw = self.kc["bk_grad_w"] or bk
+
+        self.loss_parts.zero_()
+        self.g_bias.zero_()
+
+        torch.clamp(enc, min=0.0, out=self.Wc)
+        torch.add(x, INPUT_SHIFT, out=self.x_shift)
+        ext.row_norms(self.Wc, self.norms, self.inv_norms, EPS_NORM)
+        ext.enc_fwd(self.x_shift, self.Wc, bias, self.inv_norms, self.c,
+                    self.loss_parts, self.fired, 0, bk, prio)
+        ext.dec_fwd(self.c, self.Wc, self.inv_norms, self.x_shift, self.r,
+                    self.loss_parts, self.kc["bk_dec"] or bk, prio)
+        ext.gc(self.r, self.Wc, self.inv_norms, self.c, self.l1_alpha,
+               self.gpre, self.g_bias, bk, prio)
+
+        gscale = 2.0 / (B * self.d_act)
+        ext.grad_w(self.c, self.r, self.gw, gscale, 0.0, bk_gw, prio)
+        ext.grad_w(self.gpre, self.x_shift, self.gw, 1.0, 1.0, bk_gw, prio)
+        if on_grads is not None:
+            on_grads([self.gw, self.g_bias])
+        return B
+
+    def update_phase(self, B: int):
+        ens, ext = self.ens, self.ext
+        st = ens.optim_states
+        st["step"] += 1.0
+        step_no = st["step"]
+        p = ens.params
+        ext.project_adam(p["encoder"], self.gw, self.norms,
+                         st["mu"]["encoder"], st["nu"]["encoder"], step_no,
+                         self.n_dict, self.lr, self.beta1, self.beta2,
+                         self.eps, EPS_NORM, True,
+                         w_used=self.Wc, clamp_mask=True)
+        ext.bias_adam(p["encoder_bias"], self.g_bias, self.bias_decay,
+                      st["mu"]["encoder_bias"], st["nu"]["encoder_bias"],
+                      step_no, self.lr, self.beta1, self.beta2, self.eps)
 
 
 class HipThresholdStep(HipSAEStep):

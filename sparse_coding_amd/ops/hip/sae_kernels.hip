@@ -333,8 +333,9 @@ void k_enc_fwd_t(const float* __restrict__ x,       // [B, d]
                  const float* __restrict__ act_scale,  // [M, n] (mode 2)
                  const float* __restrict__ act_gain,   // [M, n] (mode 2)
                  float* __restrict__ u_out,            // [M, B, n] (mode 2)
-                 const int* __restrict__ dict_sizes) { // [M] or nullptr:
+                 const int* __restrict__ dict_sizes, // [M] or nullptr:
                                        // masked sigs zero cols >= dict_sizes[m]
+                 long x_mstride) {     // 0: x shared [B,d]; else x is [M,B,d]
   __shared__ float As[2][TBK * BMP];
   __shared__ float Bs[2][TBK * BMP];
 
@@ -345,6 +346,7 @@ void k_enc_fwd_t(const float* __restrict__ x,       // [B, d]
   const int row0 = ty * BM;
   const int col0 = tx * BN;
   const float* W = Wenc + (long)m * n * d;
+  const float* x_m = x + (long)m * x_mstride;
   const float* inv = inv_norms ? inv_norms + (long)m * n : nullptr;
   const bool scaled = inv != nullptr;
 
@@ -352,7 +354,7 @@ void k_enc_fwd_t(const float* __restrict__ x,       // [B, d]
   zero_acc(acc);
   TStage<TBK> sa, sb;
 
-#define ENC_LA(K) stage_T_load<TBK>(x, d, row0, (K), B, d, nullptr, sa)
+#define ENC_LA(K) stage_T_load<TBK>(x_m, d, row0, (K), B, d, nullptr, sa)
 #define ENC_LB(K) stage_T_load<TBK>(W, d, col0, (K), n, d, inv, sb)
 #define ENC_WA(BUF) stage_T_write<TBK>(sa, &As[BUF][0], false)
 #define ENC_WB(BUF) stage_T_write<TBK>(sb, &Bs[BUF][0], scaled)
@@ -470,7 +472,8 @@ void k_dec_fwd_t(const float* __restrict__ c,       // [M, B, n]
                  const float* __restrict__ x,       // [B, d]
                  float* __restrict__ r_out,         // [M, B, d]
                  float* __restrict__ loss_parts,    // [M, 2]
-                 int B, int d, int n, int prio) {
+                 int B, int d, int n, int prio,
+                 long x_mstride) {  // 0: shared [B,d]; else [M,B,d]
   __shared__ float As[2][TBK * BMP];
   __shared__ float Bs[2][TBK * BM];
 
@@ -482,6 +485,7 @@ void k_dec_fwd_t(const float* __restrict__ c,       // [M, B, n]
   const int col0 = tx * BN;
   const float* c_m = c + (long)m * B * n;
   const float* W = Wdec + (long)m * n * d;
+  const float* x_m = x + (long)m * x_mstride;
   const float* inv = inv_norms + (long)m * n;
 
   f32x16 acc[2];
@@ -511,7 +515,7 @@ void k_dec_fwd_t(const float* __restrict__ c,       // [M, B, n]
     for (int r = 0; r < 16; ++r) {
       int row = row0 + g.wr + acc_row(r, g.lane);
       if (row < B && col_ok) {
-        float rv = acc[tj][r] - x[(long)row * d + col];
+        float rv = acc[tj][r] - x_m[(long)row * d + col];
         r_m[(long)row * d + col] = rv;
         mse_sum += rv * rv;
       }
@@ -762,13 +766,16 @@ void k_project_adam(float* __restrict__ W,        // [M*n, d]
                     const float* __restrict__ step_no,  // [M]
                     int n_rows_total, int n_per_model, int d,
                     float lr, float b1, float b2, float eps_adam,
-                    float eps_norm, int project) {
+                    float eps_norm, int project,
+                    const float* __restrict__ w_used_p,  // nullptr -> W
+                    int clamp_mask) {  // 1: g *= [W >= 0] (clamp backward)
   const int wave = threadIdx.x / WAVE;
   const int lane = threadIdx.x & (WAVE - 1);
   const int row = blockIdx.x * (NTHREADS / WAVE) + wave;
   if (row >= n_rows_total) return;
 
-  float* w = W + (long)row * d;
+  float* w_param = W + (long)row * d;
+  const float* w = w_used_p ? w_used_p + (long)row * d : w_param;
   const float* g_in = gw + (long)row * d;
   float* mu_r = mu + (long)row * d;
   float* nu_r = nu + (long)row * d;
@@ -799,11 +806,12 @@ void k_project_adam(float* __restrict__ W,        // [M*n, d]
       if (do_proj) g = (g - dot_scaled * w[j]) * inv_s;
       else g = g * inv_s;
     }
+    if (clamp_mask && w_param[j] < 0.f) g = 0.f;
     float m1 = b1 * mu_r[j] + (1.0f - b1) * g;
     float v1 = b2 * nu_r[j] + (1.0f - b2) * g * g;
     mu_r[j] = m1;
     nu_r[j] = v1;
-    w[j] -= lr * (m1 / bc1) / (sqrtf(v1 / bc2) + eps_adam);
+    w_param[j] -= lr * (m1 / bc1) / (sqrtf(v1 / bc2) + eps_adam);
   }
 }
 

@@ -39,7 +39,12 @@ void enc_fwd(torch::Tensor x, torch::Tensor Wenc, torch::Tensor bias,
   CHECK_IN(x); CHECK_IN(Wenc); CHECK_IN(bias); CHECK_IN(c_out);
   CHECK_IN(loss_parts); CHECK_IN(fired);
   int M = Wenc.size(0), n = Wenc.size(1), d = Wenc.size(2);
-  int B = x.size(0);
+  long x_mstride = 0;
+  if (x.dim() == 3) {
+    TORCH_CHECK(x.size(0) == M, "per-model x must be [M,B,d]");
+    x_mstride = (long)x.size(1) * x.size(2);
+  }
+  int B = x.size(x.dim() - 2);
   TORCH_CHECK(d % 4 == 0 && n % 4 == 0, "d and n must be multiples of 4 (float4 staging)");
   const float* inv = nullptr;
   if (inv_norms.has_value()) {
@@ -68,13 +73,13 @@ void enc_fwd(torch::Tensor x, torch::Tensor Wenc, torch::Tensor bias,
                        x.data_ptr<float>(), Wenc.data_ptr<float>(),
                        bias.data_ptr<float>(), inv, c_out.data_ptr<float>(),
                        loss_parts.data_ptr<float>(), fired.data_ptr<float>(),
-                       B, d, n, (int)mode, prio ? 1 : 0, a_p, gn_p, u_p, ds_p);
+                       B, d, n, (int)mode, prio ? 1 : 0, a_p, gn_p, u_p, ds_p, x_mstride);
   else
     hipLaunchKernelGGL((k_enc_fwd_t<32, 4>), grid, dim3(NTHREADS), 0, cur_stream(),
                        x.data_ptr<float>(), Wenc.data_ptr<float>(),
                        bias.data_ptr<float>(), inv, c_out.data_ptr<float>(),
                        loss_parts.data_ptr<float>(), fired.data_ptr<float>(),
-                       B, d, n, (int)mode, prio ? 1 : 0, a_p, gn_p, u_p, ds_p);
+                       B, d, n, (int)mode, prio ? 1 : 0, a_p, gn_p, u_p, ds_p, x_mstride);
 }
 
 void dec_fwd(torch::Tensor c, torch::Tensor Wdec, torch::Tensor inv_norms,
@@ -83,20 +88,25 @@ void dec_fwd(torch::Tensor c, torch::Tensor Wdec, torch::Tensor inv_norms,
   CHECK_IN(c); CHECK_IN(Wdec); CHECK_IN(inv_norms); CHECK_IN(x);
   CHECK_IN(r_out); CHECK_IN(loss_parts);
   int M = Wdec.size(0), n = Wdec.size(1), d = Wdec.size(2);
-  int B = x.size(0);
+  long x_mstride = 0;
+  if (x.dim() == 3) {
+    TORCH_CHECK(x.size(0) == M, "per-model x must be [M,B,d]");
+    x_mstride = (long)x.size(1) * x.size(2);
+  }
+  int B = x.size(x.dim() - 2);
   dim3 grid(cdiv(d, BN), cdiv(B, BM), M);
   if (bk == 16)
     hipLaunchKernelGGL((k_dec_fwd_t<16, 6>), grid, dim3(NTHREADS), 0, cur_stream(),
                        c.data_ptr<float>(), Wdec.data_ptr<float>(),
                        inv_norms.data_ptr<float>(), x.data_ptr<float>(),
                        r_out.data_ptr<float>(), loss_parts.data_ptr<float>(),
-                       B, d, n, prio ? 1 : 0);
+                       B, d, n, prio ? 1 : 0, x_mstride);
   else
     hipLaunchKernelGGL((k_dec_fwd_t<32, 4>), grid, dim3(NTHREADS), 0, cur_stream(),
                        c.data_ptr<float>(), Wdec.data_ptr<float>(),
                        inv_norms.data_ptr<float>(), x.data_ptr<float>(),
                        r_out.data_ptr<float>(), loss_parts.data_ptr<float>(),
-                       B, d, n, prio ? 1 : 0);
+                       B, d, n, prio ? 1 : 0, x_mstride);
 }
 
 void gc(torch::Tensor r, torch::Tensor Wdec, torch::Tensor inv_norms,
@@ -181,9 +191,15 @@ void grad_w(torch::Tensor P, torch::Tensor Q, torch::Tensor gw,
 void project_adam(torch::Tensor W, torch::Tensor gw, torch::Tensor norms,
                   torch::Tensor mu, torch::Tensor nu, torch::Tensor step_no,
                   long n_per_model, double lr, double b1, double b2,
-                  double eps_adam, double eps_norm, bool project) {
+                  double eps_adam, double eps_norm, bool project,
+                  c10::optional<torch::Tensor> w_used, bool clamp_mask) {
   CHECK_IN(W); CHECK_IN(gw); CHECK_IN(norms); CHECK_IN(mu); CHECK_IN(nu);
   CHECK_IN(step_no);
+  const float* wu = nullptr;
+  if (w_used.has_value()) {
+    CHECK_IN(w_used.value());
+    wu = w_used->data_ptr<float>();
+  }
   long rows = W.numel() / W.size(-1);
   int d = W.size(-1);
   dim3 grid(cdiv(rows, NTHREADS / WAVE));
@@ -193,7 +209,7 @@ void project_adam(torch::Tensor W, torch::Tensor gw, torch::Tensor norms,
                      nu.data_ptr<float>(), step_no.data_ptr<float>(),
                      (int)rows, (int)n_per_model, d, (float)lr, (float)b1,
                      (float)b2, (float)eps_adam, (float)eps_norm,
-                     project ? 1 : 0);
+                     project ? 1 : 0, wu, clamp_mask ? 1 : 0);
 }
 
 void bias_adam(torch::Tensor bias, torch::Tensor g_bias, torch::Tensor decay,
@@ -318,6 +334,11 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("grad_w", &grad_w, "gw = beta*gw + alpha * P^T Q (batched over M)",
         py::arg("P"), py::arg("Q"), py::arg("gw"), py::arg("alpha"), py::arg("beta"),
         py::arg("bk") = 32, py::arg("prio") = false);
-  m.def("project_adam", &project_adam, "renorm-gradient projection + Adam");
+  m.def("project_adam", &project_adam, "renorm-gradient projection + Adam",
+        py::arg("W"), py::arg("gw"), py::arg("norms"), py::arg("mu"),
+        py::arg("nu"), py::arg("step_no"), py::arg("n_per_model"),
+        py::arg("lr"), py::arg("b1"), py::arg("b2"), py::arg("eps_adam"),
+        py::arg("eps_norm"), py::arg("project"),
+        py::arg("w_used") = py::none(), py::arg("clamp_mask") = false);
   m.def("bias_adam", &bias_adam, "Adam on bias with L2-norm decay");
 }

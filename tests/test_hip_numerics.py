@@ -284,6 +284,66 @@ def test_thresholding_step_matches_torch():
         assert _rel_err(ens_hip.params[k], ens_ref.params[k]) < 2e-3, k
 
 
+def test_centered_step_matches_torch():
+    """Fused tied-centered step: per-model x' = x - t[m] through the tied
+    pipeline; learnable-center grad = gscale*sum_b r - g_bias @ What."""
+    from sparse_coding_amd.engine.ensemble import FunctionalEnsemble
+    from sparse_coding_amd.functional.optim import adam
+    from sparse_coding_amd.models.sae_signatures import FunctionalTiedCenteredSAE
+
+    torch.manual_seed(15)
+    M, B, d, n = 2, 256, 64, 128
+    models = [FunctionalTiedCenteredSAE.init(d, n, l1, device=DEV) for l1 in (1e-3, 3e-3)]
+    ens_hip = FunctionalEnsemble(models, FunctionalTiedCenteredSAE, adam, {"lr": 1e-3},
+                                 device=DEV, backend="hip")
+    assert type(ens_hip._hip_step).__name__ == "HipCenteredStep"
+    models2 = [({k: v.clone() for k, v in p.items()}, {k: v.clone() for k, v in b.items()})
+               for p, b in ens_hip.unstack()]
+    ens_ref = FunctionalEnsemble(models2, FunctionalTiedCenteredSAE, adam, {"lr": 1e-3},
+                                 device=DEV, backend="torch")
+    with torch.no_grad():  # nonzero center so x' differs per model
+        ens_hip.params["center"].normal_(0, 0.3)
+        ens_ref.params["center"].copy_(ens_hip.params["center"])
+    x = torch.randn(B, d, device=DEV) + 0.5
+    for i in range(4):
+        l_hip, _ = ens_hip.step_batch(x)
+        l_ref, _ = ens_ref.step_batch(x)
+        assert _rel_err(l_hip["loss"], l_ref["loss"]) < 1e-4, i
+    for k in ens_ref.params:
+        assert _rel_err(ens_hip.params[k], ens_ref.params[k]) < 2e-3, k
+
+
+def test_positive_step_matches_torch():
+    """Fused positive-tied step: pipeline on Wc = clamp(W,0) and x + 0.18,
+    clamp-masked projected gradient updating the raw encoder."""
+    from sparse_coding_amd.engine.ensemble import FunctionalEnsemble
+    from sparse_coding_amd.functional.optim import adam
+    from sparse_coding_amd.models.positive import FunctionalPositiveTiedSAE
+
+    torch.manual_seed(16)
+    M, B, d, n = 2, 256, 64, 128
+    models = [FunctionalPositiveTiedSAE.init(d, n, l1, bias_decay=0.01, device=DEV)
+              for l1 in (1e-3, 3e-3)]
+    ens_hip = FunctionalEnsemble(models, FunctionalPositiveTiedSAE, adam, {"lr": 1e-3},
+                                 device=DEV, backend="hip")
+    assert type(ens_hip._hip_step).__name__ == "HipPositiveStep"
+    models2 = [({k: v.clone() for k, v in p.items()}, {k: v.clone() for k, v in b.items()})
+               for p, b in ens_hip.unstack()]
+    ens_ref = FunctionalEnsemble(models2, FunctionalPositiveTiedSAE, adam, {"lr": 1e-3},
+                                 device=DEV, backend="torch")
+    # drive some weights negative so the clamp mask matters
+    with torch.no_grad():
+        ens_hip.params["encoder"].sub_(0.05)
+        ens_ref.params["encoder"].copy_(ens_hip.params["encoder"])
+    x = torch.randn(B, d, device=DEV)
+    for i in range(4):
+        l_hip, _ = ens_hip.step_batch(x)
+        l_ref, _ = ens_ref.step_batch(x)
+        assert _rel_err(l_hip["loss"], l_ref["loss"]) < 1e-4, i
+    for k in ens_ref.params:
+        assert _rel_err(ens_hip.params[k], ens_ref.params[k]) < 2e-3, k
+
+
 def test_reverse_step_matches_torch():
     """Fused reverse-SAE step (enc mode 3 + gc_mode 1): bias-subtracted
     codes (possibly negative), |code| L1, and NO bias grad from the code
