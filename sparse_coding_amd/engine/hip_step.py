@@ -1,20 +1,30 @@
 """Fused CDNA4 training step for SAE ensembles.
 
 Implements the reference's vmapped grad+Adam `step_batch`
-(autoencoders/ensemble.py:119-123,175-193 + the FunctionalSAE /
-FunctionalTiedSAE losses in sae_ensemble.py) as a pipeline of hand-written
-gfx950 HIP kernels (sources in ``sparse_coding_amd/ops/hip/sae_kernels.hip``):
+(autoencoders/ensemble.py:119-123,175-193 + the signature losses in
+sae_ensemble.py / topk_encoder.py / mlp_tests.py) as a pipeline of
+hand-written gfx950 HIP kernels (``sparse_coding_amd/ops/hip/sae_kernels.hip``):
 
   k_row_norms    : per-row ||w|| and 1/max(||w||,1e-8) of the dictionary
-  k_enc_fwd      : encoder GEMM -> +bias -> ReLU -> c  (+L1 partial, fired)
+  k_enc_fwd      : encoder GEMM -> epilogue by mode: bias+ReLU (0), raw
+                   scores for TopK (1), threshold gate (2), reverse
+                   bias-subtract (3); + L1 partial, fired counts, optional
+                   per-model coef mask (K9) and per-model x stride
   k_dec_fwd      : decoder GEMM (rows scaled by inv-norm) -> residual r
                    (+MSE partial)
-  k_gc           : code-grad GEMM -> relu mask -> +l1 term -> g_pre
-                   (+bias-grad column sums)
+  k_gc           : code-grad GEMM -> relu/sign mask + l1 term -> g_pre
+                   (+bias-grad column sums); k_gc_thresh adds the gate
+                   derivative and gain/scale column-sum grads
   k_grad_w       : batch-contraction GEMMs -> dL/d(W_hat) (and encoder grad)
   k_project_adam : analytic gradient of w/max(||w||,eps) (the in-forward
-                   decoder renormalization) + fused Adam
-  k_bias_adam    : Adam on the bias (+ L2-decay gradient)
+                   decoder renormalization) + fused Adam; optional separate
+                   "used" weights + clamp-derivative mask (positive SAE)
+  k_bias_adam    : Adam on any [M,k] vector param (+ L2-decay gradient)
+
+Fused-step coverage: tied, untied, masked tied/untied, thresholding,
+reverse, tied-centered, positive-tied, TopK.  LISTA and the semilinear
+(MLP-encoder) SAEs stay on the vmap backend by design: their hot ops are
+plain batched GEMMs that vmap lowers to rocBLAS strided-batched GEMM.
 
 All GEMMs run on the exact-f32 MFMA path (v_mfma_f32_32x32x2_f32): fp32 end
 to end, the reference's training dtype.  Validated against the
