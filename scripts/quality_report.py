@@ -35,6 +35,8 @@ def main():
     p.add_argument("--n-models", type=int, default=8)
     p.add_argument("--out-dir", default="docs/quality")
     p.add_argument("--backend", default="auto")
+    p.add_argument("--resample-every", type=int, default=0,
+                   help="resample dead features every N steps (0 = off)")
     args = p.parse_args()
 
     device = "cuda:0" if torch.cuda.is_available() else "cpu"
@@ -60,9 +62,21 @@ def main():
     backend = type(ens._hip_step).__name__ if ens._hip_step else "torch"
     print(f"training {args.n_models} models, backend={backend}")
 
+    resampler = None
+    if args.resample_every:
+        from sparse_coding_amd.engine.resample import EnsembleResampler
+
+        resampler = EnsembleResampler(ens, n_track=512)
+
     t0 = time.time()
     for step in range(args.steps):
-        losses, _ = ens.step_batch(gen.send(None))
+        batch = gen.send(None)
+        losses, aux = ens.step_batch(batch)
+        if resampler is not None:
+            resampler.observe(batch, aux)
+            if (step + 1) % args.resample_every == 0:
+                n_dead = resampler.resample()
+                print(f"step {step}: resampled {n_dead.tolist() if hasattr(n_dead, 'tolist') else n_dead}")
         if step % 500 == 0:
             print(f"step {step}: loss={[round(v, 4) for v in losses['loss'].tolist()]}")
     if device.startswith("cuda"):
@@ -91,6 +105,7 @@ def main():
             "batch": args.batch, "steps": args.steps, "backend": backend,
             "n_ground_truth": args.n_true, "feature_num_nonzero": args.nonzero,
             "dtype": "fp32", "data": "synthetic ground-truth sparse dict",
+            "resample_every": args.resample_every,
         },
         "train_seconds": train_s,
         "acts_per_sec": args.batch * args.steps / train_s,

@@ -159,8 +159,9 @@ class DataParallelEnsembleTrainer:
                     for t in tensors:
                         works.append((dist.all_reduce(t, async_op=True, group=self.group), t))
 
-            B = hs.grads_phase(local_batch, on_grads=on_grads if self.world_size > 1 else None)
-            if self.world_size > 1:
+            use_dp = self.world_size > 1 or self.force_dp_path
+            B = hs.grads_phase(local_batch, on_grads=on_grads if use_dp else None)
+            if use_dp:
                 for w, t in works:
                     w.wait()
                 if comm_stream is not None:

@@ -239,3 +239,43 @@ def test_cluster_dispatch_gpu(tmp_path):
     for (ens, _, _), enc0 in zip(ensembles, before):
         assert torch.isfinite(ens.params["encoder"]).all()
         assert not torch.allclose(ens.params["encoder"], enc0)
+
+
+def test_dp_rccl_one_rank_overlap_path(tmp_path):
+    """The 8-GPU scaling bench's exact code path — fused step split +
+    all-reduces launched from a side stream via RCCL — on a 1-rank nccl
+    process group (all_reduce is then an on-device no-op, but the full
+    RCCL/stream call sequence runs)."""
+    import os
+
+    import torch.distributed as dist
+
+    from sparse_coding_amd.engine.ensemble import FunctionalEnsemble
+    from sparse_coding_amd.functional.optim import adam
+    from sparse_coding_amd.models.sae_signatures import FunctionalTiedSAE
+    from sparse_coding_amd.parallel.dp import DataParallelEnsembleTrainer
+
+    os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+    os.environ.setdefault("MASTER_PORT", "29541")
+    dist.init_process_group("nccl", rank=0, world_size=1)
+    try:
+        torch.manual_seed(31)
+        M, B, d, n = 4, 1024, 128, 512
+        models = [FunctionalTiedSAE.init(d, n, 1e-3, device=DEV) for _ in range(M)]
+        ens_dp = FunctionalEnsemble(models, FunctionalTiedSAE, adam, {"lr": 1e-3}, device=DEV, backend="hip")
+        models2 = [({k: v.clone() for k, v in p.items()}, {k: v.clone() for k, v in b.items()})
+                   for p, b in ens_dp.unstack()]
+        ens_ref = FunctionalEnsemble(models2, FunctionalTiedSAE, adam, {"lr": 1e-3}, device=DEV, backend="hip")
+
+        trainer = DataParallelEnsembleTrainer(ens_dp, force_dp_path=True)
+        x = torch.randn(B, d, device=DEV)
+        for _ in range(3):
+            losses, _ = trainer.step(x)
+            ens_ref.step_batch(x)
+        torch.cuda.synchronize()
+        assert torch.isfinite(losses["loss"]).all()
+        for k in ens_ref.params:
+            err = (ens_dp.params[k] - ens_ref.params[k]).abs().max().item()
+            assert err < 1e-6, (k, err)
+    finally:
+        dist.destroy_process_group()
