@@ -113,12 +113,17 @@ class DataParallelEnsembleTrainer:
     Adam update locally — states stay bit-identical across ranks without a
     broadcast."""
 
-    def __init__(self, ensemble, bucket_bytes: int = 64 << 20, group=None):
+    def __init__(self, ensemble, bucket_bytes: int = 64 << 20, group=None,
+                 force_dp_path: bool = False):
         self.ensemble = ensemble
         self.group = group
         self._reducer: Optional[GradBucketAllReducer] = None
         self.bucket_bytes = bucket_bytes
         self.world_size = dist.get_world_size(group) if dist.is_initialized() else 1
+        # force_dp_path: run the overlapped all-reduce path even at world
+        # size 1 — lets a single-GPU box exercise the exact RCCL call
+        # sequence the 8-GPU scaling run uses (tests/test_full_stack_gpu.py)
+        self.force_dp_path = force_dp_path
 
     def broadcast_state(self) -> None:
         """One-time parameter/optimizer broadcast from rank 0 (use when
@@ -131,7 +136,7 @@ class DataParallelEnsembleTrainer:
                     dist.broadcast(leaf, src=0, group=self.group)
 
     def step(self, local_batch: torch.Tensor):
-        if self.world_size == 1:
+        if self.world_size == 1 and not self.force_dp_path:
             # single-process: take the ensemble's own step (hipGraph-captured
             # on the fused path)
             return self.ensemble.step_batch(local_batch)
