@@ -74,7 +74,9 @@ def main():
         losses, aux = ens.step_batch(batch)
         if resampler is not None:
             resampler.observe(batch, aux)
-            if (step + 1) % args.resample_every == 0:
+            # no resampling in the final quarter: freshly re-initialized
+            # features need training time before they count as alive
+            if (step + 1) % args.resample_every == 0 and step < args.steps * 3 // 4:
                 n_dead = resampler.resample()
                 print(f"step {step}: resampled {n_dead.tolist() if hasattr(n_dead, 'tolist') else n_dead}")
         if step % 500 == 0:

@@ -58,7 +58,11 @@ def unstack_dict(tree, n_models: int, device=None):
     per_model: List[list] = [[] for _ in range(n_models)]
     for leaf in leaves:
         for i in range(n_models):
-            per_model[i].append(leaf[i].to(device=device))
+            # copy=True: unstacking is a SNAPSHOT even when device is
+            # unchanged (a plain .to() would alias live training memory on
+            # same-device unstacks and every "checkpoint" would silently
+            # track the ensemble)
+            per_model[i].append(leaf[i].to(device=device, copy=True))
     return [tree_unflatten(spec, ls) for ls in per_model]
 
 
