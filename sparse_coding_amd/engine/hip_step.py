@@ -22,9 +22,10 @@ hand-written gfx950 HIP kernels (``sparse_coding_amd/ops/hip/sae_kernels.hip``):
   k_bias_adam    : Adam on any [M,k] vector param (+ L2-decay gradient)
 
 Fused-step coverage: tied, untied, masked tied/untied, thresholding,
-reverse, tied-centered, positive-tied, TopK.  LISTA and the semilinear
-(MLP-encoder) SAEs stay on the vmap backend by design: their hot ops are
-plain batched GEMMs that vmap lowers to rocBLAS strided-batched GEMM.
+reverse, tied-centered, positive-tied, TopK, and LISTA (HipLISTAStep:
+hand-derived backward through the unrolled learned-ISTA layers, every GEMM
+on the MFMA kernels).  The semilinear (MLP-encoder) and residual-denoising
+SAEs stay on the vmap backend: plain batched GEMMs, rocBLAS-served.
 
 All GEMMs run on the exact-f32 MFMA path (v_mfma_f32_32x32x2_f32): fp32 end
 to end, the reference's training dtype.  Validated against the
@@ -98,6 +99,12 @@ def maybe_make_step(ensemble, required: bool = False) -> Optional["HipSAEStep"]:
     if sig is FunctionalPositiveTiedSAE:
         ext = _ops.get_extension(required=True)
         return HipPositiveStep(ensemble, ext)
+
+    from sparse_coding_amd.models.lista import FunctionalLISTADenoisingSAE
+
+    if sig is FunctionalLISTADenoisingSAE:
+        ext = _ops.get_extension(required=True)
+        return HipLISTAStep(ensemble, ext)
 
     if sig is sigs.FunctionalMaskedTiedSAE:
         ext = _ops.get_extension(required=True)
@@ -595,6 +602,226 @@ class HipThresholdStep(HipSAEStep):
 
     def dp_grad_tensors(self):
         return [self.gw, self.g_gain, self.g_scale]
+
+
+class HipLISTAStep:
+    """Fused training step for FunctionalLISTADenoisingSAE (SURVEY.md K11;
+    reference residual_denoising_autoencoder.py:15-122).
+
+    The unrolled learned-ISTA encoder is GEMM-dominated; every GEMM here
+    runs on the hand-written MFMA kernels (enc/dec-shaped products reuse
+    k_enc_fwd mode 1 with the per-model x stride, the batch contractions
+    reuse k_grad_w), with the shrinkage/momentum elementwise glue in torch
+    ops on persistent workspaces.  The backward is derived analytically:
+
+      fwd  per layer: e = b - y A_hat;  r = y + e W_l^T;
+                      x = sign(r) relu(|r|-theta_l);  y' = x + m_l (x - x_prev)
+      bwd  per layer: g_x  = (1+m) g_y ;  g_x_prev -= m g_y
+                      g_r  = g_x * [|r| > theta];  g_theta = -sum_b g_r sign(r)
+                      g_m  = sum(g_y (x - x_prev)) * [0 < rho < 1]
+                      g_W  = g_r^T e;   g_e = g_r W_l
+                      g_y_prev = g_r - g_e A_hat^T   (+ momentum carry)
+                      g_Ahat  -= y_prev^T g_e
+      plus the decode/encode-init contributions g_Ahat += gscale c^T rr
+      + g_y0^T b, all pushed through the row-renorm projection by
+      k_project_adam.  Validated against the torch.func.grad oracle in
+      tests/test_hip_numerics.py::test_lista_step_matches_torch.
+    """
+
+    def __init__(self, ensemble, ext):
+        self.ens = ensemble
+        self.ext = ext
+        p = ensemble.params
+        self.n_models, self.n_dict, self.d_act = p["decoder"].shape
+        self.n_layers = len(p["encoder_layers"])
+
+        opt = ensemble.optimizer_kwargs
+        self.lr = float(opt.get("lr", 1e-3))
+        betas = opt.get("betas", (0.9, 0.999))
+        self.beta1, self.beta2 = float(betas[0]), float(betas[1])
+        self.eps = float(opt.get("eps", 1e-8))
+        name = getattr(ensemble.optimizer_func, "__name__", "adam")
+        if name != "adam":
+            raise RuntimeError(f"fused HIP step supports adam only, got {name}")
+        self.l1_alpha = ensemble.buffers["l1_alpha"].detach().reshape(self.n_models).contiguous()
+        self._B = None
+
+    def _alloc(self, B: int):
+        from sparse_coding_amd.ops.kconfig import kernel_config
+
+        self.kc = kernel_config()
+        M, n, d, L = self.n_models, self.n_dict, self.d_act, self.n_layers
+        dev = self.ens.params["decoder"].device
+        f = lambda *shape: torch.empty(shape, device=dev, dtype=torch.float32)
+        self.norms = f(M, n)
+        self.inv_norms = f(M, n)
+        self.ones_mn = torch.ones(M, n, device=dev)
+        self.zeros_bd = torch.zeros(B, d, device=dev)
+        self.scratch_lp = f(M, 2)
+        self.loss_parts = f(M, 2)
+        self.fired = torch.zeros(M, n, device=dev)
+        # forward saves
+        self.y = [f(M, B, n) for _ in range(L + 1)]
+        self.x = [f(M, B, n) for _ in range(L + 1)]
+        self.r = [f(M, B, n) for _ in range(L)]
+        self.neg_e = [f(M, B, d) for _ in range(L)]
+        self.s = f(M, B, n)       # e @ W_l^T scratch
+        self.rr = f(M, B, d)      # decode residual
+        # backward workspaces
+        self.g_y = f(M, B, n)
+        self.g_r = f(M, B, n)
+        self.g_e = f(M, B, d)
+        self.t_n = f(M, B, n)
+        self.gA = f(M, n, d)
+        self.gW = [f(M, n, d) for _ in range(L)]
+        self.g_theta = [f(M, n) for _ in range(L)]
+        self.g_rho = [f(M) for _ in range(L)]
+        self.zero_decay = torch.zeros(M, device=dev)
+        self._B = B
+
+    def grads_phase(self, x_in: torch.Tensor, on_grads=None):
+        ens, ext = self.ens, self.ext
+        M, n, d, L = self.n_models, self.n_dict, self.d_act, self.n_layers
+        B = x_in.shape[0]
+        if self._B != B:
+            self._alloc(B)
+        b = x_in.contiguous()
+        p = ens.params
+        A = p["decoder"]
+        layers = p["encoder_layers"]
+        bk, prio = self.kc["bk"], self.kc["prio"]
+        bk_gw = self.kc["bk_grad_w"] or bk
+
+        self.loss_parts.zero_()
+        ext.row_norms(A, self.norms, self.inv_norms, EPS_NORM)
+
+        # ---- forward ----
+        # y0 = b @ A_hat^T
+        ext.enc_fwd(b, A, self.ones_mn, self.inv_norms, self.y[0],
+                    self.scratch_lp, self.fired, 1, bk, prio)
+        self.x[0].copy_(self.y[0])
+        self.ms = [torch.clamp(layers[l]["rho"], 0.0, 1.0) for l in range(L)]
+        for l in range(L):
+            W_l, theta = layers[l]["W"], layers[l]["theta"]
+            # neg_e = y A_hat - b
+            ext.dec_fwd(self.y[l], A, self.inv_norms, b, self.neg_e[l],
+                        self.scratch_lp, self.kc["bk_dec"] or bk, prio)
+            # s = neg_e @ W_l^T ;  r = y - s
+            ext.enc_fwd(self.neg_e[l], W_l, self.ones_mn, None, self.s,
+                        self.scratch_lp, self.fired, 1, bk, prio)
+            torch.sub(self.y[l], self.s, out=self.r[l])
+            # x = shrink(r, theta);  y' = x + m (x - x_prev)
+            torch.sub(torch.abs(self.r[l]), theta.unsqueeze(1), out=self.x[l + 1])
+            self.x[l + 1].clamp_(min=0.0).mul_(torch.sign(self.r[l]))
+            m = self.ms[l].reshape(M, 1, 1)
+            torch.add(self.x[l + 1], (self.x[l + 1] - self.x[l]) * m, out=self.y[l + 1])
+
+        c = self.y[L]
+        # decode: rr = c A_hat - b (+ MSE partial)
+        ext.dec_fwd(c, A, self.inv_norms, b, self.rr, self.loss_parts,
+                    self.kc["bk_dec"] or bk, prio)
+        self._l1_sum = c.abs().sum(dim=(1, 2))
+
+        # ---- backward ----
+        gscale = 2.0 / (B * d)
+        # g_c = gscale * rr @ A_hat^T + l1/B * sign(c)
+        ext.enc_fwd(self.rr, A, self.ones_mn, self.inv_norms, self.g_y,
+                    self.scratch_lp, self.fired, 1, bk, prio)
+        self.g_y.mul_(gscale)
+        self.g_y.add_(torch.sign(c) * (self.l1_alpha / B).reshape(M, 1, 1))
+        # g_Ahat from decode path
+        ext.grad_w(c, self.rr, self.gA, gscale, 0.0, bk_gw, prio)
+
+        for l in range(L - 1, -1, -1):
+            W_l, theta = layers[l]["W"], layers[l]["theta"]
+            m = self.ms[l].reshape(M, 1, 1)
+            dx = self.x[l + 1] - self.x[l]
+            # rho grad (clamp passes only inside (0,1))
+            rho = layers[l]["rho"].reshape(M)
+            gate = ((rho > 0.0) & (rho < 1.0)).float()
+            torch.sum(self.g_y * dx, dim=(1, 2), out=self.g_rho[l])
+            self.g_rho[l].mul_(gate)
+            # g_x (into g_r buffer first), momentum carry to x_prev
+            torch.mul(self.g_y, 1.0 + m, out=self.g_r)
+            carry = self.g_y * (-m)  # g_x_prev contribution
+            # shrink backward
+            mask = (self.r[l].abs() > theta.unsqueeze(1)).float()
+            self.g_r.mul_(mask)
+            torch.sum(self.g_r * torch.sign(self.r[l]), dim=1, out=self.g_theta[l])
+            self.g_theta[l].neg_()
+            # g_W_l = g_r^T e = -(g_r^T neg_e)
+            ext.grad_w(self.g_r, self.neg_e[l], self.gW[l], -1.0, 0.0, bk_gw, prio)
+            if on_grads is not None:
+                on_grads([self.gW[l], self.g_theta[l], self.g_rho[l]])
+            # g_e = g_r @ W_l
+            ext.dec_fwd(self.g_r, W_l, self.ones_mn, self.zeros_bd, self.g_e,
+                        self.scratch_lp, self.kc["bk_dec"] or bk, prio)
+            # g_Ahat -= y_prev^T g_e
+            ext.grad_w(self.y[l], self.g_e, self.gA, -1.0, 1.0, bk_gw, prio)
+            # g_y_prev = g_r - g_e @ A_hat^T + carry (x0 = y0 folds the
+            # l = 0 momentum carry straight into g_y0)
+            ext.enc_fwd(self.g_e, A, self.ones_mn, self.inv_norms, self.t_n,
+                        self.scratch_lp, self.fired, 1, bk, prio)
+            torch.sub(self.g_r, self.t_n, out=self.g_y)
+            self.g_y.add_(carry)
+
+        # y0 = b @ A_hat^T: g_Ahat += g_y0^T b
+        ext.grad_w(self.g_y, b, self.gA, 1.0, 1.0, bk_gw, prio)
+        if on_grads is not None:
+            on_grads([self.gA])
+        return B
+
+    def update_phase(self, B: int):
+        ens, ext = self.ens, self.ext
+        st = ens.optim_states
+        st["step"] += 1.0
+        step_no = st["step"]
+        p = ens.params
+        ext.project_adam(p["decoder"], self.gA, self.norms,
+                         st["mu"]["decoder"], st["nu"]["decoder"], step_no,
+                         self.n_dict, self.lr, self.beta1, self.beta2,
+                         self.eps, EPS_NORM, True)
+        for l in range(self.n_layers):
+            mu = st["mu"]["encoder_layers"][l]["W"]
+            nu = st["nu"]["encoder_layers"][l]["W"]
+            ext.project_adam(p["encoder_layers"][l]["W"], self.gW[l], self.norms,
+                             mu, nu, step_no, self.n_dict,
+                             self.lr, self.beta1, self.beta2,
+                             self.eps, EPS_NORM, False)
+            ext.bias_adam(p["encoder_layers"][l]["theta"], self.g_theta[l],
+                          self.zero_decay,
+                          st["mu"]["encoder_layers"][l]["theta"],
+                          st["nu"]["encoder_layers"][l]["theta"],
+                          step_no, self.lr, self.beta1, self.beta2, self.eps)
+            self._apply_rho_adam(l, step_no)
+
+    def _apply_rho_adam(self, l, step_no):
+        st = self.ens.optim_states
+        rho = self.ens.params["encoder_layers"][l]["rho"]
+        mu = st["mu"]["encoder_layers"][l]["rho"]
+        nu = st["nu"]["encoder_layers"][l]["rho"]
+        g = self.g_rho[l].reshape(rho.shape)
+        t = step_no.reshape(rho.shape)
+        mu.mul_(self.beta1).add_(g, alpha=1 - self.beta1)
+        nu.mul_(self.beta2).addcmul_(g, g, value=1 - self.beta2)
+        bc1 = 1 - torch.pow(self.beta1, t)
+        bc2 = 1 - torch.pow(self.beta2, t)
+        rho.sub_(self.lr * (mu / bc1) / ((nu / bc2).sqrt() + self.eps))
+
+    def _loss_data(self, B: int):
+        mse = self.loss_parts[:, 0] / (B * self.d_act)
+        l1 = self.l1_alpha * self._l1_sum / B
+        return {"loss": mse + l1, "l_reconstruction": mse, "l_l1": l1}
+
+    def step(self, minibatches: torch.Tensor, expand_dims: bool = True):
+        if not expand_dims:
+            raise NotImplementedError("per-model batches not supported by the HIP step")
+        B = self.grads_phase(minibatches)
+        self.update_phase(B)
+        return self._loss_data(B), {"c": self.y[self.n_layers]}
+
+    def dp_grad_tensors(self):
+        return [self.gA] + self.gW + self.g_theta + self.g_rho
 
 
 class HipTopKStep:

@@ -433,6 +433,40 @@ def test_topk_step_matches_torch():
     assert _rel_err(ens_hip.params["dict"], ens_ref.params["dict"]) < 2e-3
 
 
+def test_lista_step_matches_torch():
+    """HipLISTAStep (hand-derived backward through the unrolled LISTA
+    layers) vs the torch.func.grad oracle: decoder, per-layer W/theta/rho
+    and the loss must all track over several steps."""
+    from sparse_coding_amd.engine.ensemble import FunctionalEnsemble
+    from sparse_coding_amd.functional.optim import adam
+    from sparse_coding_amd.models.lista import FunctionalLISTADenoisingSAE
+
+    torch.manual_seed(17)
+    M, B, d, n, L = 2, 256, 64, 128, 3
+    models = [FunctionalLISTADenoisingSAE.init(d, n, L, l1) for l1 in (1e-3, 3e-3)]
+    ens_hip = FunctionalEnsemble(models, FunctionalLISTADenoisingSAE, adam, {"lr": 1e-3},
+                                 device=DEV, backend="hip")
+    assert type(ens_hip._hip_step).__name__ == "HipLISTAStep"
+    models2 = [({k: (v.clone() if torch.is_tensor(v) else [{kk: vv.clone() for kk, vv in lay.items()} for lay in v])
+                 for k, v in p.items()},
+                {k: v.clone() for k, v in b.items()})
+               for p, b in ens_hip.unstack()]
+    ens_ref = FunctionalEnsemble(models2, FunctionalLISTADenoisingSAE, adam, {"lr": 1e-3},
+                                 device=DEV, backend="torch")
+    x = torch.randn(B, d, device=DEV)
+    for i in range(4):
+        l_hip, aux_hip = ens_hip.step_batch(x)
+        l_ref, aux_ref = ens_ref.step_batch(x)
+        assert _rel_err(l_hip["loss"], l_ref["loss"]) < 1e-4, i
+        assert _rel_err(aux_hip["c"], aux_ref["c"]) < 1e-3, i
+    assert _rel_err(ens_hip.params["decoder"], ens_ref.params["decoder"]) < 2e-3
+    for l in range(L):
+        for k in ("W", "theta", "rho"):
+            err = _rel_err(ens_hip.params["encoder_layers"][l][k],
+                           ens_ref.params["encoder_layers"][l][k])
+            assert err < 2e-3, (l, k, err)
+
+
 def test_resampler_gpu():
     from sparse_coding_amd.engine.ensemble import FunctionalEnsemble
     from sparse_coding_amd.engine.resample import EnsembleResampler
