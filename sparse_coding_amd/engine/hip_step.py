@@ -670,6 +670,7 @@ class HipLISTAStep:
         # backward workspaces
         self.g_y = f(M, B, n)
         self.g_r = f(M, B, n)
+        self.g_xc = torch.zeros(M, B, n, device=dev)  # momentum carry: g wrt x_{l-1}
         self.g_e = f(M, B, d)
         self.t_n = f(M, B, n)
         self.gA = f(M, n, d)
@@ -732,6 +733,10 @@ class HipLISTAStep:
         # g_Ahat from decode path
         ext.grad_w(c, self.rr, self.gA, gscale, 0.0, bk_gw, prio)
 
+        # two grad streams walk the layers together: g_y (through r/e) and
+        # g_xc, the momentum carry -m_{l+1} g_y_{l+1}, which reaches x_l and
+        # must flow through LAYER l's shrink, not through y_l
+        self.g_xc.zero_()
         for l in range(L - 1, -1, -1):
             W_l, theta = layers[l]["W"], layers[l]["theta"]
             m = self.ms[l].reshape(M, 1, 1)
@@ -741,9 +746,10 @@ class HipLISTAStep:
             gate = ((rho > 0.0) & (rho < 1.0)).float()
             torch.sum(self.g_y * dx, dim=(1, 2), out=self.g_rho[l])
             self.g_rho[l].mul_(gate)
-            # g_x (into g_r buffer first), momentum carry to x_prev
+            # g_x_l = (1+m) g_y_l + carry from layer l+1
             torch.mul(self.g_y, 1.0 + m, out=self.g_r)
-            carry = self.g_y * (-m)  # g_x_prev contribution
+            self.g_r.add_(self.g_xc)
+            torch.mul(self.g_y, -m, out=self.g_xc)  # carry for x_{l-1}
             # shrink backward
             mask = (self.r[l].abs() > theta.unsqueeze(1)).float()
             self.g_r.mul_(mask)
@@ -758,13 +764,13 @@ class HipLISTAStep:
                         self.scratch_lp, self.kc["bk_dec"] or bk, prio)
             # g_Ahat -= y_prev^T g_e
             ext.grad_w(self.y[l], self.g_e, self.gA, -1.0, 1.0, bk_gw, prio)
-            # g_y_prev = g_r - g_e @ A_hat^T + carry (x0 = y0 folds the
-            # l = 0 momentum carry straight into g_y0)
+            # g_y_prev = g_r - g_e @ A_hat^T
             ext.enc_fwd(self.g_e, A, self.ones_mn, self.inv_norms, self.t_n,
                         self.scratch_lp, self.fired, 1, bk, prio)
             torch.sub(self.g_r, self.t_n, out=self.g_y)
-            self.g_y.add_(carry)
 
+        # x0 = y0: the remaining carry lands on y0 directly
+        self.g_y.add_(self.g_xc)
         # y0 = b @ A_hat^T: g_Ahat += g_y0^T b
         ext.grad_w(self.g_y, b, self.gA, 1.0, 1.0, bk_gw, prio)
         if on_grads is not None:
