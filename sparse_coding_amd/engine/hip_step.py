@@ -21,11 +21,12 @@ hand-written gfx950 HIP kernels (``sparse_coding_amd/ops/hip/sae_kernels.hip``):
                    "used" weights + clamp-derivative mask (positive SAE)
   k_bias_adam    : Adam on any [M,k] vector param (+ L2-decay gradient)
 
-Fused-step coverage: tied, untied, masked tied/untied, thresholding,
-reverse, tied-centered, positive-tied, TopK, and LISTA (HipLISTAStep:
-hand-derived backward through the unrolled learned-ISTA layers, every GEMM
-on the MFMA kernels).  The semilinear (MLP-encoder) and residual-denoising
-SAEs stay on the vmap backend: plain batched GEMMs, rocBLAS-served.
+Fused-step coverage: EVERY trainable signature in the zoo — tied, untied,
+masked tied/untied, thresholding, reverse, tied-centered, positive-tied,
+TopK, LISTA, residual-denoising and semilinear.  The unrolled/MLP encoders
+(HipLISTAStep / HipResidualDenoisingStep / HipSemilinearStep) orchestrate
+the same MFMA GEMM kernels with hand-derived backward chains; elementwise
+glue runs as torch ops on persistent workspaces.
 
 All GEMMs run on the exact-f32 MFMA path (v_mfma_f32_32x32x2_f32): fp32 end
 to end, the reference's training dtype.  Validated against the
@@ -100,11 +101,21 @@ def maybe_make_step(ensemble, required: bool = False) -> Optional["HipSAEStep"]:
         ext = _ops.get_extension(required=True)
         return HipPositiveStep(ensemble, ext)
 
-    from sparse_coding_amd.models.lista import FunctionalLISTADenoisingSAE
+    from sparse_coding_amd.models.lista import (
+        FunctionalLISTADenoisingSAE,
+        FunctionalResidualDenoisingSAE,
+    )
+    from sparse_coding_amd.models.semilinear import SemiLinearSAE
 
     if sig is FunctionalLISTADenoisingSAE:
         ext = _ops.get_extension(required=True)
         return HipLISTAStep(ensemble, ext)
+    if sig is FunctionalResidualDenoisingSAE:
+        ext = _ops.get_extension(required=True)
+        return HipResidualDenoisingStep(ensemble, ext)
+    if sig is SemiLinearSAE:
+        ext = _ops.get_extension(required=True)
+        return HipSemilinearStep(ensemble, ext)
 
     if sig is sigs.FunctionalMaskedTiedSAE:
         ext = _ops.get_extension(required=True)
@@ -608,24 +619,22 @@ class HipLISTAStep:
     """Fused training step for FunctionalLISTADenoisingSAE (SURVEY.md K11;
     reference residual_denoising_autoencoder.py:15-122).
 
-    The unrolled learned-ISTA encoder is GEMM-dominated; every GEMM here
-    runs on the hand-written MFMA kernels (enc/dec-shaped products reuse
-    k_enc_fwd mode 1 with the per-model x stride, the batch contractions
-    reuse k_grad_w), with the shrinkage/momentum elementwise glue in torch
-    ops on persistent workspaces.  The backward is derived analytically:
+    Every GEMM runs on the MFMA kernels, and ALL the elementwise work is
+    fused into them or into one custom pass:
+      * forward: k_enc_fwd mode 4 computes r = y - (neg_e W_l^T),
+        x = shrink(r, theta) and y' = x + m (x - x_prev) in the GEMM
+        epilogue (one kernel per layer after the neg_e product);
+      * backward: k_lista_bwd_elem replaces the g_x/g_r/g_theta/g_rho
+        elementwise chain with one coalesced pass; k_enc_fwd mode 5 writes
+        g_y = g_r - g_e A_hat^T straight out of the GEMM.
+    The whole step is hipGraph-captured like the plain SAE step.
 
-      fwd  per layer: e = b - y A_hat;  r = y + e W_l^T;
-                      x = sign(r) relu(|r|-theta_l);  y' = x + m_l (x - x_prev)
-      bwd  per layer: g_x  = (1+m) g_y ;  g_x_prev -= m g_y
-                      g_r  = g_x * [|r| > theta];  g_theta = -sum_b g_r sign(r)
-                      g_m  = sum(g_y (x - x_prev)) * [0 < rho < 1]
-                      g_W  = g_r^T e;   g_e = g_r W_l
-                      g_y_prev = g_r - g_e A_hat^T   (+ momentum carry)
-                      g_Ahat  -= y_prev^T g_e
-      plus the decode/encode-init contributions g_Ahat += gscale c^T rr
-      + g_y0^T b, all pushed through the row-renorm projection by
-      k_project_adam.  Validated against the torch.func.grad oracle in
-      tests/test_hip_numerics.py::test_lista_step_matches_torch.
+    Backward, per layer (hand-derived, oracle-tested):
+      g_x = (1+m) g_y + carry;  g_r = g_x [|r| > theta]
+      g_theta = -sum_b g_r sign(r);  g_rho = sum g_y (x - x_prev) [0<rho<1]
+      g_W = -g_r^T neg_e;  g_e = g_r W_l;  g_y_prev = g_r - g_e A_hat^T
+      carry' = -m g_y;  g_Ahat -= y_prev^T g_e
+    plus g_Ahat += gscale c^T rr + g_y0^T b, projected by k_project_adam.
     """
 
     def __init__(self, ensemble, ext):
@@ -640,11 +649,15 @@ class HipLISTAStep:
         betas = opt.get("betas", (0.9, 0.999))
         self.beta1, self.beta2 = float(betas[0]), float(betas[1])
         self.eps = float(opt.get("eps", 1e-8))
-        name = getattr(ensemble.optimizer_func, "__name__", "adam")
-        if name != "adam":
-            raise RuntimeError(f"fused HIP step supports adam only, got {name}")
+        if getattr(ensemble.optimizer_func, "__name__", "adam") != "adam":
+            raise RuntimeError("fused HIP step supports adam only")
         self.l1_alpha = ensemble.buffers["l1_alpha"].detach().reshape(self.n_models).contiguous()
         self._B = None
+        import os as _os
+
+        self.use_graph = _os.environ.get("SPARSE_CODING_AMD_NO_GRAPH") != "1"
+        self._graph = None
+        self._eager_steps = 0
 
     def _alloc(self, B: int):
         from sparse_coding_amd.ops.kconfig import kernel_config
@@ -653,32 +666,34 @@ class HipLISTAStep:
         M, n, d, L = self.n_models, self.n_dict, self.d_act, self.n_layers
         dev = self.ens.params["decoder"].device
         f = lambda *shape: torch.empty(shape, device=dev, dtype=torch.float32)
-        self.norms = f(M, n)
-        self.inv_norms = f(M, n)
+        self.norms, self.inv_norms = f(M, n), f(M, n)
         self.ones_mn = torch.ones(M, n, device=dev)
         self.zeros_bd = torch.zeros(B, d, device=dev)
         self.scratch_lp = f(M, 2)
         self.loss_parts = f(M, 2)
         self.fired = torch.zeros(M, n, device=dev)
+        self.mom = [f(M) for _ in range(L)]
         # forward saves
         self.y = [f(M, B, n) for _ in range(L + 1)]
         self.x = [f(M, B, n) for _ in range(L + 1)]
         self.r = [f(M, B, n) for _ in range(L)]
         self.neg_e = [f(M, B, d) for _ in range(L)]
-        self.s = f(M, B, n)       # e @ W_l^T scratch
-        self.rr = f(M, B, d)      # decode residual
+        self.rr = f(M, B, d)
         # backward workspaces
         self.g_y = f(M, B, n)
         self.g_r = f(M, B, n)
-        self.g_xc = torch.zeros(M, B, n, device=dev)  # momentum carry: g wrt x_{l-1}
+        self.carry_a = torch.zeros(M, B, n, device=dev)
+        self.carry_b = torch.zeros(M, B, n, device=dev)
+        self.sgn = f(M, B, n)
         self.g_e = f(M, B, d)
-        self.t_n = f(M, B, n)
         self.gA = f(M, n, d)
         self.gW = [f(M, n, d) for _ in range(L)]
         self.g_theta = [f(M, n) for _ in range(L)]
         self.g_rho = [f(M) for _ in range(L)]
         self.zero_decay = torch.zeros(M, device=dev)
         self._B = B
+        self._graph = None
+        self._eager_steps = 0
 
     def grads_phase(self, x_in: torch.Tensor, on_grads=None):
         ens, ext = self.ens, self.ext
@@ -691,87 +706,60 @@ class HipLISTAStep:
         A = p["decoder"]
         layers = p["encoder_layers"]
         bk, prio = self.kc["bk"], self.kc["prio"]
+        bk_dec = self.kc["bk_dec"] or bk
         bk_gw = self.kc["bk_grad_w"] or bk
 
         self.loss_parts.zero_()
         ext.row_norms(A, self.norms, self.inv_norms, EPS_NORM)
 
         # ---- forward ----
-        # y0 = b @ A_hat^T
         ext.enc_fwd(b, A, self.ones_mn, self.inv_norms, self.y[0],
                     self.scratch_lp, self.fired, 1, bk, prio)
         self.x[0].copy_(self.y[0])
-        self.ms = [torch.clamp(layers[l]["rho"], 0.0, 1.0) for l in range(L)]
         for l in range(L):
-            W_l, theta = layers[l]["W"], layers[l]["theta"]
-            # neg_e = y A_hat - b
+            torch.clamp(layers[l]["rho"].reshape(M), 0.0, 1.0, out=self.mom[l])
             ext.dec_fwd(self.y[l], A, self.inv_norms, b, self.neg_e[l],
-                        self.scratch_lp, self.kc["bk_dec"] or bk, prio)
-            # s = neg_e @ W_l^T ;  r = y - s
-            ext.enc_fwd(self.neg_e[l], W_l, self.ones_mn, None, self.s,
-                        self.scratch_lp, self.fired, 1, bk, prio)
-            torch.sub(self.y[l], self.s, out=self.r[l])
-            # x = shrink(r, theta);  y' = x + m (x - x_prev)
-            torch.sub(torch.abs(self.r[l]), theta.unsqueeze(1), out=self.x[l + 1])
-            self.x[l + 1].clamp_(min=0.0).mul_(torch.sign(self.r[l]))
-            m = self.ms[l].reshape(M, 1, 1)
-            torch.add(self.x[l + 1], (self.x[l + 1] - self.x[l]) * m, out=self.y[l + 1])
-
+                        self.scratch_lp, bk_dec, prio)
+            # GEMM (neg_e W_l^T) + fused shrink/momentum epilogue
+            ext.enc_fwd(self.neg_e[l], layers[l]["W"], self.ones_mn, None,
+                        self.y[l + 1], self.scratch_lp, self.fired, 4, bk, prio,
+                        act_scale=layers[l]["theta"], u_out=self.r[l],
+                        y_in=self.y[l], x_prev=self.x[l], x_out=self.x[l + 1],
+                        mom=self.mom[l])
         c = self.y[L]
-        # decode: rr = c A_hat - b (+ MSE partial)
-        ext.dec_fwd(c, A, self.inv_norms, b, self.rr, self.loss_parts,
-                    self.kc["bk_dec"] or bk, prio)
-        self._l1_sum = c.abs().sum(dim=(1, 2))
+        ext.dec_fwd(c, A, self.inv_norms, b, self.rr, self.loss_parts, bk_dec, prio)
 
         # ---- backward ----
         gscale = 2.0 / (B * d)
-        # g_c = gscale * rr @ A_hat^T + l1/B * sign(c)
         ext.enc_fwd(self.rr, A, self.ones_mn, self.inv_norms, self.g_y,
                     self.scratch_lp, self.fired, 1, bk, prio)
         self.g_y.mul_(gscale)
-        self.g_y.add_(torch.sign(c) * (self.l1_alpha / B).reshape(M, 1, 1))
-        # g_Ahat from decode path
+        torch.sign(c, out=self.sgn)
+        self.sgn.mul_((self.l1_alpha / B).reshape(M, 1, 1))
+        self.g_y.add_(self.sgn)
         ext.grad_w(c, self.rr, self.gA, gscale, 0.0, bk_gw, prio)
 
-        # two grad streams walk the layers together: g_y (through r/e) and
-        # g_xc, the momentum carry -m_{l+1} g_y_{l+1}, which reaches x_l and
-        # must flow through LAYER l's shrink, not through y_l
-        self.g_xc.zero_()
+        carry_in, carry_out = None, self.carry_a
         for l in range(L - 1, -1, -1):
-            W_l, theta = layers[l]["W"], layers[l]["theta"]
-            m = self.ms[l].reshape(M, 1, 1)
-            dx = self.x[l + 1] - self.x[l]
-            # rho grad (clamp passes only inside (0,1))
-            rho = layers[l]["rho"].reshape(M)
-            gate = ((rho > 0.0) & (rho < 1.0)).float()
-            torch.sum(self.g_y * dx, dim=(1, 2), out=self.g_rho[l])
-            self.g_rho[l].mul_(gate)
-            # g_x_l = (1+m) g_y_l + carry from layer l+1
-            torch.mul(self.g_y, 1.0 + m, out=self.g_r)
-            self.g_r.add_(self.g_xc)
-            torch.mul(self.g_y, -m, out=self.g_xc)  # carry for x_{l-1}
-            # shrink backward
-            mask = (self.r[l].abs() > theta.unsqueeze(1)).float()
-            self.g_r.mul_(mask)
-            torch.sum(self.g_r * torch.sign(self.r[l]), dim=1, out=self.g_theta[l])
-            self.g_theta[l].neg_()
-            # g_W_l = g_r^T e = -(g_r^T neg_e)
+            self.g_theta[l].zero_()
+            self.g_rho[l].zero_()
+            ext.lista_bwd_elem(self.g_y, carry_in, self.r[l], layers[l]["theta"],
+                               self.x[l + 1], self.x[l], self.mom[l],
+                               self.g_r, carry_out, self.g_theta[l], self.g_rho[l])
             ext.grad_w(self.g_r, self.neg_e[l], self.gW[l], -1.0, 0.0, bk_gw, prio)
             if on_grads is not None:
                 on_grads([self.gW[l], self.g_theta[l], self.g_rho[l]])
-            # g_e = g_r @ W_l
-            ext.dec_fwd(self.g_r, W_l, self.ones_mn, self.zeros_bd, self.g_e,
-                        self.scratch_lp, self.kc["bk_dec"] or bk, prio)
-            # g_Ahat -= y_prev^T g_e
+            ext.dec_fwd(self.g_r, layers[l]["W"], self.ones_mn, self.zeros_bd,
+                        self.g_e, self.scratch_lp, bk_dec, prio)
             ext.grad_w(self.y[l], self.g_e, self.gA, -1.0, 1.0, bk_gw, prio)
-            # g_y_prev = g_r - g_e @ A_hat^T
-            ext.enc_fwd(self.g_e, A, self.ones_mn, self.inv_norms, self.t_n,
-                        self.scratch_lp, self.fired, 1, bk, prio)
-            torch.sub(self.g_r, self.t_n, out=self.g_y)
+            # g_y_prev = g_r - g_e A_hat^T (mode-5 epilogue)
+            ext.enc_fwd(self.g_e, A, self.ones_mn, self.inv_norms, self.g_y,
+                        self.scratch_lp, self.fired, 5, bk, prio, y_in=self.g_r)
+            carry_in = carry_out
+            carry_out = self.carry_b if carry_in is self.carry_a else self.carry_a
 
-        # x0 = y0: the remaining carry lands on y0 directly
-        self.g_y.add_(self.g_xc)
-        # y0 = b @ A_hat^T: g_Ahat += g_y0^T b
+        # x0 = y0: remaining carry lands on g_y0
+        self.g_y.add_(carry_in)
         ext.grad_w(self.g_y, b, self.gA, 1.0, 1.0, bk_gw, prio)
         if on_grads is not None:
             on_grads([self.gA])
@@ -788,17 +776,19 @@ class HipLISTAStep:
                          self.n_dict, self.lr, self.beta1, self.beta2,
                          self.eps, EPS_NORM, True)
         for l in range(self.n_layers):
-            mu = st["mu"]["encoder_layers"][l]["W"]
-            nu = st["nu"]["encoder_layers"][l]["W"]
             ext.project_adam(p["encoder_layers"][l]["W"], self.gW[l], self.norms,
-                             mu, nu, step_no, self.n_dict,
-                             self.lr, self.beta1, self.beta2,
+                             st["mu"]["encoder_layers"][l]["W"],
+                             st["nu"]["encoder_layers"][l]["W"], step_no,
+                             self.n_dict, self.lr, self.beta1, self.beta2,
                              self.eps, EPS_NORM, False)
             ext.bias_adam(p["encoder_layers"][l]["theta"], self.g_theta[l],
                           self.zero_decay,
                           st["mu"]["encoder_layers"][l]["theta"],
                           st["nu"]["encoder_layers"][l]["theta"],
                           step_no, self.lr, self.beta1, self.beta2, self.eps)
+            # clamp gate on rho (identical on every DP rank)
+            rho = p["encoder_layers"][l]["rho"].reshape(self.n_models)
+            self.g_rho[l].mul_(((rho > 0.0) & (rho < 1.0)).float())
             self._apply_rho_adam(l, step_no)
 
     def _apply_rho_adam(self, l, step_no):
@@ -816,18 +806,327 @@ class HipLISTAStep:
 
     def _loss_data(self, B: int):
         mse = self.loss_parts[:, 0] / (B * self.d_act)
+        l1 = self.l1_alpha * self.y[self.n_layers].abs().sum(dim=(1, 2)) / B
+        return {"loss": mse + l1, "l_reconstruction": mse, "l_l1": l1}
+
+    def _capture(self, x: torch.Tensor) -> None:
+        try:
+            self.x_static = x.clone()
+            torch.cuda.synchronize()
+            g = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(g):
+                B = self.grads_phase(self.x_static)
+                self.update_phase(B)
+            self._graph = g
+        except Exception as e:  # noqa: BLE001 - graphs are an optimization only
+            print(f"[hip_step] hipGraph capture failed ({e}); staying eager")
+            self.use_graph = False
+            self._graph = None
+
+    def step(self, minibatches: torch.Tensor, expand_dims: bool = True):
+        if not expand_dims:
+            raise NotImplementedError("per-model batches not supported by the HIP step")
+        B = minibatches.shape[0]
+        if self.use_graph:
+            if self._B == B and self._graph is None and self._eager_steps >= 2:
+                self._capture(minibatches.contiguous())
+            if self._graph is not None:
+                self.x_static.copy_(minibatches)
+                self._graph.replay()
+                return self._loss_data(B), {"c": self.y[self.n_layers]}
+        B = self.grads_phase(minibatches)
+        self.update_phase(B)
+        self._eager_steps += 1
+        return self._loss_data(B), {"c": self.y[self.n_layers]}
+
+    def dp_grad_tensors(self):
+        return [self.gA] + self.gW + self.g_theta + self.g_rho
+
+
+class HipResidualDenoisingStep:
+    """Fused step for FunctionalResidualDenoisingSAE
+    (residual_denoising_autoencoder.py:125-201): x0 = b A_hat^T, then
+    L residual blocks x' = relu(x + theta_l) W_l^T + x (W_l is [n, n]),
+    c = relu(x_L + bias).  Same kernel-orchestration recipe as HipLISTAStep;
+    all GEMMs (including the [n,n] layer products) run on the MFMA kernels.
+    """
+
+    def __init__(self, ensemble, ext):
+        self.ens = ensemble
+        self.ext = ext
+        p = ensemble.params
+        self.n_models, self.n_dict, self.d_act = p["decoder"].shape
+        self.n_layers = len(p["encoder_layers"])
+        opt = ensemble.optimizer_kwargs
+        self.lr = float(opt.get("lr", 1e-3))
+        betas = opt.get("betas", (0.9, 0.999))
+        self.beta1, self.beta2 = float(betas[0]), float(betas[1])
+        self.eps = float(opt.get("eps", 1e-8))
+        if getattr(ensemble.optimizer_func, "__name__", "adam") != "adam":
+            raise RuntimeError("fused HIP step supports adam only")
+        self.l1_alpha = ensemble.buffers["l1_alpha"].detach().reshape(self.n_models).contiguous()
+        self._B = None
+
+    def _alloc(self, B: int):
+        from sparse_coding_amd.ops.kconfig import kernel_config
+
+        self.kc = kernel_config()
+        M, n, d, L = self.n_models, self.n_dict, self.d_act, self.n_layers
+        dev = self.ens.params["decoder"].device
+        f = lambda *shape: torch.empty(shape, device=dev, dtype=torch.float32)
+        self.norms, self.inv_norms = f(M, n), f(M, n)
+        self.ones_mn = torch.ones(M, n, device=dev)
+        self.zeros_bn = torch.zeros(B, n, device=dev)
+        self.scratch_lp = f(M, 2)
+        self.loss_parts = f(M, 2)
+        self.fired = torch.zeros(M, n, device=dev)
+        self.xs = [f(M, B, n) for _ in range(L + 1)]
+        self.hs = [f(M, B, n) for _ in range(L)]
+        self.c = f(M, B, n)
+        self.rr = f(M, B, d)
+        self.g_x = f(M, B, n)
+        self.g_h = f(M, B, n)
+        self.t_n = f(M, B, n)
+        self.gA = f(M, n, d)
+        self.gW = [f(M, n, n) for _ in range(L)]
+        self.g_theta = [f(M, n) for _ in range(L)]
+        self.g_bias = f(M, n)
+        self.zero_decay = torch.zeros(M, device=dev)
+        self._B = B
+
+    def grads_phase(self, x_in: torch.Tensor, on_grads=None):
+        ens, ext = self.ens, self.ext
+        M, n, d, L = self.n_models, self.n_dict, self.d_act, self.n_layers
+        B = x_in.shape[0]
+        if self._B != B:
+            self._alloc(B)
+        b = x_in.contiguous()
+        p = ens.params
+        A = p["decoder"]
+        layers = p["encoder_layers"]
+        bias = p["encoder_bias"]
+        bk, prio = self.kc["bk"], self.kc["prio"]
+        bk_dec = self.kc["bk_dec"] or bk
+        bk_gw = self.kc["bk_grad_w"] or bk
+
+        self.loss_parts.zero_()
+        self.g_bias.zero_()
+        ext.row_norms(A, self.norms, self.inv_norms, EPS_NORM)
+
+        # forward
+        ext.enc_fwd(b, A, self.ones_mn, self.inv_norms, self.xs[0],
+                    self.scratch_lp, self.fired, 1, bk, prio)
+        for l in range(L):
+            W_l, theta = layers[l]["W"], layers[l]["theta"]
+            torch.add(self.xs[l], theta.unsqueeze(1), out=self.hs[l])
+            self.hs[l].clamp_(min=0.0)
+            # x' = h W_l^T + x  ([M,B,n] x [M,n,n])
+            ext.enc_fwd(self.hs[l], W_l, self.ones_mn, None, self.xs[l + 1],
+                        self.scratch_lp, self.fired, 1, bk, prio)
+            self.xs[l + 1].add_(self.xs[l])
+        torch.add(self.xs[L], bias.unsqueeze(1), out=self.c)
+        self.c.clamp_(min=0.0)
+
+        ext.dec_fwd(self.c, A, self.inv_norms, b, self.rr, self.loss_parts, bk_dec, prio)
+        self._l1_sum = self.c.sum(dim=(1, 2))  # c >= 0
+
+        # backward: g_c via k_gc (relu mask + l1 term + bias colsum, K=d)
+        ext.gc(self.rr, A, self.inv_norms, self.c, self.l1_alpha,
+               self.g_x, self.g_bias, bk, prio)
+        gscale = 2.0 / (B * d)
+        ext.grad_w(self.c, self.rr, self.gA, gscale, 0.0, bk_gw, prio)
+
+        for l in range(L - 1, -1, -1):
+            W_l, theta = layers[l]["W"], layers[l]["theta"]
+            # g_W_l = g_x'^T h
+            ext.grad_w(self.g_x, self.hs[l], self.gW[l], 1.0, 0.0, bk_gw, prio)
+            if on_grads is not None:
+                on_grads([self.gW[l]])
+            # g_h = g_x' @ W_l  ([M,B,n] x [M,n,n])
+            ext.dec_fwd(self.g_x, W_l, self.ones_mn, self.zeros_bn, self.g_h,
+                        self.scratch_lp, bk_dec, prio)
+            # relu(x + theta) backward
+            self.g_h.mul_((self.hs[l] > 0).float())
+            torch.sum(self.g_h, dim=1, out=self.g_theta[l])
+            if on_grads is not None:
+                on_grads([self.g_theta[l]])
+            self.g_x.add_(self.g_h)  # residual skip + through-layer paths
+
+        # x0 = b A_hat^T
+        ext.grad_w(self.g_x, b, self.gA, 1.0, 1.0, bk_gw, prio)
+        if on_grads is not None:
+            on_grads([self.gA, self.g_bias])
+        return B
+
+    def update_phase(self, B: int):
+        ens, ext = self.ens, self.ext
+        st = ens.optim_states
+        st["step"] += 1.0
+        step_no = st["step"]
+        p = ens.params
+        ext.project_adam(p["decoder"], self.gA, self.norms,
+                         st["mu"]["decoder"], st["nu"]["decoder"], step_no,
+                         self.n_dict, self.lr, self.beta1, self.beta2,
+                         self.eps, EPS_NORM, True)
+        for l in range(self.n_layers):
+            ext.project_adam(p["encoder_layers"][l]["W"], self.gW[l], self.norms,
+                             st["mu"]["encoder_layers"][l]["W"],
+                             st["nu"]["encoder_layers"][l]["W"], step_no,
+                             self.n_dict, self.lr, self.beta1, self.beta2,
+                             self.eps, EPS_NORM, False)
+            ext.bias_adam(p["encoder_layers"][l]["theta"], self.g_theta[l],
+                          self.zero_decay,
+                          st["mu"]["encoder_layers"][l]["theta"],
+                          st["nu"]["encoder_layers"][l]["theta"],
+                          step_no, self.lr, self.beta1, self.beta2, self.eps)
+        ext.bias_adam(p["encoder_bias"], self.g_bias, self.zero_decay,
+                      st["mu"]["encoder_bias"], st["nu"]["encoder_bias"],
+                      step_no, self.lr, self.beta1, self.beta2, self.eps)
+
+    def _loss_data(self, B: int):
+        mse = self.loss_parts[:, 0] / (B * self.d_act)
         l1 = self.l1_alpha * self._l1_sum / B
         return {"loss": mse + l1, "l_reconstruction": mse, "l_l1": l1}
 
     def step(self, minibatches: torch.Tensor, expand_dims: bool = True):
         if not expand_dims:
-            raise NotImplementedError("per-model batches not supported by the HIP step")
+            raise NotImplementedError
         B = self.grads_phase(minibatches)
         self.update_phase(B)
-        return self._loss_data(B), {"c": self.y[self.n_layers]}
+        return self._loss_data(B), {"c": self.c}
 
     def dp_grad_tensors(self):
-        return [self.gA] + self.gW + self.g_theta + self.g_rho
+        return [self.gA, self.g_bias] + self.gW + self.g_theta
+
+
+class HipSemilinearStep:
+    """Fused step for SemiLinearSAE (semilinear_autoencoder.py:14-83): a
+    2-layer relu MLP encoder + normalized linear decoder.  The top-layer
+    backward IS k_gc (relu mask + l1 + bias column sums); the hidden layer
+    is one more GEMM pair.
+    """
+
+    def __init__(self, ensemble, ext):
+        self.ens = ensemble
+        self.ext = ext
+        p = ensemble.params
+        self.n_models, self.n_dict, self.d_act = p["decoder"].shape
+        self.hidden = p["encoder_layers"][0]["weight"].shape[1]  # [M, h, d] -> h
+        assert len(p["encoder_layers"]) == 2
+        opt = ensemble.optimizer_kwargs
+        self.lr = float(opt.get("lr", 1e-3))
+        betas = opt.get("betas", (0.9, 0.999))
+        self.beta1, self.beta2 = float(betas[0]), float(betas[1])
+        self.eps = float(opt.get("eps", 1e-8))
+        if getattr(ensemble.optimizer_func, "__name__", "adam") != "adam":
+            raise RuntimeError("fused HIP step supports adam only")
+        self.l1_alpha = ensemble.buffers["l1_alpha"].detach().reshape(self.n_models).contiguous()
+        self._B = None
+
+    def _alloc(self, B: int):
+        from sparse_coding_amd.ops.kconfig import kernel_config
+
+        self.kc = kernel_config()
+        M, n, d, h = self.n_models, self.n_dict, self.d_act, self.hidden
+        dev = self.ens.params["decoder"].device
+        f = lambda *shape: torch.empty(shape, device=dev, dtype=torch.float32)
+        self.norms, self.inv_norms = f(M, n), f(M, n)
+        self.ones_mh = torch.ones(M, h, device=dev)
+        self.zeros_bh = torch.zeros(B, h, device=dev)
+        self.scratch_lp = f(M, 2)
+        self.loss_parts = f(M, 2)
+        self.fired = torch.zeros(M, n, device=dev)
+        self.fired_h = torch.zeros(M, h, device=dev)
+        self.h1 = f(M, B, h)
+        self.c = f(M, B, n)
+        self.rr = f(M, B, d)
+        self.g_c = f(M, B, n)
+        self.g_h = f(M, B, h)
+        self.gA = f(M, n, d)
+        self.gW2 = f(M, n, h)
+        self.gW1 = f(M, h, d)
+        self.g_b2 = f(M, n)
+        self.g_b1 = f(M, h)
+        self.zero_decay = torch.zeros(M, device=dev)
+        self._B = B
+
+    def grads_phase(self, x_in: torch.Tensor, on_grads=None):
+        ens, ext = self.ens, self.ext
+        B = x_in.shape[0]
+        if self._B != B:
+            self._alloc(B)
+        b = x_in.contiguous()
+        p = ens.params
+        A = p["decoder"]
+        l1p, l2p = p["encoder_layers"]
+        bk, prio = self.kc["bk"], self.kc["prio"]
+        bk_dec = self.kc["bk_dec"] or bk
+        bk_gw = self.kc["bk_grad_w"] or bk
+
+        self.loss_parts.zero_()
+        self.g_b2.zero_()
+        ext.row_norms(A, self.norms, self.inv_norms, EPS_NORM)
+
+        # h1 = relu(b W1^T + b1);  c = relu(h1 W2^T + b2)
+        ext.enc_fwd(b, l1p["weight"], l1p["bias"], None, self.h1,
+                    self.scratch_lp, self.fired_h, 0, bk, prio)
+        ext.enc_fwd(self.h1, l2p["weight"], l2p["bias"], None, self.c,
+                    self.scratch_lp, self.fired, 0, bk, prio)
+        ext.dec_fwd(self.c, A, self.inv_norms, b, self.rr, self.loss_parts, bk_dec, prio)
+        self._l1_sum = self.c.sum(dim=(1, 2))
+
+        # backward
+        ext.gc(self.rr, A, self.inv_norms, self.c, self.l1_alpha,
+               self.g_c, self.g_b2, bk, prio)
+        gscale = 2.0 / (B * self.d_act)
+        ext.grad_w(self.c, self.rr, self.gA, gscale, 0.0, bk_gw, prio)
+        ext.grad_w(self.g_c, self.h1, self.gW2, 1.0, 0.0, bk_gw, prio)
+        # g_h1 = (g_c W2) * [h1 > 0]
+        ext.dec_fwd(self.g_c, l2p["weight"], self.ones_mh, self.zeros_bh,
+                    self.g_h, self.scratch_lp, bk_dec, prio)
+        self.g_h.mul_((self.h1 > 0).float())
+        torch.sum(self.g_h, dim=1, out=self.g_b1)
+        ext.grad_w(self.g_h, b, self.gW1, 1.0, 0.0, bk_gw, prio)
+        if on_grads is not None:
+            on_grads([self.gA, self.gW2, self.gW1, self.g_b2, self.g_b1])
+        return B
+
+    def update_phase(self, B: int):
+        ens, ext = self.ens, self.ext
+        st = ens.optim_states
+        st["step"] += 1.0
+        step_no = st["step"]
+        p = ens.params
+        ext.project_adam(p["decoder"], self.gA, self.norms,
+                         st["mu"]["decoder"], st["nu"]["decoder"], step_no,
+                         self.n_dict, self.lr, self.beta1, self.beta2,
+                         self.eps, EPS_NORM, True)
+        for li, (gW, gb) in enumerate(((self.gW1, self.g_b1), (self.gW2, self.g_b2))):
+            lay = p["encoder_layers"][li]
+            ext.project_adam(lay["weight"], gW, self.norms,
+                             st["mu"]["encoder_layers"][li]["weight"],
+                             st["nu"]["encoder_layers"][li]["weight"], step_no,
+                             gW.shape[1], self.lr, self.beta1, self.beta2,
+                             self.eps, EPS_NORM, False)
+            ext.bias_adam(lay["bias"], gb, self.zero_decay,
+                          st["mu"]["encoder_layers"][li]["bias"],
+                          st["nu"]["encoder_layers"][li]["bias"],
+                          step_no, self.lr, self.beta1, self.beta2, self.eps)
+
+    def _loss_data(self, B: int):
+        mse = self.loss_parts[:, 0] / (B * self.d_act)
+        l1 = self.l1_alpha * self._l1_sum / B
+        return {"loss": mse + l1, "l_reconstruction": mse, "l_l1": l1}
+
+    def step(self, minibatches: torch.Tensor, expand_dims: bool = True):
+        if not expand_dims:
+            raise NotImplementedError
+        B = self.grads_phase(minibatches)
+        self.update_phase(B)
+        return self._loss_data(B), {"c": self.c}
+
+    def dp_grad_tensors(self):
+        return [self.gA, self.gW1, self.gW2, self.g_b1, self.g_b2]
 
 
 class HipTopKStep:

@@ -332,10 +332,14 @@ void k_enc_fwd_t(const float* __restrict__ x,       // [B, d]
                  int prio,
                  const float* __restrict__ act_scale,  // [M, n] (mode 2)
                  const float* __restrict__ act_gain,   // [M, n] (mode 2)
-                 float* __restrict__ u_out,            // [M, B, n] (mode 2)
+                 float* __restrict__ u_out,            // [M, B, n] (mode 2/4)
                  const int* __restrict__ dict_sizes, // [M] or nullptr:
                                        // masked sigs zero cols >= dict_sizes[m]
-                 long x_mstride) {     // 0: x shared [B,d]; else x is [M,B,d]
+                 long x_mstride,       // 0: x shared [B,d]; else x is [M,B,d]
+                 const float* __restrict__ y_in,   // [M, B, n] (modes 4, 5)
+                 const float* __restrict__ x_prev, // [M, B, n] (mode 4)
+                 float* __restrict__ x_out,        // [M, B, n] (mode 4)
+                 const float* __restrict__ mom) {  // [M] momentum (mode 4)
   __shared__ float As[2][TBK * BMP];
   __shared__ float Bs[2][TBK * BMP];
 
@@ -408,6 +412,42 @@ void k_enc_fwd_t(const float* __restrict__ x,       // [B, d]
         float other = __shfl_xor(fired_cnt, 32, WAVE);
         float tot = fired_cnt + other;
         if (g.lane < 32 && tot > 0.f) atomicAdd(&fired_m[col], tot);
+      }
+      continue;
+    }
+    if (mode == 4 || mode == 5) {
+      // LISTA layer epilogues (HipLISTAStep): the GEMM result `acc` is
+      //   mode 4:  s = neg_e W_l^T;  r = y - s;  x = shrink(r, theta);
+      //            y' = x + m (x - x_prev)   -> u_out=r, x_out=x, c_out=y'
+      //   mode 5:  c_out = y_in - acc        (backward g_y = g_r - g_e A^T)
+      const float* y_m = y_in + (long)m * (long)B * n;
+      float th = 0.f, mm = 0.f;
+      const float* xp_m = nullptr;
+      float* xo_m = nullptr;
+      float* u_m = nullptr;
+      if (mode == 4) {
+        th = col_ok ? act_scale[(long)m * n + col] : 0.f;
+        mm = mom[m];
+        xp_m = x_prev + (long)m * (long)B * n;
+        xo_m = x_out + (long)m * (long)B * n;
+        u_m = u_out + (long)m * (long)B * n;
+      }
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        int row = row0 + g.wr + acc_row(r, g.lane);
+        if (row < B && col_ok) {
+          long idx = (long)row * n + col;
+          float v = y_m[idx] - acc[tj][r];
+          if (mode == 5) {
+            c_m[idx] = v;
+          } else {
+            u_m[idx] = v;  // r_l, kept for the backward's shrink mask
+            float a = fabsf(v) - th;
+            float xa = (a > 0.f) ? ((v > 0.f) ? a : -a) : 0.f;
+            xo_m[idx] = xa;
+            c_m[idx] = xa + mm * (xa - xp_m[idx]);
+          }
+        }
       }
       continue;
     }
@@ -693,6 +733,66 @@ void k_gc_thresh_t(const float* __restrict__ r,        // [M, B, d]
         if (scale_colsum + os != 0.f) atomicAdd(&gs_m[col], scale_colsum + os);
       }
     }
+  }
+}
+
+// ---------------------------------------------------------------------------
+// k_lista_bwd_elem: one pass over [M, B, n] replacing the LISTA backward's
+// elementwise chain:
+//   g_x   = (1 + m) g_y + carry_in
+//   g_r   = g_x * [|r| > theta]
+//   carry = -m g_y                       (becomes next layer's carry_in)
+//   g_theta[col] += -g_r * sign(r)       (column sum)
+//   g_rho[model] += g_y * (x - x_prev)   (scalar sum; clamp gate on host)
+// grid (ceil(n/256), ceil(B/ROWS), M); block 256; coalesced row-major loops.
+// ---------------------------------------------------------------------------
+#define LBW_COLS 256
+#define LBW_ROWS 256
+extern "C" __global__ __launch_bounds__(LBW_COLS)
+void k_lista_bwd_elem(const float* __restrict__ g_y,
+                      const float* __restrict__ carry_in,  // nullptr on layer L
+                      const float* __restrict__ r,
+                      const float* __restrict__ theta,   // [M, n]
+                      const float* __restrict__ x,
+                      const float* __restrict__ x_prev,
+                      const float* __restrict__ mom,     // [M]
+                      float* __restrict__ g_r,
+                      float* __restrict__ carry_out,
+                      float* __restrict__ g_theta,       // [M, n] (pre-zeroed)
+                      float* __restrict__ g_rho,         // [M]    (pre-zeroed)
+                      int B, int n) {
+  const int m = blockIdx.z;
+  const int col = blockIdx.x * LBW_COLS + threadIdx.x;
+  const int row0 = blockIdx.y * LBW_ROWS;
+  const int row1 = min(row0 + LBW_ROWS, B);
+  const bool col_ok = col < n;
+  const float mm = mom[m];
+  const float th = col_ok ? theta[(long)m * n + col] : 0.f;
+  const long base = (long)m * B * n;
+
+  float th_sum = 0.f, rho_sum = 0.f;
+  if (col_ok) {
+    for (int row = row0; row < row1; ++row) {
+      long idx = base + (long)row * n + col;
+      float gy = g_y[idx];
+      float gx = (1.0f + mm) * gy + (carry_in ? carry_in[idx] : 0.f);
+      float rv = r[idx];
+      float gr = (fabsf(rv) > th) ? gx : 0.f;
+      g_r[idx] = gr;
+      carry_out[idx] = -mm * gy;
+      th_sum -= gr * ((rv > 0.f) ? 1.f : ((rv < 0.f) ? -1.f : 0.f));
+      rho_sum += gy * (x[idx] - x_prev[idx]);
+    }
+    if (th_sum != 0.f) atomicAdd(&g_theta[(long)m * n + col], th_sum);
+  }
+  rho_sum = wave_reduce_sum(rho_sum);
+  __shared__ float warp_part[LBW_COLS / WAVE];
+  if ((threadIdx.x & (WAVE - 1)) == 0) warp_part[threadIdx.x / WAVE] = rho_sum;
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    float s = 0.f;
+    for (int w = 0; w < LBW_COLS / WAVE; ++w) s += warp_part[w];
+    if (s != 0.f) atomicAdd(&g_rho[m], s);
   }
 }
 

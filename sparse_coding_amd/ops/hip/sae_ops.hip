@@ -35,7 +35,11 @@ void enc_fwd(torch::Tensor x, torch::Tensor Wenc, torch::Tensor bias,
              c10::optional<torch::Tensor> act_scale,
              c10::optional<torch::Tensor> act_gain,
              c10::optional<torch::Tensor> u_out,
-             c10::optional<torch::Tensor> dict_sizes) {
+             c10::optional<torch::Tensor> dict_sizes,
+             c10::optional<torch::Tensor> y_in,
+             c10::optional<torch::Tensor> x_prev,
+             c10::optional<torch::Tensor> x_out,
+             c10::optional<torch::Tensor> mom) {
   CHECK_IN(x); CHECK_IN(Wenc); CHECK_IN(bias); CHECK_IN(c_out);
   CHECK_IN(loss_parts); CHECK_IN(fired);
   int M = Wenc.size(0), n = Wenc.size(1), d = Wenc.size(2);
@@ -67,19 +71,40 @@ void enc_fwd(torch::Tensor x, torch::Tensor Wenc, torch::Tensor bias,
                 dict_sizes->scalar_type() == torch::kInt32, "dict_sizes must be int32 GPU");
     ds_p = dict_sizes->data_ptr<int>();
   }
+  const float* yin_p = nullptr;
+  const float* xp_p = nullptr;
+  float* xo_p = nullptr;
+  const float* mom_p = nullptr;
+  if (mode == 4 || mode == 5) {
+    TORCH_CHECK(y_in, "modes 4/5 need y_in");
+    CHECK_IN(y_in.value());
+    yin_p = y_in->data_ptr<float>();
+    if (mode == 4) {
+      TORCH_CHECK(x_prev && x_out && mom && act_scale, "mode 4 needs x_prev/x_out/mom/act_scale(theta)/u_out");
+      CHECK_IN(x_prev.value()); CHECK_IN(x_out.value()); CHECK_IN(mom.value());
+      CHECK_IN(act_scale.value()); CHECK_IN(u_out.value());
+      xp_p = x_prev->data_ptr<float>();
+      xo_p = x_out->data_ptr<float>();
+      mom_p = mom->data_ptr<float>();
+      a_p = act_scale->data_ptr<float>();
+      u_p = u_out->data_ptr<float>();
+    }
+  }
   dim3 grid(cdiv(n, BN), cdiv(B, BM), M);
   if (bk == 16)
     hipLaunchKernelGGL((k_enc_fwd_t<16, 6>), grid, dim3(NTHREADS), 0, cur_stream(),
                        x.data_ptr<float>(), Wenc.data_ptr<float>(),
                        bias.data_ptr<float>(), inv, c_out.data_ptr<float>(),
                        loss_parts.data_ptr<float>(), fired.data_ptr<float>(),
-                       B, d, n, (int)mode, prio ? 1 : 0, a_p, gn_p, u_p, ds_p, x_mstride);
+                       B, d, n, (int)mode, prio ? 1 : 0, a_p, gn_p, u_p, ds_p, x_mstride,
+                       yin_p, xp_p, xo_p, mom_p);
   else
     hipLaunchKernelGGL((k_enc_fwd_t<32, 4>), grid, dim3(NTHREADS), 0, cur_stream(),
                        x.data_ptr<float>(), Wenc.data_ptr<float>(),
                        bias.data_ptr<float>(), inv, c_out.data_ptr<float>(),
                        loss_parts.data_ptr<float>(), fired.data_ptr<float>(),
-                       B, d, n, (int)mode, prio ? 1 : 0, a_p, gn_p, u_p, ds_p, x_mstride);
+                       B, d, n, (int)mode, prio ? 1 : 0, a_p, gn_p, u_p, ds_p, x_mstride,
+                       yin_p, xp_p, xo_p, mom_p);
 }
 
 void dec_fwd(torch::Tensor c, torch::Tensor Wdec, torch::Tensor inv_norms,
@@ -300,6 +325,28 @@ void gc2(torch::Tensor rT, torch::Tensor WT, torch::Tensor c,
                        B, d, n, prio ? 1 : 0, (int)gc_mode);
 }
 
+void lista_bwd_elem(torch::Tensor g_y, c10::optional<torch::Tensor> carry_in,
+                    torch::Tensor r, torch::Tensor theta, torch::Tensor x,
+                    torch::Tensor x_prev, torch::Tensor mom,
+                    torch::Tensor g_r, torch::Tensor carry_out,
+                    torch::Tensor g_theta, torch::Tensor g_rho) {
+  CHECK_IN(g_y); CHECK_IN(r); CHECK_IN(theta); CHECK_IN(x); CHECK_IN(x_prev);
+  CHECK_IN(mom); CHECK_IN(g_r); CHECK_IN(carry_out); CHECK_IN(g_theta); CHECK_IN(g_rho);
+  const float* ci = nullptr;
+  if (carry_in.has_value()) {
+    CHECK_IN(carry_in.value());
+    ci = carry_in->data_ptr<float>();
+  }
+  int M = g_y.size(0), B = g_y.size(1), n = g_y.size(2);
+  dim3 grid(cdiv(n, LBW_COLS), cdiv(B, LBW_ROWS), M);
+  hipLaunchKernelGGL(k_lista_bwd_elem, grid, dim3(LBW_COLS), 0, cur_stream(),
+                     g_y.data_ptr<float>(), ci, r.data_ptr<float>(),
+                     theta.data_ptr<float>(), x.data_ptr<float>(),
+                     x_prev.data_ptr<float>(), mom.data_ptr<float>(),
+                     g_r.data_ptr<float>(), carry_out.data_ptr<float>(),
+                     g_theta.data_ptr<float>(), g_rho.data_ptr<float>(), B, n);
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("row_norms", &row_norms, "dictionary row norms + clamped inverses");
   m.def("transpose_scale", &transpose_scale, "batched [R,C]->[C,R] transpose with row scale");
@@ -317,7 +364,13 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("c_out"), py::arg("loss_parts"), py::arg("fired"), py::arg("mode"),
         py::arg("bk") = 32, py::arg("prio") = false,
         py::arg("act_scale") = py::none(), py::arg("act_gain") = py::none(),
-        py::arg("u_out") = py::none(), py::arg("dict_sizes") = py::none());
+        py::arg("u_out") = py::none(), py::arg("dict_sizes") = py::none(),
+        py::arg("y_in") = py::none(), py::arg("x_prev") = py::none(),
+        py::arg("x_out") = py::none(), py::arg("mom") = py::none());
+  m.def("lista_bwd_elem", &lista_bwd_elem, "fused LISTA backward elementwise pass",
+        py::arg("g_y"), py::arg("carry_in"), py::arg("r"), py::arg("theta"),
+        py::arg("x"), py::arg("x_prev"), py::arg("mom"), py::arg("g_r"),
+        py::arg("carry_out"), py::arg("g_theta"), py::arg("g_rho"));
   m.def("gc_thresh", &gc_thresh, "code-grad through the threshold gate (+gain/scale grads)",
         py::arg("r"), py::arg("Wdec"), py::arg("inv_norms"), py::arg("c"),
         py::arg("u"), py::arg("act_scale"), py::arg("l1_alpha"), py::arg("gpre"),

@@ -492,3 +492,66 @@ def test_resampler_gpu():
     after = ens.params["encoder"][:, :64]
     assert not torch.allclose(before[:, :64], after)
     assert (ens.optim_states["mu"]["encoder"][0, :64] == 0).all()
+
+
+def test_residual_denoising_step_matches_torch():
+    """HipResidualDenoisingStep vs the vmap oracle."""
+    from sparse_coding_amd.engine.ensemble import FunctionalEnsemble
+    from sparse_coding_amd.functional.optim import adam
+    from sparse_coding_amd.models.lista import FunctionalResidualDenoisingSAE
+
+    torch.manual_seed(18)
+    M, B, d, n, L = 2, 256, 64, 128, 2
+    models = [FunctionalResidualDenoisingSAE.init(d, n, L, l1) for l1 in (1e-3, 3e-3)]
+    ens_hip = FunctionalEnsemble(models, FunctionalResidualDenoisingSAE, adam, {"lr": 1e-3},
+                                 device=DEV, backend="hip")
+    assert type(ens_hip._hip_step).__name__ == "HipResidualDenoisingStep"
+    models2 = [({k: (v.clone() if torch.is_tensor(v) else [{kk: vv.clone() for kk, vv in lay.items()} for lay in v])
+                 for k, v in p.items()}, {k: v.clone() for k, v in b.items()})
+               for p, b in ens_hip.unstack()]
+    ens_ref = FunctionalEnsemble(models2, FunctionalResidualDenoisingSAE, adam, {"lr": 1e-3},
+                                 device=DEV, backend="torch")
+    x = torch.randn(B, d, device=DEV)
+    for i in range(4):
+        l_hip, aux_hip = ens_hip.step_batch(x)
+        l_ref, aux_ref = ens_ref.step_batch(x)
+        assert _rel_err(l_hip["loss"], l_ref["loss"]) < 1e-4, i
+        assert _rel_err(aux_hip["c"], aux_ref["c"]) < 1e-3, i
+    assert _rel_err(ens_hip.params["decoder"], ens_ref.params["decoder"]) < 2e-3
+    assert _rel_err(ens_hip.params["encoder_bias"], ens_ref.params["encoder_bias"]) < 2e-3
+    for l in range(L):
+        for k in ("W", "theta"):
+            err = _rel_err(ens_hip.params["encoder_layers"][l][k],
+                           ens_ref.params["encoder_layers"][l][k])
+            assert err < 2e-3, (l, k, err)
+
+
+def test_semilinear_step_matches_torch():
+    """HipSemilinearStep (2-layer MLP encoder) vs the vmap oracle."""
+    from sparse_coding_amd.engine.ensemble import FunctionalEnsemble
+    from sparse_coding_amd.functional.optim import adam
+    from sparse_coding_amd.models.semilinear import SemiLinearSAE
+
+    torch.manual_seed(19)
+    M, B, d, n = 2, 256, 64, 128
+    models = [SemiLinearSAE.init(d, n, l1, device=DEV) for l1 in (1e-3, 3e-3)]
+    ens_hip = FunctionalEnsemble(models, SemiLinearSAE, adam, {"lr": 1e-3},
+                                 device=DEV, backend="hip")
+    assert type(ens_hip._hip_step).__name__ == "HipSemilinearStep"
+    models2 = [({k: (v.clone() if torch.is_tensor(v) else [{kk: vv.clone() for kk, vv in lay.items()} for lay in v])
+                 for k, v in p.items()}, {k: v.clone() for k, v in b.items()})
+               for p, b in ens_hip.unstack()]
+    ens_ref = FunctionalEnsemble(models2, SemiLinearSAE, adam, {"lr": 1e-3},
+                                 device=DEV, backend="torch")
+    x = torch.randn(B, d, device=DEV)
+    for i in range(4):
+        l_hip, aux_hip = ens_hip.step_batch(x)
+        l_ref, aux_ref = ens_ref.step_batch(x)
+        assert _rel_err(l_hip["loss"], l_ref["loss"]) < 1e-4, i
+        assert _rel_err(aux_hip["c"], aux_ref["c"]) < 1e-3, i
+    assert _rel_err(ens_hip.params["decoder"], ens_ref.params["decoder"]) < 2e-3
+    for li in (0, 1):
+        for k in ("weight", "bias"):
+            err = _rel_err(ens_hip.params["encoder_layers"][li][k],
+                           ens_ref.params["encoder_layers"][li][k])
+            assert err < 2e-3, (li, k, err)
