@@ -57,6 +57,16 @@ def build(sig_name, d, n, M, device):
     elif sig_name == "lista":
         sig = FunctionalLISTADenoisingSAE
         models = [sig.init(d, n, 3, float(l1)) for l1 in l1s]
+    elif sig_name == "residual":
+        from sparse_coding_amd.models.lista import FunctionalResidualDenoisingSAE
+
+        sig = FunctionalResidualDenoisingSAE
+        models = [sig.init(d, n, 2, float(l1)) for l1 in l1s]
+    elif sig_name == "semilinear":
+        from sparse_coding_amd.models.semilinear import SemiLinearSAE
+
+        sig = SemiLinearSAE
+        models = [sig.init(d, n, float(l1), device=device) for l1 in l1s]
     else:
         raise ValueError(sig_name)
     return sig, models, no_stacking
@@ -96,7 +106,7 @@ def main():
     device = "cuda:0"
 
     for sig_name in ("tied", "untied", "masked_tied", "thresholding", "reverse",
-                     "centered", "positive", "topk", "lista"):
+                     "centered", "positive", "topk", "lista", "residual", "semilinear"):
         row = {"sig": sig_name}
         for backend in ("hip", "torch"):
             try:
