@@ -286,10 +286,14 @@ class HipSAEStep:
 
         gscale = 2.0 / (B * d)
         if self.tied and on_grads is not None and M > 1:
-            # chunk the weight-grad GEMMs over model halves so the first
-            # half's all-reduce rides under the second half's compute
-            half = M // 2
-            for sl in (slice(0, half), slice(half, M)):
+            # chunk the weight-grad GEMMs over model groups so each chunk's
+            # all-reduce rides under the next chunk's compute; quarters keep
+            # the exposed tail to one chunk's reduce (xGMI ring: ~2*S*7/8 /
+            # 153 GB/s per link for an S-byte chunk)
+            n_chunks = min(M, 4)
+            bounds = [M * i // n_chunks for i in range(n_chunks + 1)]
+            for lo, hi in zip(bounds[:-1], bounds[1:]):
+                sl = slice(lo, hi)
                 ext.grad_w(self.c[sl], self.r[sl], self.gw[sl], gscale, 0.0, bk_gw, prio)
                 ext.grad_w(self.gpre[sl], x, self.gw[sl], 1.0, 1.0, bk_gw, prio)
                 on_grads([self.gw[sl]])
