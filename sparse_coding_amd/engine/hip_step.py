@@ -151,7 +151,6 @@ class HipSAEStep:
         # |code| in the L1, sign-aware backward, NO bias gradient from the
         # code path (only the L2-decay term moves the bias)
         self.reverse = reverse
-        self._ws = {}
 
         p = ensemble.params
         self.n_models, self.n_dict, self.d_act = p["encoder"].shape
@@ -890,7 +889,6 @@ class HipResidualDenoisingStep:
         self.rr = f(M, B, d)
         self.g_x = f(M, B, n)
         self.g_h = f(M, B, n)
-        self.t_n = f(M, B, n)
         self.gA = f(M, n, d)
         self.gW = [f(M, n, n) for _ in range(L)]
         self.g_theta = [f(M, n) for _ in range(L)]
