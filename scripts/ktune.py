@@ -25,13 +25,13 @@ if _ROOT not in sys.path:
     sys.path.insert(0, _ROOT)
 
 
-def bench_combo(args, staging, bk, prio, bk_gw=None):
+def bench_combo(args, staging, bk, prio, bk_gw=None, bk_dec=None):
     from sparse_coding_amd.engine.ensemble import FunctionalEnsemble
     from sparse_coding_amd.functional.optim import adam
     from sparse_coding_amd.models.sae_signatures import FunctionalTiedSAE
     from sparse_coding_amd.ops.kconfig import set_kernel_config
 
-    set_kernel_config(staging=staging, bk=bk, prio=prio, bk_grad_w=bk_gw)
+    set_kernel_config(staging=staging, bk=bk, prio=prio, bk_grad_w=bk_gw, bk_dec=bk_dec)
     device = "cuda:0"
     d, n_dict, M, B = args.d_model, args.d_model * args.dict_ratio, args.n_models, args.batch
     torch.manual_seed(0)
@@ -72,15 +72,18 @@ def main():
     best = min(results, key=lambda r: r["ms_per_step"])
     print(json.dumps({"winner": best}), flush=True)
 
-    # mixed: winner's staging/prio with per-kernel bk for grad_w (K=B long
-    # reduction may prefer the deeper tile even when enc/gc prefer bk16)
-    for bk_gw in (16, 32):
-        if bk_gw == best["bk"]:
-            continue
-        ms, acts = bench_combo(args, best["staging"], best["bk"], best["prio"], bk_gw=bk_gw)
-        rec = {"staging": best["staging"], "bk": best["bk"], "prio": best["prio"],
-               "bk_grad_w": bk_gw, "ms_per_step": round(ms, 4), "acts_per_sec": round(acts)}
-        print(json.dumps(rec), flush=True)
+    # mixed: winner's staging/prio with per-kernel bk for the long-K
+    # kernels (grad_w K=B, dec K=n): deep tiles can win there even when
+    # occupancy wins the short-K kernels
+    for key in ("bk_grad_w", "bk_dec"):
+        for bk_k in (16, 32):
+            if bk_k == best["bk"]:
+                continue
+            ms, acts = bench_combo(args, best["staging"], best["bk"], best["prio"],
+                                   **({"bk_gw": bk_k} if key == "bk_grad_w" else {"bk_dec": bk_k}))
+            rec = {"staging": best["staging"], "bk": best["bk"], "prio": best["prio"],
+                   key: bk_k, "ms_per_step": round(ms, 4), "acts_per_sec": round(acts)}
+            print(json.dumps(rec), flush=True)
 
 
 if __name__ == "__main__":

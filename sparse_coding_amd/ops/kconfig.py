@@ -31,9 +31,12 @@ DEFAULTS = {
     "bk": 16,
     "prio": False,
     "staging": "t",
-    # per-kernel overrides, None -> use "bk"
+    # per-kernel overrides, None -> use "bk".  dec_fwd reduces over the
+    # dictionary (K = n, long): the deeper TBK=32 tile wins there even
+    # though TBK=16's occupancy wins the short-K kernels (643.6k vs 639.0k
+    # acts/s on the flagship bench).
     "bk_grad_w": None,
-    "bk_dec": None,
+    "bk_dec": 32,
 }
 
 _cfg = dict(DEFAULTS)

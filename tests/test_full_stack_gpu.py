@@ -279,3 +279,32 @@ def test_dp_rccl_one_rank_overlap_path(tmp_path):
             assert err < 1e-6, (k, err)
     finally:
         dist.destroy_process_group()
+
+
+def test_flagship_soak_1000_steps():
+    """1000 fused steps at the flagship shape: finite losses throughout,
+    loss improves, fired counters accumulate monotonically."""
+    import numpy as np
+
+    from sparse_coding_amd.engine.ensemble import FunctionalEnsemble
+    from sparse_coding_amd.functional.optim import adam
+    from sparse_coding_amd.models.sae_signatures import FunctionalTiedSAE
+
+    torch.manual_seed(41)
+    M, B, d, n = 4, 2048, 512, 4096
+    models = [FunctionalTiedSAE.init(d, n, float(l1), device=DEV)
+              for l1 in np.logspace(-4, -3, M)]
+    ens = FunctionalEnsemble(models, FunctionalTiedSAE, adam, {"lr": 1e-3}, device=DEV, backend="hip")
+    pool = [torch.randn(B, d, device=DEV) for _ in range(4)]
+    l0 = None
+    for i in range(1000):
+        losses, _ = ens.step_batch(pool[i % 4])
+        if i == 0:
+            l0 = losses["loss"].clone()
+        if i % 250 == 0:
+            assert torch.isfinite(losses["loss"]).all(), i
+    torch.cuda.synchronize()
+    assert torch.isfinite(losses["loss"]).all()
+    assert (losses["loss"] < l0).all()
+    fired = ens._hip_step.fired
+    assert (fired.sum(dim=1) > 0).all()

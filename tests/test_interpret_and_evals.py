@@ -199,3 +199,17 @@ def test_deep_ae():
     ae.train_on([x] * 200, lr=5e-3)
     loss1, *_ = ae(x)
     assert loss1 < loss0
+
+
+@pytest.mark.timeout(600)
+def test_example_notebooks_execute(tmp_path, monkeypatch):
+    """The example notebooks run top to bottom on CPU (code cells exec'd in
+    order; plots land in tmp)."""
+    import json
+
+    monkeypatch.chdir(tmp_path)
+    root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    for name in ("dict_across_time", "dict_compare", "interpreting_sparse_dictionaries"):
+        nb = json.load(open(os.path.join(root, "examples", f"{name}.ipynb")))
+        src = "\n".join("".join(c["source"]) for c in nb["cells"] if c["cell_type"] == "code")
+        exec(compile(src, name, "exec"), {})  # noqa: S102 - our own notebooks
