@@ -189,3 +189,107 @@ def run_gpt2_small_mlp_sweep(cfg):
     cfg.model_name = "gpt2"
     cfg.layer_loc = "mlpout"
     return sweep(lambda c: make_grid_ensembles(c, FunctionalTiedSAE, np.logspace(-4, -2, 8), (1, 2, 4, 8)), cfg)
+
+
+def lista_init(cfg, n_hidden_layers: int = 3):
+    """LISTA denoising SAEs across l1 (reference :341-400)."""
+    from sparse_coding_amd.models.lista import FunctionalLISTADenoisingSAE
+
+    d = cfg.activation_width
+    devices = default_devices()
+    n_dict = int(d * cfg.learned_dict_ratio)
+    l1s = np.logspace(-4, -2, 8)
+    models = [FunctionalLISTADenoisingSAE.init(d, n_dict, n_hidden_layers, float(l1)) for l1 in l1s]
+    ens = FunctionalEnsemble(models, FunctionalLISTADenoisingSAE, adam, {"lr": cfg.lr}, device=devices[0])
+    args = {"batch_size": cfg.batch_size, "device": devices[0], "dict_size": n_dict}
+    return ([(ens, args, "lista")], ["dict_size"], ["l1_alpha"],
+            {"l1_alpha": [float(x) for x in l1s], "dict_size": [n_dict]})
+
+
+def residual_denoising_init(cfg, n_hidden_layers: int = 2):
+    """Residual-denoising SAEs (reference residual_denoising_experiment /
+    run_resid_denoise :1280-area)."""
+    from sparse_coding_amd.models.lista import FunctionalResidualDenoisingSAE
+
+    d = cfg.activation_width
+    devices = default_devices()
+    n_dict = int(d * cfg.learned_dict_ratio)
+    l1s = np.logspace(-4, -2, 8)
+    models = [FunctionalResidualDenoisingSAE.init(d, n_dict, n_hidden_layers, float(l1)) for l1 in l1s]
+    ens = FunctionalEnsemble(models, FunctionalResidualDenoisingSAE, adam, {"lr": cfg.lr}, device=devices[0])
+    args = {"batch_size": cfg.batch_size, "device": devices[0], "dict_size": n_dict}
+    return ([(ens, args, "resid_denoise")], ["dict_size"], ["l1_alpha"],
+            {"l1_alpha": [float(x) for x in l1s], "dict_size": [n_dict]})
+
+
+def positive_init(cfg):
+    """Non-negative-encoder tied SAEs with the reference's +0.18 input shift
+    (reference run_positive :1034-1171; mlp_tests.py)."""
+    from sparse_coding_amd.models.positive import FunctionalPositiveTiedSAE
+
+    return make_grid_ensembles(cfg, FunctionalPositiveTiedSAE, np.logspace(-4, -2, 8))
+
+
+def run_thresholding(cfg):
+    return sweep(thresholding_init, cfg)
+
+
+def run_lista(cfg):
+    return sweep(lista_init, cfg)
+
+
+def run_resid_denoise(cfg):
+    return sweep(residual_denoising_init, cfg)
+
+
+def run_positive(cfg):
+    return sweep(positive_init, cfg)
+
+
+def run_zero_l1_baseline(cfg):
+    return sweep(zero_l1_baseline_init, cfg)
+
+
+def run_dict_ratio(cfg):
+    return sweep(dict_ratio_range_init, cfg)
+
+
+def run_synthetic(cfg):
+    cfg.use_synthetic_dataset = True
+    return sweep(synthetic_sweep_init, cfg)
+
+
+def run_across_layers_attn(cfg, layers: Sequence[int]):
+    cfg.layer_loc = "attn"
+    return run_across_layers(cfg, layers)
+
+
+def run_across_layers_mlp_out(cfg, layers: Sequence[int]):
+    cfg.layer_loc = "mlpout"
+    return run_across_layers(cfg, layers)
+
+
+def run_across_layers_mlp_untied(cfg, layers: Sequence[int]):
+    cfg.layer_loc = "mlp"
+    cfg.tied_ae = False
+    return run_across_layers(cfg, layers)
+
+
+def run_across_layers_mlp_long(cfg, layers: Sequence[int], n_chunks: int = 60):
+    """Long MLP sweep (reference long_mlp_sweep :956-1031): more chunks,
+    mlp location."""
+    cfg.layer_loc = "mlp"
+    cfg.n_chunks = n_chunks
+    return run_across_layers(cfg, layers)
+
+
+def run_single_layer(cfg, layer: int = 2):
+    """One layer end to end on the default model (reference run_single_layer)."""
+    cfg.layer = layer
+    return sweep(dense_l1_range_init, cfg)
+
+
+def run_single_layer_gpt2(cfg, layer: int = 5):
+    cfg.model_name = "gpt2"
+    cfg.layer = layer
+    return sweep(dense_l1_range_init, cfg)
