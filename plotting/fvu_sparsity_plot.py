@@ -8,6 +8,11 @@ ICA / identity baselines as reference points.
 
 from __future__ import annotations
 
+import os as _os
+import sys as _sys
+
+_sys.path.insert(0, _os.path.dirname(_os.path.dirname(_os.path.abspath(__file__))))
+
 import argparse
 import os
 from collections import defaultdict

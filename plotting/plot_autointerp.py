@@ -7,6 +7,11 @@ neuron baselines), on the fixed -0.2..0.6 axis used throughout the reference.
 
 from __future__ import annotations
 
+import os as _os
+import sys as _sys
+
+_sys.path.insert(0, _os.path.dirname(_os.path.dirname(_os.path.abspath(__file__))))
+
 import argparse
 import os
 from typing import Dict, List

@@ -7,6 +7,11 @@ of the output distribution against the clean model, plotted against L0.
 
 from __future__ import annotations
 
+import os as _os
+import sys as _sys
+
+_sys.path.insert(0, _os.path.dirname(_os.path.dirname(_os.path.abspath(__file__))))
+
 import argparse
 from typing import Dict, List, Tuple
 

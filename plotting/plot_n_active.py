@@ -6,6 +6,11 @@ number of ever-active features (and dead fraction) per saved checkpoint.
 
 from __future__ import annotations
 
+import os as _os
+import sys as _sys
+
+_sys.path.insert(0, _os.path.dirname(_os.path.dirname(_os.path.abspath(__file__))))
+
 import argparse
 import os
 import re

@@ -6,6 +6,11 @@ metrics per hyperparameter setting from a learned_dicts.pt checkpoint.
 
 from __future__ import annotations
 
+import os as _os
+import sys as _sys
+
+_sys.path.insert(0, _os.path.dirname(_os.path.dirname(_os.path.abspath(__file__))))
+
 import argparse
 from collections import defaultdict
 
