@@ -245,6 +245,9 @@ def init_synthetic_dataset(cfg) -> None:
 
 def sweep(ensemble_init_func, cfg):
     """The trainer entry point (reference :298-386)."""
+    # the reference flips grad mode globally and leaves it off
+    # (big_sweep.py:299); restore it on exit so library callers are safe
+    _prev_grad_mode = torch.is_grad_enabled()
     torch.set_grad_enabled(False)
     if torch.cuda.is_available():
         torch.cuda.empty_cache()
@@ -343,6 +346,7 @@ def sweep(ensemble_init_func, cfg):
             _save_resume_state(resume_path, ensembles, chunk_order, i + 1)
 
     cfg.logger.close()
+    torch.set_grad_enabled(_prev_grad_mode)
     return learned_dicts
 
 

@@ -170,10 +170,12 @@ def test_project_adam_matches_autograd(shapes):
     inv = torch.empty(M, n, device=DEV)
     ext.row_norms(W, norms, inv, 1e-8)
 
-    # autograd reference for the projected gradient
-    W_ref = W.clone().requires_grad_()
-    What = W_ref / torch.clamp(torch.norm(W_ref, dim=-1, keepdim=True), 1e-8)
-    (What * gw).sum().backward()
+    # autograd reference for the projected gradient (grad mode explicitly on:
+    # a prior sweep() in the same process must not poison this)
+    with torch.enable_grad():
+        W_ref = W.clone().requires_grad_()
+        What = W_ref / torch.clamp(torch.norm(W_ref, dim=-1, keepdim=True), 1e-8)
+        (What * gw).sum().backward()
     g_ref = W_ref.grad
 
     W_out = W.clone()
