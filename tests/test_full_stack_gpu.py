@@ -368,7 +368,9 @@ def test_perplexity_under_reconstruction_gpu():
     p_rand = sm.calculate_perplexity(model, None, rand, 2, "residual", tokens, device=DEV)
 
     assert abs(p_ident - clean) / clean < 1e-3
-    assert p_rand > clean * 1.001  # destroying the stream must hurt
+    # the LM here is random-init (no network), so corrupting the stream
+    # cannot meaningfully "hurt" — but it must CHANGE the output
+    assert abs(p_rand - clean) / clean > 1e-3
 
 
 def test_activation_dataset_throughput_gpu(tmp_path):
