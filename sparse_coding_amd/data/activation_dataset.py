@@ -95,14 +95,22 @@ def make_tensor_name(layer: int, layer_loc: str, model_name: str) -> str:
     raise AssertionError
 
 
-def load_model(model_name: str, device="cpu", random_init: bool = True, local_path: Optional[str] = None):
+def load_model(model_name: str, device="cpu", random_init: bool = True, local_path: Optional[str] = None,
+               dtype: Optional[torch.dtype] = None):
     """Build the host LM.  With no network access, models come up random-init
     from the config table (BASELINE.json configs specify random-init weights);
-    pass ``local_path`` to load real weights from disk."""
+    pass ``local_path`` to load real weights from disk.
+
+    ``dtype``: run the host LM in bf16/fp16 to roughly double data-plane
+    throughput — the captured activations are stored fp16 regardless
+    (reference activation_dataset.py:364-388), so the chunk format and the
+    fp32 TRAINING dtype are unaffected."""
     from transformers import AutoModelForCausalLM, GPT2Config, GPT2LMHeadModel, GPTNeoXConfig, GPTNeoXForCausalLM
 
     if local_path is not None:
         model = AutoModelForCausalLM.from_pretrained(local_path)
+        if dtype is not None:
+            model = model.to(dtype)
         return model.to(device)
 
     family, hidden, n_layer, n_head, inter, vocab = _lookup(model_name)
@@ -127,6 +135,8 @@ def load_model(model_name: str, device="cpu", random_init: bool = True, local_pa
         )
         model = GPT2LMHeadModel(cfg)
     model.eval()
+    if dtype is not None:
+        model = model.to(dtype)
     return model.to(device)
 
 

@@ -384,21 +384,22 @@ def test_activation_dataset_throughput_gpu(tmp_path):
         synthetic_token_batches,
     )
 
-    model = load_model("pythia-70m", device=DEV)
     n_batches, bsz, seq = 24, 16, 256
-    t0 = time.perf_counter()
-    total = make_activation_dataset_hf(
-        synthetic_token_batches(model.config.vocab_size, bsz, seq, n_batches),
-        model, [2], "residual",
-        chunk_size=65536, n_chunks=2,
-        output_folder=str(tmp_path), device=DEV, model_name="pythia-70m",
-    )
-    torch.cuda.synchronize()
-    dt = time.perf_counter() - t0
-    import os
-
-    files = sorted(os.listdir(tmp_path))
-    assert "0.pt" in files
-    chunk = torch.load(tmp_path / "0.pt")
-    assert chunk.dtype == torch.float16 and chunk.shape[1] == model.config.hidden_size
-    print(f"[data-plane] {total} activations in {dt:.2f}s = {total/dt:,.0f} acts/s")
+    for dtype, tag in ((None, "fp32"), (torch.bfloat16, "bf16")):
+        model = load_model("pythia-70m", device=DEV, dtype=dtype)
+        out = tmp_path / tag
+        t0 = time.perf_counter()
+        total = make_activation_dataset_hf(
+            synthetic_token_batches(model.config.vocab_size, bsz, seq, n_batches),
+            model, [2], "residual",
+            chunk_size=65536, n_chunks=2,
+            output_folder=str(out), device=DEV, model_name="pythia-70m",
+        )
+        torch.cuda.synchronize()
+        dt = time.perf_counter() - t0
+        chunk = torch.load(out / "0.pt")
+        assert chunk.dtype == torch.float16 and chunk.shape[1] == model.config.hidden_size
+        assert torch.isfinite(chunk.float()).all()
+        print(f"[data-plane {tag}] {total} activations in {dt:.2f}s = {total/dt:,.0f} acts/s")
+        del model
+        torch.cuda.empty_cache()
