@@ -1162,6 +1162,18 @@ class HipTopKStep:
         if name != "adam":
             raise RuntimeError(f"fused HIP step supports adam only, got {name}")
         self._B = None
+        # models grouped by k: one batched torch.topk per distinct k instead
+        # of a python loop per model
+        self.k_groups = {}
+        for m, k in enumerate(self.ks):
+            self.k_groups.setdefault(k, []).append(m)
+        self.k_groups = {k: torch.tensor(ms, device=p["dict"].device)
+                         for k, ms in self.k_groups.items()}
+        import os as _os
+
+        self.use_graph = _os.environ.get("SPARSE_CODING_AMD_NO_GRAPH") != "1"
+        self._graph = None
+        self._eager_steps = 0
 
     def _alloc(self, B):
         M, n, d = self.n_models, self.n_dict, self.d_act
@@ -1178,6 +1190,8 @@ class HipTopKStep:
         self.g_bias_scratch = f(M, n)
         self.gw = f(M, n, d)
         self._B = B
+        self._graph = None
+        self._eager_steps = 0
 
     def grads_phase(self, x):
         ens, ext = self.ens, self.ext
@@ -1193,13 +1207,15 @@ class HipTopKStep:
         ext.row_norms(W, self.norms, self.inv_norms, self.EPS)
         ext.enc_fwd(x, W, self.dummy_bias, self.inv_norms,
                     self.scores, self.loss_parts, self.fired, 1)
-        # top-k selection per model (k varies across the ensemble)
+        # top-k selection, one batched topk per distinct k
         self.c.zero_()
-        for m, k in enumerate(self.ks):
-            sc = self.scores[m]
+        for k, midx in self.k_groups.items():
+            sc = self.scores.index_select(0, midx)  # [G, B, n]
             top = torch.topk(sc, k, dim=-1)
             vals = torch.clamp(top.values, min=0.0)
-            self.c[m].scatter_(-1, top.indices, vals)
+            cg = torch.zeros_like(sc)
+            cg.scatter_(-1, top.indices, vals)
+            self.c.index_copy_(0, midx, cg)
         self.fired += (self.c > 0).float().sum(dim=1)
 
         ext.dec_fwd(self.c, W, self.inv_norms, x, self.r, self.loss_parts)
@@ -1223,12 +1239,35 @@ class HipTopKStep:
         mse = self.loss_parts[:, 0] / (B * self.d_act)
         return {"loss": mse}
 
+    def _capture(self, x):
+        try:
+            self.x_static = x.clone()
+            torch.cuda.synchronize()
+            g = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(g):
+                B = self.grads_phase(self.x_static)
+                self.update_phase(B)
+            self._graph = g
+        except Exception as e:  # noqa: BLE001 - graphs are an optimization only
+            print(f"[hip_step] hipGraph capture failed ({e}); staying eager")
+            self.use_graph = False
+            self._graph = None
+
     def step(self, minibatches, expand_dims=True):
         if not expand_dims:
             # per-model batches not needed: TopK stacks fine on this path
             raise NotImplementedError
+        B = minibatches.shape[0]
+        if self.use_graph:
+            if self._B == B and self._graph is None and self._eager_steps >= 2:
+                self._capture(minibatches.contiguous())
+            if self._graph is not None:
+                self.x_static.copy_(minibatches)
+                self._graph.replay()
+                return self._loss_data(B), {"c": self.c}
         B = self.grads_phase(minibatches)
         self.update_phase(B)
+        self._eager_steps += 1
         return self._loss_data(B), {"c": self.c}
 
     def dp_grad_tensors(self):
