@@ -1150,6 +1150,7 @@ class HipTopKStep:
         self.n_models, self.n_dict, self.d_act = p["dict"].shape
         dev = p["dict"].device
         self.ks = [int(k) for k in ensemble.buffers["sparsity"].reshape(-1).tolist()]
+        self.ks_i32 = torch.tensor(self.ks, device=dev, dtype=torch.int32)
         self.zero_l1 = torch.zeros(self.n_models, device=dev)
         self.dummy_bias = torch.zeros(self.n_models, self.n_dict, device=dev)
 
@@ -1162,13 +1163,6 @@ class HipTopKStep:
         if name != "adam":
             raise RuntimeError(f"fused HIP step supports adam only, got {name}")
         self._B = None
-        # models grouped by k: one batched torch.topk per distinct k instead
-        # of a python loop per model
-        self.k_groups = {}
-        for m, k in enumerate(self.ks):
-            self.k_groups.setdefault(k, []).append(m)
-        self.k_groups = {k: torch.tensor(ms, device=p["dict"].device)
-                         for k, ms in self.k_groups.items()}
         import os as _os
 
         self.use_graph = _os.environ.get("SPARSE_CODING_AMD_NO_GRAPH") != "1"
@@ -1207,16 +1201,9 @@ class HipTopKStep:
         ext.row_norms(W, self.norms, self.inv_norms, self.EPS)
         ext.enc_fwd(x, W, self.dummy_bias, self.inv_norms,
                     self.scores, self.loss_parts, self.fired, 1)
-        # top-k selection, one batched topk per distinct k
-        self.c.zero_()
-        for k, midx in self.k_groups.items():
-            sc = self.scores.index_select(0, midx)  # [G, B, n]
-            top = torch.topk(sc, k, dim=-1)
-            vals = torch.clamp(top.values, min=0.0)
-            cg = torch.zeros_like(sc)
-            cg.scatter_(-1, top.indices, vals)
-            self.c.index_copy_(0, midx, cg)
-        self.fired += (self.c > 0).float().sum(dim=1)
+        # exact per-row radix top-k + scatter + fired counts in ONE kernel
+        # (k_topk_select) — replaces the torch.topk/scatter chain
+        ext.topk_select(self.scores, self.c, self.fired, self.ks_i32)
 
         ext.dec_fwd(self.c, W, self.inv_norms, x, self.r, self.loss_parts)
         ext.gc(self.r, W, self.inv_norms, self.c, self.zero_l1,

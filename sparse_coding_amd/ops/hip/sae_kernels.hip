@@ -797,6 +797,93 @@ void k_lista_bwd_elem(const float* __restrict__ g_y,
 }
 
 // ---------------------------------------------------------------------------
+// k_topk_select: per-row top-k selection + scatter for the TopK encoder
+// (SURVEY.md K8).  One 256-thread block per (model, batch-row); exact
+// radix select over the order-preserving uint mapping of fp32 (4 passes of
+// 8-bit histograms), then one write pass producing the dense code row
+//   c[j] = j in top-k ? max(score[j], 0) : 0     (reference topk_encoder.py
+// keeps clamped values in the selected slots), with fired counts fused.
+// Exact-equal ties at the threshold are admitted in arbitrary order via an
+// LDS slot counter (measure-zero for continuous scores).
+// ---------------------------------------------------------------------------
+#define TOPK_T 256
+__device__ __forceinline__ unsigned f32_ord(float f) {
+  unsigned u = __float_as_uint(f);
+  return (u & 0x80000000u) ? ~u : (u | 0x80000000u);
+}
+
+extern "C" __global__ __launch_bounds__(TOPK_T)
+void k_topk_select(const float* __restrict__ scores,  // [M, B, n]
+                   float* __restrict__ c_out,         // [M, B, n]
+                   float* __restrict__ fired,         // [M, n]
+                   const int* __restrict__ ks,        // [M]
+                   int B, int n) {
+  const int m = blockIdx.z;
+  const int row = blockIdx.x;
+  if (row >= B) return;
+  const float* s = scores + ((long)m * B + row) * n;
+  float* c = c_out + ((long)m * B + row) * n;
+  float* fired_m = fired + (long)m * n;
+  int k = ks[m];
+  if (k >= n) k = n;
+
+  __shared__ unsigned hist[256];
+  __shared__ unsigned sh_prefix, sh_mask, sh_kleft, sh_slots;
+
+  if (threadIdx.x == 0) {
+    sh_prefix = 0u;
+    sh_mask = 0u;  // bits of the prefix that are decided
+    sh_kleft = (unsigned)k;
+  }
+  __syncthreads();
+
+  // 4 radix passes, MSB first
+  for (int shift = 24; shift >= 0; shift -= 8) {
+    if (threadIdx.x < 256) hist[threadIdx.x] = 0u;
+    __syncthreads();
+    unsigned prefix = sh_prefix, mask = sh_mask;
+    for (int j = threadIdx.x; j < n; j += TOPK_T) {
+      unsigned u = f32_ord(s[j]);
+      if ((u & mask) == prefix) atomicAdd(&hist[(u >> shift) & 0xFFu], 1u);
+    }
+    __syncthreads();
+    if (threadIdx.x == 0) {
+      unsigned kleft = sh_kleft;
+      int bin = 255;
+      for (; bin >= 0; --bin) {
+        if (hist[bin] >= kleft) break;
+        kleft -= hist[bin];
+      }
+      if (bin < 0) bin = 0;  // defensive; cannot happen for k <= n
+      sh_kleft = kleft;
+      sh_prefix = sh_prefix | ((unsigned)bin << shift);
+      sh_mask = sh_mask | (0xFFu << shift);
+    }
+    __syncthreads();
+  }
+
+  // threshold = exact k-th largest key; elements with key > thr are all
+  // selected; sh_kleft of the == thr ties are selected in index order
+  unsigned thr = sh_prefix;
+  if (threadIdx.x == 0) sh_slots = sh_kleft;
+  __syncthreads();
+  for (int j = threadIdx.x; j < n; j += TOPK_T) {
+    unsigned u = f32_ord(s[j]);
+    float v = 0.f;
+    bool sel = u > thr;
+    if (!sel && u == thr) {
+      unsigned slot = atomicSub(&sh_slots, 1u);
+      sel = (slot != 0u && slot <= (unsigned)n);  // old value; wraps reject
+    }
+    if (sel) {
+      v = fmaxf(s[j], 0.f);
+      if (v > 0.f) atomicAdd(&fired_m[j], 1.f);
+    }
+    c[j] = v;
+  }
+}
+
+// ---------------------------------------------------------------------------
 // k_grad_w_t
 // ---------------------------------------------------------------------------
 template <int TBK, int MINW>

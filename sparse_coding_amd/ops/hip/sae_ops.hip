@@ -325,6 +325,18 @@ void gc2(torch::Tensor rT, torch::Tensor WT, torch::Tensor c,
                        B, d, n, prio ? 1 : 0, (int)gc_mode);
 }
 
+void topk_select(torch::Tensor scores, torch::Tensor c_out, torch::Tensor fired,
+                 torch::Tensor ks) {
+  CHECK_IN(scores); CHECK_IN(c_out); CHECK_IN(fired);
+  TORCH_CHECK(ks.is_cuda() && ks.is_contiguous() && ks.scalar_type() == torch::kInt32,
+              "ks must be int32 GPU");
+  int M = scores.size(0), B = scores.size(1), n = scores.size(2);
+  dim3 grid(B, 1, M);
+  hipLaunchKernelGGL(k_topk_select, grid, dim3(TOPK_T), 0, cur_stream(),
+                     scores.data_ptr<float>(), c_out.data_ptr<float>(),
+                     fired.data_ptr<float>(), ks.data_ptr<int>(), B, n);
+}
+
 void lista_bwd_elem(torch::Tensor g_y, c10::optional<torch::Tensor> carry_in,
                     torch::Tensor r, torch::Tensor theta, torch::Tensor x,
                     torch::Tensor x_prev, torch::Tensor mom,
@@ -367,6 +379,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("u_out") = py::none(), py::arg("dict_sizes") = py::none(),
         py::arg("y_in") = py::none(), py::arg("x_prev") = py::none(),
         py::arg("x_out") = py::none(), py::arg("mom") = py::none());
+  m.def("topk_select", &topk_select, "per-row radix top-k + scatter (+fired)",
+        py::arg("scores"), py::arg("c_out"), py::arg("fired"), py::arg("ks"));
   m.def("lista_bwd_elem", &lista_bwd_elem, "fused LISTA backward elementwise pass",
         py::arg("g_y"), py::arg("carry_in"), py::arg("r"), py::arg("theta"),
         py::arg("x"), py::arg("x_prev"), py::arg("mom"), py::arg("g_r"),
