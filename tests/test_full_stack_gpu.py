@@ -403,3 +403,26 @@ def test_activation_dataset_throughput_gpu(tmp_path):
         print(f"[data-plane {tag}] {total} activations in {dt:.2f}s = {total/dt:,.0f} acts/s")
         del model
         torch.cuda.empty_cache()
+
+
+def test_baselines_runner_gpu(tmp_path):
+    """C21: per-layer baseline suite (streaming PCA on device + topk-PCA /
+    random / identity-relu exports) against a real fp16 chunk file."""
+    from sparse_coding_amd.sweep.baselines import run_layer_baselines
+
+    torch.manual_seed(44)
+    d = 64
+    chunk = (torch.randn(4096, d) @ torch.randn(d, d)).to(torch.float16)
+    chunk_path = str(tmp_path / "0.pt")
+    torch.save(chunk, chunk_path)
+
+    out = run_layer_baselines(2, chunk_path, str(tmp_path / "base"),
+                              device=DEV, sparsity=8, do_ica=False)
+    import os
+
+    assert set(out) == {"pca", "pca_topk", "pca_rot", "random", "identity_relu"}
+    ld = torch.load(os.path.join(tmp_path / "base", "pca_topk_l2.pt"), weights_only=False)
+    # saved baselines load as reference-module-path LearnedDicts
+    assert type(ld).__module__.startswith("autoencoders.")
+    c = ld.encode(torch.randn(16, d, device=DEV))
+    assert (c != 0).sum(dim=-1).max() <= 8
