@@ -424,5 +424,6 @@ def test_baselines_runner_gpu(tmp_path):
     ld = torch.load(os.path.join(tmp_path / "base", "pca_topk_l2.pt"), weights_only=False)
     # saved baselines load as reference-module-path LearnedDicts
     assert type(ld).__module__.startswith("autoencoders.")
+    ld.to_device(DEV)  # baselines are saved as CPU copies
     c = ld.encode(torch.randn(16, d, device=DEV))
     assert (c != 0).sum(dim=-1).max() <= 8
