@@ -45,9 +45,15 @@ def job_wrapper(job, ensemble_state_dict, cfg, args, tag, dataset, done_flag, pr
 
 
 def dispatch_job_on_chunk(ensembles: List[Tuple[Any, dict, str]], cfg, chunk: torch.Tensor, job: Callable, poll_s: float = 0.1):
-    """Run `job` on every (ensemble, args, name) against one shared chunk."""
-    if torch.cuda.is_available():
-        chunk.pin_memory()
+    """Run `job` on every (ensemble, args, name) against one shared chunk.
+
+    NOTE the reference "pins" the chunk with a discarded `chunk.pin_memory()`
+    (cluster_runs.py:101 — pin_memory returns a copy, so it is a no-op) and
+    then streams every batch over unpinned H2D.  Here the chunk goes to
+    shared CPU memory for the spawn children, and each child stages it into
+    its GPU's HBM ONCE (big_sweep.ensemble_train_loop) — with 288 GB per GPU,
+    whole-chunk residency is the right call.
+    """
     chunk.share_memory_()
 
     processes = []
@@ -94,8 +100,6 @@ def dispatch_job_on_chunk(ensembles: List[Tuple[Any, dict, str]], cfg, chunk: to
 
 def dispatch_lite(cfg, chunk: torch.Tensor, ensemble, name: str, job: Callable):
     """Single-ensemble async dispatch (reference cluster_runs.py:50-86)."""
-    if torch.cuda.is_available():
-        chunk.pin_memory()
     chunk.share_memory_()
     ensemble.to_shared_memory()
     done = _ctx.Value("i", 0)
