@@ -31,6 +31,8 @@ def main():
     p.add_argument("--n-chunks", type=int, default=4)
     p.add_argument("--chunk-gb", type=float, default=0.5)
     p.add_argument("--n-repetitions", type=int, default=2)
+    p.add_argument("--persistent", action="store_true",
+                   help="persistent dispatch workers (one spawn per ensemble)")
     args = p.parse_args()
 
     from sparse_coding_amd.config import SyntheticEnsembleArgs
@@ -56,6 +58,7 @@ def main():
     cfg.output_folder = os.path.join(args.out_dir, "out")
     cfg.use_wandb = False
     cfg.wandb_images = False
+    cfg.persistent_workers = args.persistent
 
     l1s = np.logspace(-4.5, -3.2, 8)
 
@@ -86,6 +89,7 @@ def main():
                                 weights_only=False).shape[0]
                      for i in range(cfg.n_chunks)) * args.n_repetitions
     summary = {
+        "persistent_workers": args.persistent,
         "wall_seconds": round(wall, 1),
         "total_activations_seen": int(total_acts),
         "end_to_end_acts_per_sec": round(total_acts / wall),

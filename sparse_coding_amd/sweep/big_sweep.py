@@ -320,6 +320,15 @@ def sweep(ensemble_init_func, cfg):
     if getattr(cfg, "center_activations", False) and start_chunk > 0:
         means = torch.load(os.path.join(cfg.output_folder, "means.pt"), map_location="cpu")
 
+    # persistent workers: one spawn per ensemble for the WHOLE sweep instead
+    # of one per (ensemble, chunk) — kills the per-chunk CUDA-context spawn
+    # cost (sweep/worker_pool.py)
+    pool = None
+    if getattr(cfg, "persistent_workers", False):
+        from sparse_coding_amd.sweep.worker_pool import PersistentWorkerPool
+
+        pool = PersistentWorkerPool(ensembles, cfg, ensemble_train_loop)
+
     learned_dicts = []
     if start_chunk > 0:
         # a fully-resumed run (nothing left to train) still returns dicts
@@ -336,7 +345,10 @@ def sweep(ensemble_init_func, cfg):
                 torch.save(means, os.path.join(cfg.output_folder, "means.pt"))
             chunk -= means
 
-        dispatch_job_on_chunk(ensembles, cfg, chunk, ensemble_train_loop)
+        if pool is not None:
+            pool.run_chunk(chunk)
+        else:
+            dispatch_job_on_chunk(ensembles, cfg, chunk, ensemble_train_loop)
 
         learned_dicts = []
         for ensemble, arg, _ in ensembles:
@@ -356,6 +368,8 @@ def sweep(ensemble_init_func, cfg):
                 yaml.dump({k: v for k, v in cfg_dict.items() if isinstance(v, (int, float, str, bool, list, type(None)))}, f)
             _save_resume_state(resume_path, ensembles, chunk_order, i + 1)
 
+    if pool is not None:
+        pool.close()
     cfg.logger.close()
     torch.set_grad_enabled(_prev_grad_mode)
     return learned_dicts

@@ -159,3 +159,42 @@ def test_grad_bucket_allreducer_layout():
     red.all_reduce_(grads)
     for k in grads:
         assert torch.equal(before[k], grads[k])
+
+
+@pytest.mark.timeout(300)
+def test_persistent_pool_matches_dispatch(tmp_path):
+    """cfg.persistent_workers=True (one spawn per ensemble for the whole
+    sweep) must produce bit-identical dicts to the per-chunk dispatcher."""
+    from sparse_coding_amd.config import SyntheticEnsembleArgs
+    from sparse_coding_amd.sweep import big_sweep
+    from sparse_coding_amd.sweep.experiments import make_grid_ensembles
+
+    def make_cfg(sub, persistent):
+        cfg = SyntheticEnsembleArgs()
+        cfg.use_synthetic_dataset = True
+        cfg.activation_width = 16
+        cfg.n_ground_truth_components = 24
+        cfg.gen_batch_size = 256
+        cfg.feature_num_nonzero = 3
+        cfg.noise_magnitude_scale = 0.0
+        cfg.chunk_size_gb = 16 * 256 * 4 * 2 / 1024**3
+        cfg.n_chunks = 2
+        cfg.n_repetitions = 2
+        cfg.batch_size = 128
+        cfg.device = "cpu"
+        cfg.dataset_folder = str(tmp_path / sub / "data")
+        cfg.output_folder = str(tmp_path / sub / "out")
+        cfg.use_wandb = False
+        cfg.wandb_images = False
+        cfg.persistent_workers = persistent
+        return cfg
+
+    def init_func(c):
+        return make_grid_ensembles(c, FunctionalTiedSAE, [1e-4, 1e-3], [1.0], devices=["cpu"])
+
+    dicts_a = big_sweep.sweep(init_func, make_cfg("a", False))
+    dicts_b = big_sweep.sweep(init_func, make_cfg("b", True))
+    for (ld_a, hp_a), (ld_b, hp_b) in zip(dicts_a, dicts_b):
+        assert hp_a == hp_b
+        assert torch.equal(ld_a.get_learned_dict(), ld_b.get_learned_dict())
+        assert torch.equal(ld_a.encoder_bias, ld_b.encoder_bias)
