@@ -123,8 +123,8 @@ def ensemble_train_loop(ensemble, cfg, args, ensemble_name, sampler, dataset, pr
 
         progress_counter.value = i
 
-    if torch.cuda.is_available():
-        torch.cuda.synchronize(args["device"])
+    if device.type == "cuda":
+        torch.cuda.synchronize(device)
     dt = _time.perf_counter() - t0
     if logger is not None and dt > 0:
         logger.log({f"{ensemble_name}_acts_per_sec": n_acts / dt,
