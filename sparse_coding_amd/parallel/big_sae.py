@@ -187,3 +187,58 @@ def train_big_sae(
                 log_fn(f"[big_sae] resampled {n_dead} dead features")
 
     return model
+
+
+def main():
+    """torchrun entry point (reference huge_batch_size.py main :358-366):
+
+    torchrun --standalone --nproc-per-node N -m sparse_coding_amd.parallel.big_sae \\
+        --chunk-dir activation_data [--n-features 16384] [...]
+
+    Falls back to synthetic chunks when --chunk-dir has none (no network).
+    """
+    import argparse
+    import glob
+
+    p = argparse.ArgumentParser()
+    p.add_argument("--chunk-dir", default="activation_data")
+    p.add_argument("--activation-size", type=int, default=1024)
+    p.add_argument("--n-features", type=int, default=16384)
+    p.add_argument("--l1-alpha", type=float, default=1e-3)
+    p.add_argument("--lr", type=float, default=1e-3)
+    p.add_argument("--batch-size", type=int, default=4096)
+    p.add_argument("--reinit-every-chunks", type=int, default=10)
+    p.add_argument("--synthetic-chunks", type=int, default=0,
+                   help="generate N synthetic chunks into --chunk-dir first")
+    p.add_argument("--save", default="big_sae_state.pt")
+    args = p.parse_args()
+
+    if args.synthetic_chunks:
+        os.makedirs(args.chunk_dir, exist_ok=True)
+        for i in range(args.synthetic_chunks):
+            torch.save(torch.randn(65536, args.activation_size, dtype=torch.float16),
+                       os.path.join(args.chunk_dir, f"{i}.pt"))
+
+    paths = sorted(glob.glob(os.path.join(args.chunk_dir, "*.pt")))
+    if not paths:
+        raise SystemExit(f"no chunks in {args.chunk_dir} (use --synthetic-chunks N)")
+
+    model = train_big_sae(
+        paths,
+        activation_size=args.activation_size,
+        n_features=args.n_features,
+        l1_alpha=args.l1_alpha,
+        lr=args.lr,
+        batch_size=args.batch_size,
+        reinit_every_chunks=args.reinit_every_chunks,
+    )
+    rank = int(os.environ.get("RANK", "0"))
+    if rank == 0 and args.save:
+        torch.save(model.state_dict(), args.save)
+        print(f"saved {args.save}")
+    if dist.is_initialized():
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()
