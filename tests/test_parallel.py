@@ -198,3 +198,64 @@ def test_persistent_pool_matches_dispatch(tmp_path):
         assert hp_a == hp_b
         assert torch.equal(ld_a.get_learned_dict(), ld_b.get_learned_dict())
         assert torch.equal(ld_a.encoder_bias, ld_b.encoder_bias)
+
+
+def _shard_worker(rank, world_size, port, chunk, out_q):
+    try:
+        os.environ.update(MASTER_ADDR="127.0.0.1", MASTER_PORT=str(port),
+                          WORLD_SIZE=str(world_size), RANK=str(rank), LOCAL_RANK=str(rank))
+        dist.init_process_group("gloo", rank=rank, world_size=world_size)
+        from sparse_coding_amd.parallel.chunk_feed import ShardedEnsembleRunner
+        from sparse_coding_amd.sweep.big_sweep import ensemble_train_loop
+
+        torch.manual_seed(100 + rank)  # DIFFERENT ensemble per rank
+        l1 = [1e-4, 1e-3][rank]
+        models = [FunctionalTiedSAE.init(D, N, l1) for _ in range(2)]
+        ens = FunctionalEnsemble(models, FunctionalTiedSAE, adam, {"lr": 1e-3},
+                                 device="cpu", backend="torch")
+
+        class Cfg:
+            batch_size = 64
+            logger = None
+            log_every = 1000
+
+        runner = ShardedEnsembleRunner(ens, Cfg(), {"batch_size": 64, "device": "cpu",
+                                                    "l1_alpha": l1, "dict_size": N},
+                                       f"shard{rank}", ensemble_train_loop, "cpu")
+        # only rank 0 holds the chunk; others receive it by broadcast
+        runner.run_chunk(chunk if rank == 0 else None)
+        dicts = runner.gather_learned_dicts([], ["l1_alpha"])
+        if rank == 0:
+            out_q.put({
+                "n_dicts": len(dicts),
+                "l1s": sorted(round(hp["l1_alpha"], 6) for _, hp in dicts),
+                "first_dict": dicts[0][0].get_learned_dict().numpy(),
+            })
+        dist.destroy_process_group()
+    except Exception:  # noqa: BLE001
+        import traceback
+
+        out_q.put({"_error": f"rank {rank}: {traceback.format_exc()}"})
+        raise
+
+
+@pytest.mark.timeout(120)
+def test_sharded_ensemble_chunk_broadcast():
+    """P1 over collectives: rank 0 broadcasts the chunk, each rank trains a
+    DIFFERENT ensemble, dicts gather back to rank 0."""
+    torch.manual_seed(7)
+    chunk = torch.randn(512, D)
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    procs = [ctx.Process(target=_shard_worker, args=(r, 2, 29553, chunk, q)) for r in range(2)]
+    for p in procs:
+        p.start()
+    res = q.get(timeout=110)
+    for p in procs:
+        p.join(timeout=30)
+    assert "_error" not in res, res.get("_error")
+    assert res["n_dicts"] == 4  # 2 ranks x 2 models
+    assert res["l1s"] == [0.0001, 0.0001, 0.001, 0.001]
+    import numpy as np
+
+    assert np.isfinite(res["first_dict"]).all()
