@@ -427,3 +427,25 @@ def test_baselines_runner_gpu(tmp_path):
     ld.to_device(DEV)  # baselines are saved as CPU copies
     c = ld.encode(torch.randn(16, d, device=DEV))
     assert (c != 0).sum(dim=-1).max() <= 8
+
+
+def test_chunk_broadcast_rccl_one_rank():
+    """BroadcastChunkFeeder's RCCL call sequence on a 1-rank nccl group
+    (shape broadcast + payload broadcast land on-device)."""
+    import os
+
+    import torch.distributed as dist
+
+    from sparse_coding_amd.parallel.chunk_feed import BroadcastChunkFeeder
+
+    os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+    os.environ.setdefault("MASTER_PORT", "29559")
+    dist.init_process_group("nccl", rank=0, world_size=1)
+    try:
+        feeder = BroadcastChunkFeeder(DEV, src=0)
+        chunk = torch.randn(4096, 64)
+        out = feeder.feed(chunk)
+        assert out.is_cuda and out.shape == chunk.shape
+        assert torch.allclose(out.cpu(), chunk)
+    finally:
+        dist.destroy_process_group()
