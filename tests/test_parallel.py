@@ -259,3 +259,66 @@ def test_sharded_ensemble_chunk_broadcast():
     import numpy as np
 
     assert np.isfinite(res["first_dict"]).all()
+
+
+def _sharded_sweep_worker(rank, world_size, port, tmpdir, out_q):
+    try:
+        os.environ.update(MASTER_ADDR="127.0.0.1", MASTER_PORT=str(port),
+                          WORLD_SIZE=str(world_size), RANK=str(rank), LOCAL_RANK=str(rank))
+        from sparse_coding_amd.config import SyntheticEnsembleArgs
+        from sparse_coding_amd.sweep.sharded_sweep import sharded_sweep
+
+        cfg = SyntheticEnsembleArgs()
+        cfg.use_synthetic_dataset = True
+        cfg.activation_width = 16
+        cfg.n_ground_truth_components = 24
+        cfg.gen_batch_size = 256
+        cfg.feature_num_nonzero = 3
+        cfg.noise_magnitude_scale = 0.0
+        cfg.chunk_size_gb = 16 * 256 * 4 * 2 / 1024**3
+        cfg.n_chunks = 2
+        cfg.batch_size = 128
+        cfg.dataset_folder = os.path.join(tmpdir, "data")
+        cfg.output_folder = os.path.join(tmpdir, "out")
+        cfg.use_wandb = False
+        cfg.ensemble_hyperparams = ["dict_size"]
+        cfg.buffer_hyperparams = ["l1_alpha"]
+
+        def init_for_rank(c, r, w):
+            torch.manual_seed(200 + r)
+            l1 = [1e-4, 1e-3][r]
+            models = [FunctionalTiedSAE.init(16, 32, l1) for _ in range(2)]
+            ens = FunctionalEnsemble(models, FunctionalTiedSAE, adam, {"lr": 1e-3},
+                                     device="cpu", backend="torch")
+            return ens, {"batch_size": 128, "device": "cpu", "dict_size": 32, "l1_alpha": l1}, f"r{r}"
+
+        dicts = sharded_sweep(init_for_rank, cfg)
+        if rank == 0:
+            out_q.put({"n": len(dicts),
+                       "ckpt": os.path.exists(os.path.join(cfg.output_folder, "_1", "learned_dicts.pt"))})
+        import torch.distributed as dist_
+
+        dist_.destroy_process_group()
+    except Exception:  # noqa: BLE001
+        import traceback
+
+        out_q.put({"_error": f"rank {rank}: {traceback.format_exc()}"})
+        raise
+
+
+@pytest.mark.timeout(180)
+def test_sharded_sweep_end_to_end(tmp_path):
+    """The torchrun-able sharded sweep on gloo world 2: rank 0 generates +
+    broadcasts chunks, each rank trains its grid slice, checkpoints gather
+    in the reference layout."""
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    procs = [ctx.Process(target=_sharded_sweep_worker, args=(r, 2, 29563, str(tmp_path), q))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    res = q.get(timeout=170)
+    for p in procs:
+        p.join(timeout=30)
+    assert "_error" not in res, res.get("_error")
+    assert res["n"] == 4 and res["ckpt"]
