@@ -7,7 +7,7 @@ broadcasts it over xGMI; every rank trains ITS OWN ensemble (a slice of the
 hyperparameter grid) against its HBM-resident copy; checkpoints gather to
 rank 0 in the reference ``_{i}/learned_dicts.pt`` layout.
 
-    torchrun --standalone --nproc-per-node 8 -m sparse_coding_amd.sweep.sharded_sweep_demo
+    torchrun --standalone --nproc-per-node 8 -m sparse_coding_amd.sweep.sharded_sweep
 
 Library use: ``sharded_sweep(init_func_for_rank, cfg)`` where
 ``init_func_for_rank(cfg, rank, world) -> (ensemble, args, name)``.
@@ -86,3 +86,63 @@ def sharded_sweep(init_func_for_rank: Callable, cfg):
     if rank == 0 and cfg.logger is not None:
         cfg.logger.close()
     return learned_dicts
+
+
+def _demo_init_for_rank(cfg, rank: int, world: int):
+    """Default grid slice: rank r trains an 8-way L1 ensemble at dict ratio
+    2^r (so 8 ranks cover ratios 1..128)."""
+    from sparse_coding_amd.engine.ensemble import FunctionalEnsemble
+    from sparse_coding_amd.functional.optim import adam
+    from sparse_coding_amd.models.sae_signatures import FunctionalTiedSAE
+
+    torch.manual_seed(1000 + rank)
+    d = cfg.activation_width
+    ratio = 2 ** rank
+    n_dict = int(d * ratio)
+    l1s = np.logspace(-4, -2, 8)
+    models = [FunctionalTiedSAE.init(d, n_dict, float(l1), device=cfg.device) for l1 in l1s]
+    ens = FunctionalEnsemble(models, FunctionalTiedSAE, adam, {"lr": cfg.lr}, device=cfg.device)
+    args = {"batch_size": cfg.batch_size, "device": cfg.device, "dict_size": n_dict}
+    return ens, args, f"ratio_{ratio}"
+
+
+def main():
+    """Synthetic sharded-sweep demo:
+
+    torchrun --standalone --nproc-per-node N -m sparse_coding_amd.sweep.sharded_sweep
+    """
+    import argparse
+
+    from sparse_coding_amd.config import SyntheticEnsembleArgs
+
+    p = argparse.ArgumentParser()
+    p.add_argument("--dataset-folder", default="activation_data_sharded")
+    p.add_argument("--output-folder", default="output_sharded")
+    p.add_argument("--n-chunks", type=int, default=4)
+    p.add_argument("--chunk-gb", type=float, default=0.5)
+    p.add_argument("--activation-width", type=int, default=512)
+    args = p.parse_args()
+
+    cfg = SyntheticEnsembleArgs()
+    cfg.use_synthetic_dataset = True
+    cfg.activation_width = args.activation_width
+    cfg.n_ground_truth_components = 2 * args.activation_width
+    cfg.gen_batch_size = 4096
+    cfg.feature_num_nonzero = 32
+    cfg.noise_magnitude_scale = 0.0
+    cfg.chunk_size_gb = args.chunk_gb
+    cfg.n_chunks = args.n_chunks
+    cfg.batch_size = 2048
+    cfg.dataset_folder = args.dataset_folder
+    cfg.output_folder = args.output_folder
+    cfg.use_wandb = False
+    cfg.ensemble_hyperparams = ["dict_size"]
+    cfg.buffer_hyperparams = ["l1_alpha"]
+
+    dicts = sharded_sweep(_demo_init_for_rank, cfg)
+    if dicts is not None:
+        print(f"rank 0: gathered {len(dicts)} learned dicts")
+
+
+if __name__ == "__main__":
+    main()
