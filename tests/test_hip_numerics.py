@@ -557,3 +557,31 @@ def test_semilinear_step_matches_torch():
             err = _rel_err(ens_hip.params["encoder_layers"][li][k],
                            ens_ref.params["encoder_layers"][li][k])
             assert err < 2e-3, (li, k, err)
+
+
+def test_extreme_dict_ratios():
+    """SURVEY.md scale range: dict ratios 0.25x-96x.  The 96x grid at d=512
+    (n=49152) and the 0.25x grid (n=128, one column tile) must both train
+    finite and match the oracle for one step."""
+    from sparse_coding_amd.engine.ensemble import FunctionalEnsemble
+    from sparse_coding_amd.functional.optim import adam
+    from sparse_coding_amd.models.sae_signatures import FunctionalTiedSAE
+
+    for d, ratio, B in ((512, 96, 256), (512, 0.25, 256), (2048, 8, 256)):
+        n = int(d * ratio)
+        torch.manual_seed(23)
+        models = [FunctionalTiedSAE.init(d, n, 1e-3, device=DEV) for _ in range(2)]
+        ens_hip = FunctionalEnsemble(models, FunctionalTiedSAE, adam, {"lr": 1e-3},
+                                     device=DEV, backend="hip")
+        models2 = [({k: v.clone() for k, v in p.items()}, {k: v.clone() for k, v in b.items()})
+                   for p, b in ens_hip.unstack()]
+        ens_ref = FunctionalEnsemble(models2, FunctionalTiedSAE, adam, {"lr": 1e-3},
+                                     device=DEV, backend="torch")
+        x = torch.randn(B, d, device=DEV)
+        l_hip, _ = ens_hip.step_batch(x)
+        l_ref, _ = ens_ref.step_batch(x)
+        assert _rel_err(l_hip["loss"], l_ref["loss"]) < 1e-4, (d, ratio)
+        for k in ens_ref.params:
+            assert _rel_err(ens_hip.params[k], ens_ref.params[k]) < 2e-3, (d, ratio, k)
+        del ens_hip, ens_ref
+        torch.cuda.empty_cache()
