@@ -25,13 +25,13 @@ if _ROOT not in sys.path:
     sys.path.insert(0, _ROOT)
 
 
-def bench_combo(args, staging, bk, prio, bk_gw=None, bk_dec=None):
+def bench_combo(args, staging, bk, prio, bk_gw=None, bk_dec=None, bn=128):
     from sparse_coding_amd.engine.ensemble import FunctionalEnsemble
     from sparse_coding_amd.functional.optim import adam
     from sparse_coding_amd.models.sae_signatures import FunctionalTiedSAE
     from sparse_coding_amd.ops.kconfig import set_kernel_config
 
-    set_kernel_config(staging=staging, bk=bk, prio=prio, bk_grad_w=bk_gw, bk_dec=bk_dec)
+    set_kernel_config(staging=staging, bk=bk, prio=prio, bk_grad_w=bk_gw, bk_dec=bk_dec, bn=bn)
     device = "cuda:0"
     d, n_dict, M, B = args.d_model, args.d_model * args.dict_ratio, args.n_models, args.batch
     torch.manual_seed(0)
@@ -75,6 +75,11 @@ def main():
     # mixed: winner's staging/prio with per-kernel bk for the long-K
     # kernels (grad_w K=B, dec K=n): deep tiles can win there even when
     # occupancy wins the short-K kernels
+    # wide-tile probe (128x256): 4 accumulators/wave, 2 blocks/CU
+    ms, acts = bench_combo(args, best["staging"], 16, best["prio"], bn=256)
+    print(json.dumps({"staging": best["staging"], "bn": 256,
+                      "ms_per_step": round(ms, 4), "acts_per_sec": round(acts)}), flush=True)
+
     for key in ("bk_grad_w", "bk_dec"):
         for bk_k in (16, 32):
             if bk_k == best["bk"]:

@@ -251,6 +251,7 @@ class HipSAEStep:
 
         kc = self.kc
         bk, prio = kc["bk"], kc["prio"]
+        bn = kc.get("bn", 128)
         bk_dec = kc["bk_dec"] or bk
         bk_gw = kc["bk_grad_w"] or bk
 
@@ -274,11 +275,11 @@ class HipSAEStep:
             # transpose-in-staging GEMMs (no separate transpose kernels)
             enc_inv = self.inv_norms if self.tied else None
             mode = 3 if self.reverse else 0
-            ext.enc_fwd(x, enc, bias, enc_inv, self.c, self.loss_parts, self.fired, mode, bk, prio,
+            ext.enc_fwd(x, enc, bias, enc_inv, self.c, self.loss_parts, self.fired, mode, bk, prio, bn,
                         dict_sizes=self.dict_sizes)
-            ext.dec_fwd(self.c, dict_w, self.inv_norms, x, self.r, self.loss_parts, bk_dec, prio)
+            ext.dec_fwd(self.c, dict_w, self.inv_norms, x, self.r, self.loss_parts, bk_dec, prio, bn)
             ext.gc(self.r, dict_w, self.inv_norms, self.c, self.l1_alpha, self.gpre, self.g_bias, bk, prio,
-                   gc_mode=1 if self.reverse else 0)
+                   gc_mode=1 if self.reverse else 0, bn=bn)
 
         if on_grads is not None:
             on_grads([self.g_bias])  # final after k_gc
@@ -293,17 +294,17 @@ class HipSAEStep:
             bounds = [M * i // n_chunks for i in range(n_chunks + 1)]
             for lo, hi in zip(bounds[:-1], bounds[1:]):
                 sl = slice(lo, hi)
-                ext.grad_w(self.c[sl], self.r[sl], self.gw[sl], gscale, 0.0, bk_gw, prio)
-                ext.grad_w(self.gpre[sl], x, self.gw[sl], 1.0, 1.0, bk_gw, prio)
+                ext.grad_w(self.c[sl], self.r[sl], self.gw[sl], gscale, 0.0, bk_gw, prio, bn)
+                ext.grad_w(self.gpre[sl], x, self.gw[sl], 1.0, 1.0, bk_gw, prio, bn)
                 on_grads([self.gw[sl]])
         elif self.tied:
-            ext.grad_w(self.c, self.r, self.gw, gscale, 0.0, bk_gw, prio)
-            ext.grad_w(self.gpre, x, self.gw, 1.0, 1.0, bk_gw, prio)
+            ext.grad_w(self.c, self.r, self.gw, gscale, 0.0, bk_gw, prio, bn)
+            ext.grad_w(self.gpre, x, self.gw, 1.0, 1.0, bk_gw, prio, bn)
         else:
-            ext.grad_w(self.c, self.r, self.gw, gscale, 0.0, bk_gw, prio)
+            ext.grad_w(self.c, self.r, self.gw, gscale, 0.0, bk_gw, prio, bn)
             if on_grads is not None:
                 on_grads([self.gw])
-            ext.grad_w(self.gpre, x, self.gw_enc, 1.0, 0.0, bk_gw, prio)
+            ext.grad_w(self.gpre, x, self.gw_enc, 1.0, 0.0, bk_gw, prio, bn)
             if on_grads is not None:
                 on_grads([self.gw_enc])
         return B

@@ -82,23 +82,23 @@ __device__ __forceinline__ void maybe_prio(int prio) {
 // overlap the previous tile's MFMA phase.  TBK/16 float4 passes per thread.
 // ---------------------------------------------------------------------------
 
-// Transposed tile: lds[k][i] = src[i0+i][k0+k] (* scale[i0+i]), i<128, k<TBK.
-template <int TBK>
+// Transposed tile: lds[k][i] = src[i0+i][k0+k] (* scale[i0+i]), i<TROWS, k<TBK.
+template <int TBK, int TROWS = BM>
 struct TStage {
-  float4 v[TBK / 16];
-  float sc[TBK / 16];
+  float4 v[TBK * TROWS / 2048];
+  float sc[TBK * TROWS / 2048];
 };
 
-template <int TBK>
+template <int TBK, int TROWS = BM>
 __device__ __forceinline__ void stage_T_load(const float* __restrict__ src, long ld,
                                              int i0, int k0, int n_rows, int n_k,
                                              const float* __restrict__ scale,
-                                             TStage<TBK>& st) {
+                                             TStage<TBK, TROWS>& st) {
   const int t = threadIdx.x;
   constexpr int TPR = TBK / 4;                 // threads covering one row's K
   constexpr int RPP = NTHREADS / TPR;          // rows per pass
 #pragma unroll
-  for (int p = 0; p < TBK / 16; ++p) {
+  for (int p = 0; p < TBK * TROWS / 2048; ++p) {
     int i = p * RPP + t / TPR;
     int kc = (t % TPR) * 4;
     int gi = i0 + i;
@@ -120,14 +120,14 @@ __device__ __forceinline__ void stage_T_load(const float* __restrict__ src, long
   }
 }
 
-template <int TBK>
-__device__ __forceinline__ void stage_T_write(const TStage<TBK>& st, float* __restrict__ lds,
+template <int TBK, int TROWS = BM>
+__device__ __forceinline__ void stage_T_write(const TStage<TBK, TROWS>& st, float* __restrict__ lds,
                                               bool scaled) {
   const int t = threadIdx.x;
   constexpr int TPR = TBK / 4;
   constexpr int RPP = NTHREADS / TPR;
 #pragma unroll
-  for (int p = 0; p < TBK / 16; ++p) {
+  for (int p = 0; p < TBK * TROWS / 2048; ++p) {
     int i = p * RPP + t / TPR;
     int kc = (t % TPR) * 4;
     float4 v = st.v[p];
@@ -135,27 +135,29 @@ __device__ __forceinline__ void stage_T_write(const TStage<TBK>& st, float* __re
       v.x *= st.sc[p]; v.y *= st.sc[p]; v.z *= st.sc[p]; v.w *= st.sc[p];
     }
 #pragma unroll
-    for (int s = 0; s < 4; ++s) lds[(kc + s) * BMP + i] = ((float*)&v)[s];
+    for (int s = 0; s < 4; ++s) lds[(kc + s) * (TROWS + 1) + i] = ((float*)&v)[s];
   }
 }
 
-// Direct tile: lds[k][j] = src[k0+k][j0+j] (* scale[k0+k]), k<TBK, j<128.
-template <int TBK>
+// Direct tile: lds[k][j] = src[k0+k][j0+j] (* scale[k0+k]), k<TBK, j<TCOLS.
+template <int TBK, int TCOLS = BM>
 struct DStage {
-  float4 v[TBK / 16];
-  float sc[TBK / 16];
+  float4 v[TBK * TCOLS / 2048];
+  float sc[TBK * TCOLS / 2048];
 };
 
-template <int TBK>
+template <int TBK, int TCOLS = BM>
 __device__ __forceinline__ void stage_D_load(const float* __restrict__ src, long ld,
                                              int k0, int j0, int n_k, int n_cols,
                                              const float* __restrict__ scale,
-                                             DStage<TBK>& st) {
+                                             DStage<TBK, TCOLS>& st) {
   const int t = threadIdx.x;
+  constexpr int TPC = TCOLS / 4;               // threads covering one k's cols
+  constexpr int KPP = NTHREADS / TPC;          // k rows per pass
 #pragma unroll
-  for (int p = 0; p < TBK / 16; ++p) {
-    int k = p * 16 + t / 32;
-    int j = (t % 32) * 4;
+  for (int p = 0; p < TBK * TCOLS / 2048; ++p) {
+    int k = p * KPP + t / TPC;
+    int j = (t % TPC) * 4;
     int gk = k0 + k;
     float4 v = make_float4(0.f, 0.f, 0.f, 0.f);
     float sc = 1.f;
@@ -175,49 +177,53 @@ __device__ __forceinline__ void stage_D_load(const float* __restrict__ src, long
   }
 }
 
-template <int TBK>
-__device__ __forceinline__ void stage_D_write(const DStage<TBK>& st, float* __restrict__ lds,
+template <int TBK, int TCOLS = BM>
+__device__ __forceinline__ void stage_D_write(const DStage<TBK, TCOLS>& st, float* __restrict__ lds,
                                               bool scaled) {
   const int t = threadIdx.x;
+  constexpr int TPC = TCOLS / 4;
+  constexpr int KPP = NTHREADS / TPC;
 #pragma unroll
-  for (int p = 0; p < TBK / 16; ++p) {
-    int k = p * 16 + t / 32;
-    int j = (t % 32) * 4;
+  for (int p = 0; p < TBK * TCOLS / 2048; ++p) {
+    int k = p * KPP + t / TPC;
+    int j = (t % TPC) * 4;
     float4 v = st.v[p];
     if (scaled) {
       v.x *= st.sc[p]; v.y *= st.sc[p]; v.z *= st.sc[p]; v.w *= st.sc[p];
     }
-    *reinterpret_cast<float4*>(&lds[k * BM + j]) = v;
+    *reinterpret_cast<float4*>(&lds[k * TCOLS + j]) = v;
   }
 }
 
 // ---------------------------------------------------------------------------
 // MFMA phase: 8 waves; wave w covers rows [(w&3)*32, +32), cols [(w>>2)*64, +64)
 // ---------------------------------------------------------------------------
-template <int TBK, int ASTRIDE, int BSTRIDE>
+template <int TBK, int NACC, int ASTRIDE, int BSTRIDE>
 __device__ __forceinline__ void mfma_tile(const float* __restrict__ As,
                                           const float* __restrict__ Bs,
-                                          f32x16 acc[2]) {
+                                          f32x16 acc[NACC]) {
   const int lane = threadIdx.x & (WAVE - 1);
   const int wave = threadIdx.x / WAVE;
   const int wr = (wave & 3) * 32;
-  const int wc = (wave >> 2) * 64;
+  const int wc = (wave >> 2) * (NACC * 32);
   const int l31 = lane & 31;
   const int h = lane >> 5;
 
 #pragma unroll
   for (int kk = 0; kk < TBK; kk += 2) {
     float a0 = As[(kk + h) * ASTRIDE + wr + l31];
-    float b0 = Bs[(kk + h) * BSTRIDE + wc + l31];
-    float b1 = Bs[(kk + h) * BSTRIDE + wc + 32 + l31];
-    acc[0] = __builtin_amdgcn_mfma_f32_32x32x2f32(a0, b0, acc[0], 0, 0, 0);
-    acc[1] = __builtin_amdgcn_mfma_f32_32x32x2f32(a0, b1, acc[1], 0, 0, 0);
+#pragma unroll
+    for (int j = 0; j < NACC; ++j) {
+      float b = Bs[(kk + h) * BSTRIDE + wc + j * 32 + l31];
+      acc[j] = __builtin_amdgcn_mfma_f32_32x32x2f32(a0, b, acc[j], 0, 0, 0);
+    }
   }
 }
 
-__device__ __forceinline__ void zero_acc(f32x16 acc[2]) {
+template <int NACC>
+__device__ __forceinline__ void zero_acc(f32x16 acc[NACC]) {
 #pragma unroll
-  for (int j = 0; j < 2; ++j)
+  for (int j = 0; j < NACC; ++j)
 #pragma unroll
     for (int r = 0; r < 16; ++r) acc[j][r] = 0.f;
 }
@@ -232,12 +238,13 @@ __device__ __forceinline__ float wave_reduce_sum(float v) {
 struct EpiGeom {
   int lane, wave, wr, wc, l31;
 };
+template <int NACC = 2>
 __device__ __forceinline__ EpiGeom epi_geom() {
   EpiGeom g;
   g.lane = threadIdx.x & (WAVE - 1);
   g.wave = threadIdx.x / WAVE;
   g.wr = (g.wave & 3) * 32;
-  g.wc = (g.wave >> 2) * 64;
+  g.wc = (g.wave >> 2) * (NACC * 32);
   g.l31 = g.lane & 31;
   return g;
 }
@@ -290,13 +297,13 @@ extern "C" __global__ void k_row_norms(const float* __restrict__ W,
     for (int k0 = (TBK); k0 < (K_TOTAL); k0 += (TBK)) {                        \
       LOAD_A(k0);                                                              \
       LOAD_B(k0);                                                              \
-      mfma_tile<TBK, ASTR, BSTR>(&As[cur][0], &Bs[cur][0], acc);               \
+      mfma_tile<TBK, NACC, ASTR, BSTR>(&As[cur][0], &Bs[cur][0], acc);         \
       WRITE_A(cur ^ 1);                                                        \
       WRITE_B(cur ^ 1);                                                        \
       __syncthreads();                                                         \
       cur ^= 1;                                                                \
     }                                                                          \
-    mfma_tile<TBK, ASTR, BSTR>(&As[cur][0], &Bs[cur][0], acc);                 \
+    mfma_tile<TBK, NACC, ASTR, BSTR>(&As[cur][0], &Bs[cur][0], acc);           \
   }
 
 // The soft-threshold gate of the thresholding SAE (reference
@@ -317,7 +324,7 @@ __device__ __forceinline__ float gate_gp(float u) {  // dg/du (0 at the kinks)
 // ---------------------------------------------------------------------------
 // k_enc_fwd_t
 // ---------------------------------------------------------------------------
-template <int TBK, int MINW>
+template <int TBK, int MINW, int TBN = BN>
 __global__ __launch_bounds__(NTHREADS, MINW)
 void k_enc_fwd_t(const float* __restrict__ x,       // [B, d]
                  const float* __restrict__ Wenc,    // [M, n, d]
@@ -341,41 +348,43 @@ void k_enc_fwd_t(const float* __restrict__ x,       // [B, d]
                  float* __restrict__ x_out,        // [M, B, n] (mode 4)
                  const float* __restrict__ mom) {  // [M] momentum (mode 4)
   __shared__ float As[2][TBK * BMP];
-  __shared__ float Bs[2][TBK * BMP];
+  __shared__ float Bs[2][TBK * (TBN + 1)];
 
   maybe_prio(prio);
   const int m = blockIdx.z;
   int tx, ty;
   tile_coords(tx, ty);
   const int row0 = ty * BM;
-  const int col0 = tx * BN;
+  const int col0 = tx * TBN;
   const float* W = Wenc + (long)m * n * d;
   const float* x_m = x + (long)m * x_mstride;
   const float* inv = inv_norms ? inv_norms + (long)m * n : nullptr;
   const bool scaled = inv != nullptr;
 
-  f32x16 acc[2];
-  zero_acc(acc);
-  TStage<TBK> sa, sb;
+  constexpr int NACC = TBN / 64;
+  f32x16 acc[NACC];
+  zero_acc<NACC>(acc);
+  TStage<TBK> sa;
+  TStage<TBK, TBN> sb;
 
 #define ENC_LA(K) stage_T_load<TBK>(x_m, d, row0, (K), B, d, nullptr, sa)
-#define ENC_LB(K) stage_T_load<TBK>(W, d, col0, (K), n, d, inv, sb)
+#define ENC_LB(K) stage_T_load<TBK, TBN>(W, d, col0, (K), n, d, inv, sb)
 #define ENC_WA(BUF) stage_T_write<TBK>(sa, &As[BUF][0], false)
-#define ENC_WB(BUF) stage_T_write<TBK>(sb, &Bs[BUF][0], scaled)
-  PREFETCH_LOOP(TBK, d, ENC_LA, ENC_LB, ENC_WA, ENC_WB, BMP, BMP)
+#define ENC_WB(BUF) stage_T_write<TBK, TBN>(sb, &Bs[BUF][0], scaled)
+  PREFETCH_LOOP(TBK, d, ENC_LA, ENC_LB, ENC_WA, ENC_WB, BMP, TBN + 1)
 #undef ENC_LA
 #undef ENC_LB
 #undef ENC_WA
 #undef ENC_WB
 
-  const EpiGeom g = epi_geom();
+  const EpiGeom g = epi_geom<NACC>();
   float* c_m = c_out + (long)m * B * n;
   const float* bias_m = bias + (long)m * n;
   float* fired_m = fired + (long)m * n;
 
   float l1_sum = 0.f;
 #pragma unroll
-  for (int tj = 0; tj < 2; ++tj) {
+  for (int tj = 0; tj < NACC; ++tj) {
     int col = col0 + g.wc + tj * 32 + g.l31;
     bool col_ok = col < n;
     if (mode == 1) {
@@ -504,7 +513,7 @@ void k_enc_fwd_t(const float* __restrict__ x,       // [B, d]
 // ---------------------------------------------------------------------------
 // k_dec_fwd_t
 // ---------------------------------------------------------------------------
-template <int TBK, int MINW>
+template <int TBK, int MINW, int TBN = BN>
 __global__ __launch_bounds__(NTHREADS, MINW)
 void k_dec_fwd_t(const float* __restrict__ c,       // [M, B, n]
                  const float* __restrict__ Wdec,    // [M, n, d]
@@ -515,40 +524,41 @@ void k_dec_fwd_t(const float* __restrict__ c,       // [M, B, n]
                  int B, int d, int n, int prio,
                  long x_mstride) {  // 0: shared [B,d]; else [M,B,d]
   __shared__ float As[2][TBK * BMP];
-  __shared__ float Bs[2][TBK * BM];
+  __shared__ float Bs[2][TBK * TBN];
 
   maybe_prio(prio);
   const int m = blockIdx.z;
   int tx, ty;
   tile_coords(tx, ty);
   const int row0 = ty * BM;
-  const int col0 = tx * BN;
+  const int col0 = tx * TBN;
   const float* c_m = c + (long)m * B * n;
   const float* W = Wdec + (long)m * n * d;
   const float* x_m = x + (long)m * x_mstride;
   const float* inv = inv_norms + (long)m * n;
 
-  f32x16 acc[2];
-  zero_acc(acc);
+  constexpr int NACC = TBN / 64;
+  f32x16 acc[NACC];
+  zero_acc<NACC>(acc);
   TStage<TBK> sa;
-  DStage<TBK> sb;
+  DStage<TBK, TBN> sb;
 
 #define DEC_LA(K) stage_T_load<TBK>(c_m, n, row0, (K), B, n, nullptr, sa)
-#define DEC_LB(K) stage_D_load<TBK>(W, d, (K), col0, n, d, inv, sb)
+#define DEC_LB(K) stage_D_load<TBK, TBN>(W, d, (K), col0, n, d, inv, sb)
 #define DEC_WA(BUF) stage_T_write<TBK>(sa, &As[BUF][0], false)
-#define DEC_WB(BUF) stage_D_write<TBK>(sb, &Bs[BUF][0], true)
-  PREFETCH_LOOP(TBK, n, DEC_LA, DEC_LB, DEC_WA, DEC_WB, BMP, BM)
+#define DEC_WB(BUF) stage_D_write<TBK, TBN>(sb, &Bs[BUF][0], true)
+  PREFETCH_LOOP(TBK, n, DEC_LA, DEC_LB, DEC_WA, DEC_WB, BMP, TBN)
 #undef DEC_LA
 #undef DEC_LB
 #undef DEC_WA
 #undef DEC_WB
 
-  const EpiGeom g = epi_geom();
+  const EpiGeom g = epi_geom<NACC>();
   float* r_m = r_out + (long)m * B * d;
 
   float mse_sum = 0.f;
 #pragma unroll
-  for (int tj = 0; tj < 2; ++tj) {
+  for (int tj = 0; tj < NACC; ++tj) {
     int col = col0 + g.wc + tj * 32 + g.l31;
     bool col_ok = col < d;
 #pragma unroll
@@ -568,7 +578,7 @@ void k_dec_fwd_t(const float* __restrict__ c,       // [M, B, n]
 // ---------------------------------------------------------------------------
 // k_gc_t
 // ---------------------------------------------------------------------------
-template <int TBK, int MINW>
+template <int TBK, int MINW, int TBN = BN>
 __global__ __launch_bounds__(NTHREADS, MINW)
 void k_gc_t(const float* __restrict__ r,        // [M, B, d]
             const float* __restrict__ Wdec,     // [M, n, d]
@@ -579,14 +589,14 @@ void k_gc_t(const float* __restrict__ r,        // [M, B, d]
             float* __restrict__ g_bias,         // [M, n]
             int B, int d, int n, int prio, int gc_mode) {
   __shared__ float As[2][TBK * BMP];
-  __shared__ float Bs[2][TBK * BMP];
+  __shared__ float Bs[2][TBK * (TBN + 1)];
 
   maybe_prio(prio);
   const int m = blockIdx.z;
   int tx, ty;
   tile_coords(tx, ty);
   const int row0 = ty * BM;
-  const int col0 = tx * BN;
+  const int col0 = tx * TBN;
   const float* r_m = r + (long)m * B * d;
   const float* W = Wdec + (long)m * n * d;
   const float* inv = inv_norms + (long)m * n;
@@ -594,26 +604,28 @@ void k_gc_t(const float* __restrict__ r,        // [M, B, d]
   const float gscale = 2.0f / ((float)B * (float)d);
   const float l1_term = l1_alpha[m] / (float)B;
 
-  f32x16 acc[2];
-  zero_acc(acc);
-  TStage<TBK> sa, sb;
+  constexpr int NACC = TBN / 64;
+  f32x16 acc[NACC];
+  zero_acc<NACC>(acc);
+  TStage<TBK> sa;
+  TStage<TBK, TBN> sb;
 
 #define GC_LA(K) stage_T_load<TBK>(r_m, d, row0, (K), B, d, nullptr, sa)
-#define GC_LB(K) stage_T_load<TBK>(W, d, col0, (K), n, d, inv, sb)
+#define GC_LB(K) stage_T_load<TBK, TBN>(W, d, col0, (K), n, d, inv, sb)
 #define GC_WA(BUF) stage_T_write<TBK>(sa, &As[BUF][0], false)
-#define GC_WB(BUF) stage_T_write<TBK>(sb, &Bs[BUF][0], true)
-  PREFETCH_LOOP(TBK, d, GC_LA, GC_LB, GC_WA, GC_WB, BMP, BMP)
+#define GC_WB(BUF) stage_T_write<TBK, TBN>(sb, &Bs[BUF][0], true)
+  PREFETCH_LOOP(TBK, d, GC_LA, GC_LB, GC_WA, GC_WB, BMP, TBN + 1)
 #undef GC_LA
 #undef GC_LB
 #undef GC_WA
 #undef GC_WB
 
-  const EpiGeom g = epi_geom();
+  const EpiGeom g = epi_geom<NACC>();
   float* g_m = gpre_out + (long)m * B * n;
   float* gb_m = g_bias + (long)m * n;
 
 #pragma unroll
-  for (int tj = 0; tj < 2; ++tj) {
+  for (int tj = 0; tj < NACC; ++tj) {
     int col = col0 + g.wc + tj * 32 + g.l31;
     bool col_ok = col < n;
     float colsum = 0.f;
@@ -651,7 +663,7 @@ void k_gc_t(const float* __restrict__ r,        // [M, B, d]
 //   g_scale= sum_b dL/da = gv * 2a * (g(u) - [a^2>eps] g'(u) u a^2/max(..))
 // with gv = gscale*acc + l1/B * [code>0].
 // ---------------------------------------------------------------------------
-template <int TBK, int MINW>
+template <int TBK, int MINW, int TBN = BN>
 __global__ __launch_bounds__(NTHREADS, MINW)
 void k_gc_thresh_t(const float* __restrict__ r,        // [M, B, d]
                    const float* __restrict__ Wdec,     // [M, n, d]
@@ -665,14 +677,14 @@ void k_gc_thresh_t(const float* __restrict__ r,        // [M, B, d]
                    float* __restrict__ g_scale,        // [M, n]
                    int B, int d, int n, int prio) {
   __shared__ float As[2][TBK * BMP];
-  __shared__ float Bs[2][TBK * BMP];
+  __shared__ float Bs[2][TBK * (TBN + 1)];
 
   maybe_prio(prio);
   const int m = blockIdx.z;
   int tx, ty;
   tile_coords(tx, ty);
   const int row0 = ty * BM;
-  const int col0 = tx * BN;
+  const int col0 = tx * TBN;
   const float* r_m = r + (long)m * B * d;
   const float* W = Wdec + (long)m * n * d;
   const float* inv = inv_norms + (long)m * n;
@@ -681,27 +693,29 @@ void k_gc_thresh_t(const float* __restrict__ r,        // [M, B, d]
   const float gscale = 2.0f / ((float)B * (float)d);
   const float l1_term = l1_alpha[m] / (float)B;
 
-  f32x16 acc[2];
-  zero_acc(acc);
-  TStage<TBK> sa, sb;
+  constexpr int NACC = TBN / 64;
+  f32x16 acc[NACC];
+  zero_acc<NACC>(acc);
+  TStage<TBK> sa;
+  TStage<TBK, TBN> sb;
 
 #define GCT_LA(K) stage_T_load<TBK>(r_m, d, row0, (K), B, d, nullptr, sa)
-#define GCT_LB(K) stage_T_load<TBK>(W, d, col0, (K), n, d, inv, sb)
+#define GCT_LB(K) stage_T_load<TBK, TBN>(W, d, col0, (K), n, d, inv, sb)
 #define GCT_WA(BUF) stage_T_write<TBK>(sa, &As[BUF][0], false)
-#define GCT_WB(BUF) stage_T_write<TBK>(sb, &Bs[BUF][0], true)
-  PREFETCH_LOOP(TBK, d, GCT_LA, GCT_LB, GCT_WA, GCT_WB, BMP, BMP)
+#define GCT_WB(BUF) stage_T_write<TBK, TBN>(sb, &Bs[BUF][0], true)
+  PREFETCH_LOOP(TBK, d, GCT_LA, GCT_LB, GCT_WA, GCT_WB, BMP, TBN + 1)
 #undef GCT_LA
 #undef GCT_LB
 #undef GCT_WA
 #undef GCT_WB
 
-  const EpiGeom g = epi_geom();
+  const EpiGeom g = epi_geom<NACC>();
   float* g_m = gpre_out + (long)m * B * n;
   float* gg_m = g_gain + (long)m * n;
   float* gs_m = g_scale + (long)m * n;
 
 #pragma unroll
-  for (int tj = 0; tj < 2; ++tj) {
+  for (int tj = 0; tj < NACC; ++tj) {
     int col = col0 + g.wc + tj * 32 + g.l31;
     bool col_ok = col < n;
     float a = col_ok ? act_scale[(long)m * n + col] : 1.f;
@@ -886,7 +900,7 @@ void k_topk_select(const float* __restrict__ scores,  // [M, B, n]
 // ---------------------------------------------------------------------------
 // k_grad_w_t
 // ---------------------------------------------------------------------------
-template <int TBK, int MINW>
+template <int TBK, int MINW, int TBN = BN>
 __global__ __launch_bounds__(NTHREADS, MINW)
 void k_grad_w_t(const float* __restrict__ P, long p_batch_stride,
                 const float* __restrict__ Q, long q_batch_stride,
@@ -894,36 +908,38 @@ void k_grad_w_t(const float* __restrict__ P, long p_batch_stride,
                 float alpha, float beta,
                 int B, int n, int d, int prio) {
   __shared__ float As[2][TBK * BM];
-  __shared__ float Bs[2][TBK * BM];
+  __shared__ float Bs[2][TBK * TBN];
 
   maybe_prio(prio);
   const int m = blockIdx.z;
   int tx, ty;
   tile_coords(tx, ty);
   const int row0 = ty * BM;  // n rows
-  const int col0 = tx * BN;  // d cols
+  const int col0 = tx * TBN;  // d cols
   const float* P_m = P + (long)m * p_batch_stride;
   const float* Q_m = Q + (long)m * q_batch_stride;
 
-  f32x16 acc[2];
-  zero_acc(acc);
-  DStage<TBK> sa, sb;
+  constexpr int NACC = TBN / 64;
+  f32x16 acc[NACC];
+  zero_acc<NACC>(acc);
+  DStage<TBK> sa;
+  DStage<TBK, TBN> sb;
 
 #define GW_LA(K) stage_D_load<TBK>(P_m, n, (K), row0, B, n, nullptr, sa)
-#define GW_LB(K) stage_D_load<TBK>(Q_m, d, (K), col0, B, d, nullptr, sb)
+#define GW_LB(K) stage_D_load<TBK, TBN>(Q_m, d, (K), col0, B, d, nullptr, sb)
 #define GW_WA(BUF) stage_D_write<TBK>(sa, &As[BUF][0], false)
-#define GW_WB(BUF) stage_D_write<TBK>(sb, &Bs[BUF][0], false)
-  PREFETCH_LOOP(TBK, B, GW_LA, GW_LB, GW_WA, GW_WB, BM, BM)
+#define GW_WB(BUF) stage_D_write<TBK, TBN>(sb, &Bs[BUF][0], false)
+  PREFETCH_LOOP(TBK, B, GW_LA, GW_LB, GW_WA, GW_WB, BM, TBN)
 #undef GW_LA
 #undef GW_LB
 #undef GW_WA
 #undef GW_WB
 
-  const EpiGeom g = epi_geom();
+  const EpiGeom g = epi_geom<NACC>();
   float* gw_m = gw + (long)m * n * d;
 
 #pragma unroll
-  for (int tj = 0; tj < 2; ++tj) {
+  for (int tj = 0; tj < NACC; ++tj) {
     int col = col0 + g.wc + tj * 32 + g.l31;
     bool col_ok = col < d;
 #pragma unroll
@@ -1095,7 +1111,7 @@ void k_transpose_scale(const float* __restrict__ src, float* __restrict__ dst,
 // k_enc_fwd2_t: enc forward with PRE-TRANSPOSED operands (xT [d,B], WT [M,d,n],
 // already inv-norm-scaled for tied) — both tiles stage DIRECT (b128 writes).
 // ---------------------------------------------------------------------------
-template <int TBK, int MINW>
+template <int TBK, int MINW, int TBN = BN>
 __global__ __launch_bounds__(NTHREADS, MINW)
 void k_enc_fwd2_t(const float* __restrict__ xT,      // [d, B]
                   const float* __restrict__ WT,      // [M, d, n]
@@ -1106,38 +1122,40 @@ void k_enc_fwd2_t(const float* __restrict__ xT,      // [d, B]
                   int B, int d, int n, int mode, int prio,
                   const int* __restrict__ dict_sizes) {
   __shared__ float As[2][TBK * BM];
-  __shared__ float Bs[2][TBK * BM];
+  __shared__ float Bs[2][TBK * TBN];
 
   maybe_prio(prio);
   const int m = blockIdx.z;
   int tx, ty;
   tile_coords(tx, ty);
   const int row0 = ty * BM;   // batch rows
-  const int col0 = tx * BN;   // dict cols
+  const int col0 = tx * TBN;   // dict cols
   const float* WT_m = WT + (long)m * d * n;
 
-  f32x16 acc[2];
-  zero_acc(acc);
-  DStage<TBK> sa, sb;
+  constexpr int NACC = TBN / 64;
+  f32x16 acc[NACC];
+  zero_acc<NACC>(acc);
+  DStage<TBK> sa;
+  DStage<TBK, TBN> sb;
 
 #define ENC2_LA(K) stage_D_load<TBK>(xT, B, (K), row0, d, B, nullptr, sa)
-#define ENC2_LB(K) stage_D_load<TBK>(WT_m, n, (K), col0, d, n, nullptr, sb)
+#define ENC2_LB(K) stage_D_load<TBK, TBN>(WT_m, n, (K), col0, d, n, nullptr, sb)
 #define ENC2_WA(BUF) stage_D_write<TBK>(sa, &As[BUF][0], false)
-#define ENC2_WB(BUF) stage_D_write<TBK>(sb, &Bs[BUF][0], false)
-  PREFETCH_LOOP(TBK, d, ENC2_LA, ENC2_LB, ENC2_WA, ENC2_WB, BM, BM)
+#define ENC2_WB(BUF) stage_D_write<TBK, TBN>(sb, &Bs[BUF][0], false)
+  PREFETCH_LOOP(TBK, d, ENC2_LA, ENC2_LB, ENC2_WA, ENC2_WB, BM, TBN)
 #undef ENC2_LA
 #undef ENC2_LB
 #undef ENC2_WA
 #undef ENC2_WB
 
-  const EpiGeom g = epi_geom();
+  const EpiGeom g = epi_geom<NACC>();
   float* c_m = c_out + (long)m * B * n;
   const float* bias_m = bias + (long)m * n;
   float* fired_m = fired + (long)m * n;
 
   float l1_sum = 0.f;
 #pragma unroll
-  for (int tj = 0; tj < 2; ++tj) {
+  for (int tj = 0; tj < NACC; ++tj) {
     int col = col0 + g.wc + tj * 32 + g.l31;
     bool col_ok = col < n;
     if (mode == 1) {
@@ -1176,7 +1194,7 @@ void k_enc_fwd2_t(const float* __restrict__ xT,      // [d, B]
 // ---------------------------------------------------------------------------
 // k_gc2_t: code-grad with pre-transposed rT [M,d,B] and WT (scaled) [M,d,n].
 // ---------------------------------------------------------------------------
-template <int TBK, int MINW>
+template <int TBK, int MINW, int TBN = BN>
 __global__ __launch_bounds__(NTHREADS, MINW)
 void k_gc2_t(const float* __restrict__ rT,       // [M, d, B]
              const float* __restrict__ WT,       // [M, d, n] (inv-norm scaled)
@@ -1186,40 +1204,42 @@ void k_gc2_t(const float* __restrict__ rT,       // [M, d, B]
              float* __restrict__ g_bias,         // [M, n]
              int B, int d, int n, int prio, int gc_mode) {
   __shared__ float As[2][TBK * BM];
-  __shared__ float Bs[2][TBK * BM];
+  __shared__ float Bs[2][TBK * TBN];
 
   maybe_prio(prio);
   const int m = blockIdx.z;
   int tx, ty;
   tile_coords(tx, ty);
   const int row0 = ty * BM;
-  const int col0 = tx * BN;
+  const int col0 = tx * TBN;
   const float* rT_m = rT + (long)m * d * B;
   const float* WT_m = WT + (long)m * d * n;
   const float* c_m = c + (long)m * B * n;
   const float gscale = 2.0f / ((float)B * (float)d);
   const float l1_term = l1_alpha[m] / (float)B;
 
-  f32x16 acc[2];
-  zero_acc(acc);
-  DStage<TBK> sa, sb;
+  constexpr int NACC = TBN / 64;
+  f32x16 acc[NACC];
+  zero_acc<NACC>(acc);
+  DStage<TBK> sa;
+  DStage<TBK, TBN> sb;
 
 #define GC2_LA(K) stage_D_load<TBK>(rT_m, B, (K), row0, d, B, nullptr, sa)
-#define GC2_LB(K) stage_D_load<TBK>(WT_m, n, (K), col0, d, n, nullptr, sb)
+#define GC2_LB(K) stage_D_load<TBK, TBN>(WT_m, n, (K), col0, d, n, nullptr, sb)
 #define GC2_WA(BUF) stage_D_write<TBK>(sa, &As[BUF][0], false)
-#define GC2_WB(BUF) stage_D_write<TBK>(sb, &Bs[BUF][0], false)
-  PREFETCH_LOOP(TBK, d, GC2_LA, GC2_LB, GC2_WA, GC2_WB, BM, BM)
+#define GC2_WB(BUF) stage_D_write<TBK, TBN>(sb, &Bs[BUF][0], false)
+  PREFETCH_LOOP(TBK, d, GC2_LA, GC2_LB, GC2_WA, GC2_WB, BM, TBN)
 #undef GC2_LA
 #undef GC2_LB
 #undef GC2_WA
 #undef GC2_WB
 
-  const EpiGeom g = epi_geom();
+  const EpiGeom g = epi_geom<NACC>();
   float* g_m = gpre_out + (long)m * B * n;
   float* gb_m = g_bias + (long)m * n;
 
 #pragma unroll
-  for (int tj = 0; tj < 2; ++tj) {
+  for (int tj = 0; tj < NACC; ++tj) {
     int col = col0 + g.wc + tj * 32 + g.l31;
     bool col_ok = col < n;
     float colsum = 0.f;

@@ -32,6 +32,7 @@ void enc_fwd(torch::Tensor x, torch::Tensor Wenc, torch::Tensor bias,
              c10::optional<torch::Tensor> inv_norms, torch::Tensor c_out,
              torch::Tensor loss_parts, torch::Tensor fired, int64_t mode,
              int64_t bk, bool prio,
+             int64_t bn,
              c10::optional<torch::Tensor> act_scale,
              c10::optional<torch::Tensor> act_gain,
              c10::optional<torch::Tensor> u_out,
@@ -90,8 +91,15 @@ void enc_fwd(torch::Tensor x, torch::Tensor Wenc, torch::Tensor bias,
       u_p = u_out->data_ptr<float>();
     }
   }
-  dim3 grid(cdiv(n, BN), cdiv(B, BM), M);
-  if (bk == 16)
+  dim3 grid(cdiv(n, bn == 256 ? 256 : BN), cdiv(B, BM), M);
+  if (bn == 256)
+    hipLaunchKernelGGL((k_enc_fwd_t<16, 4, 256>), grid, dim3(NTHREADS), 0, cur_stream(),
+                       x.data_ptr<float>(), Wenc.data_ptr<float>(),
+                       bias.data_ptr<float>(), inv, c_out.data_ptr<float>(),
+                       loss_parts.data_ptr<float>(), fired.data_ptr<float>(),
+                       B, d, n, (int)mode, prio ? 1 : 0, a_p, gn_p, u_p, ds_p, x_mstride,
+                       yin_p, xp_p, xo_p, mom_p);
+  else if (bk == 16)
     hipLaunchKernelGGL((k_enc_fwd_t<16, 6>), grid, dim3(NTHREADS), 0, cur_stream(),
                        x.data_ptr<float>(), Wenc.data_ptr<float>(),
                        bias.data_ptr<float>(), inv, c_out.data_ptr<float>(),
@@ -109,7 +117,7 @@ void enc_fwd(torch::Tensor x, torch::Tensor Wenc, torch::Tensor bias,
 
 void dec_fwd(torch::Tensor c, torch::Tensor Wdec, torch::Tensor inv_norms,
              torch::Tensor x, torch::Tensor r_out, torch::Tensor loss_parts,
-             int64_t bk, bool prio) {
+             int64_t bk, bool prio, int64_t bn) {
   CHECK_IN(c); CHECK_IN(Wdec); CHECK_IN(inv_norms); CHECK_IN(x);
   CHECK_IN(r_out); CHECK_IN(loss_parts);
   int M = Wdec.size(0), n = Wdec.size(1), d = Wdec.size(2);
@@ -119,8 +127,14 @@ void dec_fwd(torch::Tensor c, torch::Tensor Wdec, torch::Tensor inv_norms,
     x_mstride = (long)x.size(1) * x.size(2);
   }
   int B = x.size(x.dim() - 2);
-  dim3 grid(cdiv(d, BN), cdiv(B, BM), M);
-  if (bk == 16)
+  dim3 grid(cdiv(d, bn == 256 ? 256 : BN), cdiv(B, BM), M);
+  if (bn == 256)
+    hipLaunchKernelGGL((k_dec_fwd_t<16, 4, 256>), grid, dim3(NTHREADS), 0, cur_stream(),
+                       c.data_ptr<float>(), Wdec.data_ptr<float>(),
+                       inv_norms.data_ptr<float>(), x.data_ptr<float>(),
+                       r_out.data_ptr<float>(), loss_parts.data_ptr<float>(),
+                       B, d, n, prio ? 1 : 0, x_mstride);
+  else if (bk == 16)
     hipLaunchKernelGGL((k_dec_fwd_t<16, 6>), grid, dim3(NTHREADS), 0, cur_stream(),
                        c.data_ptr<float>(), Wdec.data_ptr<float>(),
                        inv_norms.data_ptr<float>(), x.data_ptr<float>(),
@@ -136,13 +150,19 @@ void dec_fwd(torch::Tensor c, torch::Tensor Wdec, torch::Tensor inv_norms,
 
 void gc(torch::Tensor r, torch::Tensor Wdec, torch::Tensor inv_norms,
         torch::Tensor c, torch::Tensor l1_alpha, torch::Tensor gpre,
-        torch::Tensor g_bias, int64_t bk, bool prio, int64_t gc_mode) {
+        torch::Tensor g_bias, int64_t bk, bool prio, int64_t gc_mode, int64_t bn) {
   CHECK_IN(r); CHECK_IN(Wdec); CHECK_IN(inv_norms); CHECK_IN(c);
   CHECK_IN(l1_alpha); CHECK_IN(gpre); CHECK_IN(g_bias);
   int M = Wdec.size(0), n = Wdec.size(1), d = Wdec.size(2);
   int B = r.size(1);
-  dim3 grid(cdiv(n, BN), cdiv(B, BM), M);
-  if (bk == 16)
+  dim3 grid(cdiv(n, bn == 256 ? 256 : BN), cdiv(B, BM), M);
+  if (bn == 256)
+    hipLaunchKernelGGL((k_gc_t<16, 4, 256>), grid, dim3(NTHREADS), 0, cur_stream(),
+                       r.data_ptr<float>(), Wdec.data_ptr<float>(),
+                       inv_norms.data_ptr<float>(), c.data_ptr<float>(),
+                       l1_alpha.data_ptr<float>(), gpre.data_ptr<float>(),
+                       g_bias.data_ptr<float>(), B, d, n, prio ? 1 : 0, (int)gc_mode);
+  else if (bk == 16)
     hipLaunchKernelGGL((k_gc_t<16, 6>), grid, dim3(NTHREADS), 0, cur_stream(),
                        r.data_ptr<float>(), Wdec.data_ptr<float>(),
                        inv_norms.data_ptr<float>(), c.data_ptr<float>(),
@@ -187,7 +207,7 @@ void gc_thresh(torch::Tensor r, torch::Tensor Wdec, torch::Tensor inv_norms,
 
 // gw[m] = beta * gw[m] + alpha * P[m]^T @ Q[m]; Q may be rank-shared [B, d]
 void grad_w(torch::Tensor P, torch::Tensor Q, torch::Tensor gw,
-            double alpha, double beta, int64_t bk, bool prio) {
+            double alpha, double beta, int64_t bk, bool prio, int64_t bn) {
   CHECK_IN(P); CHECK_IN(Q); CHECK_IN(gw);
   int M = gw.size(0), n = gw.size(1), d = gw.size(2);
   int B;
@@ -200,8 +220,13 @@ void grad_w(torch::Tensor P, torch::Tensor Q, torch::Tensor gw,
   } else {
     q_stride = 0;  // shared across models
   }
-  dim3 grid(cdiv(d, BN), cdiv(n, BM), M);
-  if (bk == 16)
+  dim3 grid(cdiv(d, bn == 256 ? 256 : BN), cdiv(n, BM), M);
+  if (bn == 256)
+    hipLaunchKernelGGL((k_grad_w_t<16, 4, 256>), grid, dim3(NTHREADS), 0, cur_stream(),
+                       P.data_ptr<float>(), p_stride, Q.data_ptr<float>(),
+                       q_stride, gw.data_ptr<float>(), (float)alpha,
+                       (float)beta, B, n, d, prio ? 1 : 0);
+  else if (bk == 16)
     hipLaunchKernelGGL((k_grad_w_t<16, 6>), grid, dim3(NTHREADS), 0, cur_stream(),
                        P.data_ptr<float>(), p_stride, Q.data_ptr<float>(),
                        q_stride, gw.data_ptr<float>(), (float)alpha,
@@ -374,7 +399,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("enc_fwd", &enc_fwd, "fused encoder GEMM + bias + ReLU/TopK/gate (+L1, fired)",
         py::arg("x"), py::arg("Wenc"), py::arg("bias"), py::arg("inv_norms"),
         py::arg("c_out"), py::arg("loss_parts"), py::arg("fired"), py::arg("mode"),
-        py::arg("bk") = 32, py::arg("prio") = false,
+        py::arg("bk") = 32, py::arg("prio") = false, py::arg("bn") = 128,
         py::arg("act_scale") = py::none(), py::arg("act_gain") = py::none(),
         py::arg("u_out") = py::none(), py::arg("dict_sizes") = py::none(),
         py::arg("y_in") = py::none(), py::arg("x_prev") = py::none(),
@@ -393,14 +418,15 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("dec_fwd", &dec_fwd, "fused decoder GEMM - x (+MSE partial)",
         py::arg("c"), py::arg("Wdec"), py::arg("inv_norms"), py::arg("x"),
         py::arg("r_out"), py::arg("loss_parts"),
-        py::arg("bk") = 32, py::arg("prio") = false);
+        py::arg("bk") = 32, py::arg("prio") = false, py::arg("bn") = 128);
   m.def("gc", &gc, "code-gradient GEMM + relu/reverse mask + l1 term (+bias grad)",
         py::arg("r"), py::arg("Wdec"), py::arg("inv_norms"), py::arg("c"),
         py::arg("l1_alpha"), py::arg("gpre"), py::arg("g_bias"),
-        py::arg("bk") = 32, py::arg("prio") = false, py::arg("gc_mode") = 0);
+        py::arg("bk") = 32, py::arg("prio") = false, py::arg("gc_mode") = 0,
+        py::arg("bn") = 128);
   m.def("grad_w", &grad_w, "gw = beta*gw + alpha * P^T Q (batched over M)",
         py::arg("P"), py::arg("Q"), py::arg("gw"), py::arg("alpha"), py::arg("beta"),
-        py::arg("bk") = 32, py::arg("prio") = false);
+        py::arg("bk") = 32, py::arg("prio") = false, py::arg("bn") = 128);
   m.def("project_adam", &project_adam, "renorm-gradient projection + Adam",
         py::arg("W"), py::arg("gw"), py::arg("norms"), py::arg("mu"),
         py::arg("nu"), py::arg("step_no"), py::arg("n_per_model"),

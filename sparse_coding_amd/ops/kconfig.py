@@ -31,6 +31,11 @@ DEFAULTS = {
     "bk": 16,
     "prio": False,
     "staging": "t",
+    # output-tile width of the GEMM kernels: 128 (default) or 256
+    # (128x256 tiles: 4 accumulators/wave, 2 blocks/CU — measured slower
+    # than bk16's 4-block co-residency on the flagship shape, kept as a
+    # variant; see profiles/README.md)
+    "bn": 128,
     # per-kernel overrides, None -> use "bk".  dec_fwd reduces over the
     # dictionary (K = n, long): the deeper TBK=32 tile wins there even
     # though TBK=16's occupancy wins the short-K kernels (643.6k vs 639.0k
@@ -43,6 +48,8 @@ _cfg = dict(DEFAULTS)
 
 if os.environ.get("SC_AMD_BK"):
     _cfg["bk"] = int(os.environ["SC_AMD_BK"])
+if os.environ.get("SC_AMD_BN"):
+    _cfg["bn"] = int(os.environ["SC_AMD_BN"])
 if os.environ.get("SC_AMD_PRIO"):
     _cfg["prio"] = os.environ["SC_AMD_PRIO"] == "1"
 if os.environ.get("SC_AMD_STAGING"):
@@ -63,3 +70,4 @@ def set_kernel_config(**kwargs) -> None:
             raise KeyError(f"unknown kernel-config key {k!r}; valid: {sorted(_cfg)}")
         _cfg[k] = v
     assert _cfg["bk"] in (16, 32) and _cfg["staging"] in ("pre", "t")
+    assert _cfg["bn"] in (128, 256)

@@ -33,16 +33,17 @@ def shapes():
 
 # every test in this module runs under each compiled kernel variant:
 # (staging, tile depth TBK, younger-half s_setprio) — see ops/kconfig.py
-@pytest.fixture(autouse=True, params=[("t", 32, False), ("t", 16, True),
-                                      ("pre", 32, True), ("pre", 16, False)],
-                ids=lambda p: f"{p[0]}-bk{p[1]}-p{int(p[2])}")
+@pytest.fixture(autouse=True, params=[("t", 32, False, 128), ("t", 16, True, 128),
+                                      ("pre", 32, True, 128), ("pre", 16, False, 128),
+                                      ("t", 16, False, 256)],
+                ids=lambda p: f"{p[0]}-bk{p[1]}-p{int(p[2])}-bn{p[3]}")
 def kcfg(request):
     from sparse_coding_amd.ops.kconfig import kernel_config, set_kernel_config
 
     old = kernel_config()
-    staging, bk, prio = request.param
-    set_kernel_config(staging=staging, bk=bk, prio=prio)
-    yield {"bk": bk, "prio": prio, "staging": staging}
+    staging, bk, prio, bn = request.param
+    set_kernel_config(staging=staging, bk=bk, prio=prio, bn=bn)
+    yield {"bk": bk, "prio": prio, "staging": staging, "bn": bn}
     set_kernel_config(**old)
 
 
@@ -71,7 +72,7 @@ def test_enc_fwd_untied(shapes, kcfg):
     c = torch.empty(M, B, n, device=DEV)
     loss_parts = torch.zeros(M, 2, device=DEV)
     fired = torch.zeros(M, n, device=DEV)
-    ext.enc_fwd(x, W, bias, None, c, loss_parts, fired, 0, kcfg["bk"], kcfg["prio"])
+    ext.enc_fwd(x, W, bias, None, c, loss_parts, fired, 0, kcfg["bk"], kcfg["prio"], kcfg["bn"])
     ref = torch.clamp(torch.einsum("mnd,bd->mbn", W, x) + bias[:, None, :], min=0)
     assert _rel_err(c, ref) < 1e-5
     assert _rel_err(loss_parts[:, 1], ref.sum(dim=(1, 2))) < 1e-4
@@ -92,7 +93,7 @@ def test_enc_fwd_tied_scaled(shapes, kcfg):
     c = torch.empty(M, B, n, device=DEV)
     lp = torch.zeros(M, 2, device=DEV)
     fired = torch.zeros(M, n, device=DEV)
-    ext.enc_fwd(x, W, bias, inv, c, lp, fired, 0, kcfg["bk"], kcfg["prio"])
+    ext.enc_fwd(x, W, bias, inv, c, lp, fired, 0, kcfg["bk"], kcfg["prio"], kcfg["bn"])
     What = W / torch.clamp(torch.norm(W, dim=-1, keepdim=True), 1e-8)
     ref = torch.clamp(torch.einsum("mnd,bd->mbn", What, x), min=0)
     assert _rel_err(c, ref) < 1e-5
@@ -110,7 +111,7 @@ def test_dec_fwd(shapes, kcfg):
     ext.row_norms(W, norms, inv, 1e-8)
     r = torch.empty(M, B, d, device=DEV)
     lp = torch.zeros(M, 2, device=DEV)
-    ext.dec_fwd(c, W, inv, x, r, lp, kcfg["bk"], kcfg["prio"])
+    ext.dec_fwd(c, W, inv, x, r, lp, kcfg["bk"], kcfg["prio"], kcfg["bn"])
     What = W / torch.clamp(torch.norm(W, dim=-1, keepdim=True), 1e-8)
     ref_r = torch.einsum("mnd,mbn->mbd", What, c) - x
     assert _rel_err(r, ref_r) < 1e-5
@@ -130,7 +131,7 @@ def test_gc(shapes, kcfg):
     ext.row_norms(W, norms, inv, 1e-8)
     gpre = torch.empty(M, B, n, device=DEV)
     g_bias = torch.zeros(M, n, device=DEV)
-    ext.gc(r, W, inv, c, l1, gpre, g_bias, kcfg["bk"], kcfg["prio"])
+    ext.gc(r, W, inv, c, l1, gpre, g_bias, kcfg["bk"], kcfg["prio"], bn=kcfg["bn"])
     What = W / torch.clamp(torch.norm(W, dim=-1, keepdim=True), 1e-8)
     gscale = 2.0 / (B * d)
     g = gscale * torch.einsum("mnd,mbd->mbn", What, r) + l1[:, None, None] / B
@@ -146,12 +147,12 @@ def test_grad_w(shapes, kcfg):
     P = torch.randn(M, B, n, device=DEV)
     Q = torch.randn(M, B, d, device=DEV)
     gw = torch.zeros(M, n, d, device=DEV)
-    ext.grad_w(P, Q, gw, 0.5, 0.0, kcfg["bk"], kcfg["prio"])
+    ext.grad_w(P, Q, gw, 0.5, 0.0, kcfg["bk"], kcfg["prio"], kcfg["bn"])
     ref = 0.5 * torch.einsum("mbn,mbd->mnd", P, Q)
     assert _rel_err(gw, ref) < 1e-5
     # beta accumulate + shared Q
     x = torch.randn(B, d, device=DEV)
-    ext.grad_w(P, x, gw, 1.0, 1.0, kcfg["bk"], kcfg["prio"])
+    ext.grad_w(P, x, gw, 1.0, 1.0, kcfg["bk"], kcfg["prio"], kcfg["bn"])
     ref = ref + torch.einsum("mbn,bd->mnd", P, x)
     assert _rel_err(gw, ref) < 1e-5
 
