@@ -579,7 +579,8 @@ class HipThresholdStep(HipSAEStep):
 
         ext.row_norms(enc, self.norms, self.inv_norms, EPS_NORM)
         ext.enc_fwd(x, enc, self.dummy_bias, self.inv_norms, self.c,
-                    self.loss_parts, self.fired, 2, bk, prio, a, gain, self.u)
+                    self.loss_parts, self.fired, 2, bk, prio,
+                    act_scale=a, act_gain=gain, u_out=self.u)
         ext.dec_fwd(self.c, enc, self.inv_norms, x, self.r, self.loss_parts,
                     self.kc["bk_dec"] or bk, prio)
         ext.gc_thresh(self.r, enc, self.inv_norms, self.c, self.u, a,
