@@ -310,10 +310,17 @@ def make_activation_dataset_hf(
 
     def drain(l: int):
         # NOTE: the capture hook holds a reference to stores[l]; mutate it in
-        # place, never rebind it.
+        # place, never rebind it.  D2H lands in PINNED host buffers so the
+        # non_blocking copies actually overlap the next forward (a pageable
+        # destination would serialize).
         while stores[l]:
             t = stores[l].pop(0).to(dtype)
-            host_parts[l].append(t.to("cpu", non_blocking=True))
+            if t.is_cuda:
+                host = torch.empty(t.shape, dtype=t.dtype, pin_memory=True)
+                host.copy_(t, non_blocking=True)
+            else:
+                host = t
+            host_parts[l].append(host)
             host_counts[l] += t.shape[0]
 
     def flush_chunk(l: int, final: bool = False):
