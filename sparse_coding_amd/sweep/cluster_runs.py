@@ -76,7 +76,11 @@ def dispatch_job_on_chunk(ensembles: List[Tuple[Any, dict, str]], cfg, chunk: to
         done_flags.append(done)
         progress_counters.append(counter)
 
-    show_progress = bool(getattr(cfg, "show_progress", True))
+    import sys as _sys
+
+    # carriage-return progress only on a real terminal; plain logs get one
+    # line per chunk instead of thousands of \r frames
+    show_progress = bool(getattr(cfg, "show_progress", True)) and _sys.stdout.isatty()
     total = sum(n_batches)
     while not all(f.value == 1 for f in done_flags):
         # a crashed child must not hang the sweep (reference busy-polls
