@@ -593,3 +593,41 @@ def test_extreme_dict_ratios():
             assert torch.isfinite(ens_hip.params[k]).all(), (d, ratio, k)
         del ens_hip, ens_ref
         torch.cuda.empty_cache()
+
+
+@pytest.mark.parametrize("case", range(10))
+def test_shape_fuzz_gemm_kernels(case, kcfg):
+    """Seeded random shapes (d,n multiples of 4; arbitrary B, M) through
+    enc/dec/gc/grad_w vs einsum — edge-guard fuzzing beyond the two fixed
+    odd shapes."""
+    ext = _ext()
+    g = torch.Generator().manual_seed(1000 + case)
+
+    def r(lo, hi, mult=1):
+        return int(torch.randint(lo, hi, (1,), generator=g)) * mult
+
+    M, B, d, n = r(1, 5), r(3, 70), r(1, 80, 4), r(1, 90, 4)
+    torch.manual_seed(case)
+    x = torch.randn(B, d, device=DEV)
+    W = torch.randn(M, n, d, device=DEV) * 0.3
+    bias = torch.randn(M, n, device=DEV) * 0.1
+    c = torch.empty(M, B, n, device=DEV)
+    lp = torch.zeros(M, 2, device=DEV)
+    fired = torch.zeros(M, n, device=DEV)
+    ext.enc_fwd(x, W, bias, None, c, lp, fired, 0, kcfg["bk"], kcfg["prio"], kcfg["bn"])
+    ref_c = torch.clamp(torch.einsum("mnd,bd->mbn", W, x) + bias[:, None, :], min=0)
+    assert _rel_err(c, ref_c) < 1e-4, (M, B, d, n)
+
+    norms = torch.empty(M, n, device=DEV)
+    inv = torch.empty(M, n, device=DEV)
+    ext.row_norms(W, norms, inv, 1e-8)
+    rr = torch.empty(M, B, d, device=DEV)
+    ext.dec_fwd(c, W, inv, x, rr, lp, kcfg["bk"], kcfg["prio"], kcfg["bn"])
+    What = W / torch.clamp(torch.norm(W, dim=-1, keepdim=True), 1e-8)
+    ref_r = torch.einsum("mnd,mbn->mbd", What, ref_c) - x
+    assert _rel_err(rr, ref_r) < 1e-4, (M, B, d, n)
+
+    gw = torch.zeros(M, n, d, device=DEV)
+    ext.grad_w(c, rr, gw, 1.0, 0.0, kcfg["bk"], kcfg["prio"], kcfg["bn"])
+    ref_gw = torch.einsum("mbn,mbd->mnd", ref_c, ref_r)
+    assert _rel_err(gw, ref_gw) < 1e-4, (M, B, d, n)
