@@ -88,6 +88,12 @@ class EnsembleResampler:
         st = ens.optim_states
         wk = self.w_key
         M, n, d = p[wk].shape
+
+        if p[wk].is_cuda:
+            from sparse_coding_amd import ops as _ops
+
+            ext = _ops.get_extension(required=True)
+            return self._resample_fused(ext)
         counts = torch.zeros(M, dtype=torch.long)
 
         for m in range(M):
@@ -116,3 +122,36 @@ class EnsembleResampler:
         self.fired.zero_()
         self.worst_losses.fill_(-float("inf"))
         return counts
+
+    @torch.no_grad()
+    def _resample_fused(self, ext) -> torch.Tensor:
+        """K14 fully on device (ops k_resample): one launch ranks the dead
+        features per model in index order, rewrites rows from the worst-
+        example pool, and zeroes the Adam slices — no host loop or sync."""
+        ens = self.ens
+        p = ens.params
+        st = ens.optim_states
+        wk = self.w_key
+        M, n, d = p[wk].shape
+        dev = p[wk].device
+
+        worst_unit = self.worst_examples / torch.clamp(
+            torch.norm(self.worst_examples, dim=-1, keepdim=True), 1e-8)
+        avg_norm = torch.norm(p[wk], dim=-1).mean(dim=1)  # [M]
+        enc_scale = (self.encoder_norm_ratio * avg_norm).contiguous()
+        counts = torch.zeros(M, device=dev, dtype=torch.int32)
+
+        kwargs = {}
+        if "decoder" in p:
+            kwargs = dict(dec=p["decoder"], mu_d=st["mu"]["decoder"], nu_d=st["nu"]["decoder"])
+        bkw = {}
+        if "encoder_bias" in p:
+            bkw = dict(bias=p["encoder_bias"], mu_b=st["mu"]["encoder_bias"],
+                       nu_b=st["nu"]["encoder_bias"])
+        ext.resample(self.fired, worst_unit.contiguous(), enc_scale,
+                     p[wk], st["mu"][wk], st["nu"][wk],
+                     counts_out=counts, **kwargs, **bkw)
+
+        self.fired.zero_()
+        self.worst_losses.fill_(-float("inf"))
+        return counts.long().cpu()

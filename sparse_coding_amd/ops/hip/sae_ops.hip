@@ -362,6 +362,41 @@ void topk_select(torch::Tensor scores, torch::Tensor c_out, torch::Tensor fired,
                      fired.data_ptr<float>(), ks.data_ptr<int>(), B, n);
 }
 
+void resample(torch::Tensor fired, torch::Tensor new_rows, torch::Tensor enc_scale,
+              torch::Tensor W, torch::Tensor mu_w, torch::Tensor nu_w,
+              c10::optional<torch::Tensor> dec,
+              c10::optional<torch::Tensor> mu_d, c10::optional<torch::Tensor> nu_d,
+              c10::optional<torch::Tensor> bias,
+              c10::optional<torch::Tensor> mu_b, c10::optional<torch::Tensor> nu_b,
+              torch::Tensor counts_out) {
+  CHECK_IN(fired); CHECK_IN(new_rows); CHECK_IN(enc_scale);
+  CHECK_IN(W); CHECK_IN(mu_w); CHECK_IN(nu_w);
+  TORCH_CHECK(counts_out.is_cuda() && counts_out.scalar_type() == torch::kInt32,
+              "counts_out must be int32 GPU");
+  int M = W.size(0), n = W.size(1), d = W.size(2);
+  int n_track = new_rows.size(1);
+  float* dec_p = nullptr; float* mud_p = nullptr; float* nud_p = nullptr;
+  if (dec.has_value()) {
+    CHECK_IN(dec.value()); CHECK_IN(mu_d.value()); CHECK_IN(nu_d.value());
+    dec_p = dec->data_ptr<float>();
+    mud_p = mu_d->data_ptr<float>();
+    nud_p = nu_d->data_ptr<float>();
+  }
+  float* b_p = nullptr; float* mub_p = nullptr; float* nub_p = nullptr;
+  if (bias.has_value()) {
+    CHECK_IN(bias.value()); CHECK_IN(mu_b.value()); CHECK_IN(nu_b.value());
+    b_p = bias->data_ptr<float>();
+    mub_p = mu_b->data_ptr<float>();
+    nub_p = nu_b->data_ptr<float>();
+  }
+  hipLaunchKernelGGL(k_resample, dim3(M), dim3(RSMP_T), 0, cur_stream(),
+                     fired.data_ptr<float>(), new_rows.data_ptr<float>(),
+                     enc_scale.data_ptr<float>(), W.data_ptr<float>(),
+                     mu_w.data_ptr<float>(), nu_w.data_ptr<float>(),
+                     dec_p, mud_p, nud_p, b_p, mub_p, nub_p,
+                     counts_out.data_ptr<int>(), n, d, n_track);
+}
+
 void lista_bwd_elem(torch::Tensor g_y, c10::optional<torch::Tensor> carry_in,
                     torch::Tensor r, torch::Tensor theta, torch::Tensor x,
                     torch::Tensor x_prev, torch::Tensor mom,
@@ -406,6 +441,12 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("x_out") = py::none(), py::arg("mom") = py::none());
   m.def("topk_select", &topk_select, "per-row radix top-k + scatter (+fired)",
         py::arg("scores"), py::arg("c_out"), py::arg("fired"), py::arg("ks"));
+  m.def("resample", &resample, "fused dead-neuron resample (K14): rank dead, rewrite rows, zero Adam state",
+        py::arg("fired"), py::arg("new_rows"), py::arg("enc_scale"),
+        py::arg("W"), py::arg("mu_w"), py::arg("nu_w"),
+        py::arg("dec") = py::none(), py::arg("mu_d") = py::none(), py::arg("nu_d") = py::none(),
+        py::arg("bias") = py::none(), py::arg("mu_b") = py::none(), py::arg("nu_b") = py::none(),
+        py::arg("counts_out"));
   m.def("lista_bwd_elem", &lista_bwd_elem, "fused LISTA backward elementwise pass",
         py::arg("g_y"), py::arg("carry_in"), py::arg("r"), py::arg("theta"),
         py::arg("x"), py::arg("x_prev"), py::arg("mom"), py::arg("g_r"),
