@@ -90,5 +90,6 @@ def test_mfma_count_matches_tiling(disasm):
 
 def test_small_kernels_present(disasm):
     for stem in ("k_row_norms", "k_project_adam", "k_bias_adam",
-                 "k_transpose_scale", "k_lista_bwd_elem"):
+                 "k_transpose_scale", "k_lista_bwd_elem", "k_resample",
+                 "k_topk_select"):
         assert _find(disasm, stem), f"{stem} missing from code object"
