@@ -322,3 +322,24 @@ def test_sharded_sweep_end_to_end(tmp_path):
         p.join(timeout=30)
     assert "_error" not in res, res.get("_error")
     assert res["n"] == 4 and res["ckpt"]
+
+
+@pytest.mark.timeout(120)
+def test_dispatch_lite_collect():
+    """dispatch_lite/collect_lite (reference cluster_runs.py:50-97): async
+    single-ensemble dispatch, join via collect."""
+    from types import SimpleNamespace
+
+    from sparse_coding_amd.sweep.big_sweep import ensemble_train_loop
+    from sparse_coding_amd.sweep.cluster_runs import collect_lite, dispatch_lite
+
+    ens = _make_ensemble(seed=3)
+    before = ens.params["encoder"].clone()
+    chunk = torch.randn(256, D)
+    cfg = SimpleNamespace(batch_size=64, show_progress=False, logger=None,
+                          ensemble_hyperparams=[], buffer_hyperparams=["l1_alpha"],
+                          log_every=1000)
+    handle = dispatch_lite(cfg, chunk, ens, "lite", ensemble_train_loop)
+    collect_lite([handle])
+    assert not torch.allclose(ens.params["encoder"], before)
+    assert torch.isfinite(ens.params["encoder"]).all()
