@@ -167,3 +167,22 @@ def test_sweep_resume_noop(tmp_path):
     (ld, _), = dicts
     (ld2, _), = dicts2
     assert torch.equal(ld.get_learned_dict(), ld2.get_learned_dict())
+
+
+def test_sweep_image_metrics(tmp_path):
+    """The periodic image-metrics path (reference big_sweep.py:86-156):
+    MMCS grids across dict sizes + sparsity histograms land as PNGs in the
+    run folder (wandb-free)."""
+    cfg = _mini_cfg(tmp_path, 1)
+    cfg.wandb_images = True
+
+    def init_func(c):
+        return make_grid_ensembles(c, FunctionalTiedSAE, [1e-4, 1e-3], [1.0, 2.0], devices=["cpu"])
+
+    dicts = big_sweep.sweep(init_func, cfg)
+    assert len(dicts) == 4  # 2 l1 x 2 dict sizes
+    img_dir = os.path.join(cfg.output_folder, "images")
+    assert os.path.isdir(img_dir)
+    names = os.listdir(img_dir)
+    assert any("sparsity_hist" in n for n in names)
+    assert any("mmcs_grid" in n for n in names)
