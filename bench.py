@@ -40,6 +40,9 @@ def parse_args():
     p.add_argument("--n-models", type=int, default=8, help="L1-ensemble size")
     p.add_argument("--tied", action="store_true", default=True)
     p.add_argument("--untied", dest="tied", action="store_false")
+    p.add_argument("--sig", choices=["sae", "topk", "thresholding"], default="sae",
+                   help="model family: sae (tied/untied per --tied; the BASELINE default), "
+                        "topk (k=32 + dead-neuron resampling, config 5), thresholding")
     p.add_argument("--backend", choices=["auto", "hip", "torch"], default="auto")
     p.add_argument("--dtype", choices=["fp32"], default="fp32",
                    help="compute dtype; fp32 matches the reference (fp32 params, BASELINE.md)")
@@ -93,13 +96,32 @@ def main():
     n_pool = 8
     data_pool = [gen.send(None).contiguous() for _ in range(n_pool)]
 
-    sig = FunctionalTiedSAE if args.tied else FunctionalSAE
     l1_values = np.logspace(-4, -2, M)
     # build on the target device directly (288 GB HBM: everything stays resident)
     torch.manual_seed(1234)  # identical replicas on every rank
-    models = [sig.init(d, n_dict, float(l1), device=device) for l1 in l1_values]
-    ensemble = FunctionalEnsemble(models, sig, adam, {"lr": 1e-3}, device=device, backend=args.backend)
+    no_stacking = False
+    resampler = None
+    if args.sig == "topk":
+        from sparse_coding_amd.models.topk import TopKEncoder
+
+        sig = TopKEncoder
+        models = [TopKEncoder.init(d, n_dict, 32) for _ in range(M)]
+        no_stacking = not use_cuda  # vmap cannot stack data-dependent topk
+    elif args.sig == "thresholding":
+        from sparse_coding_amd.models.sae_signatures import FunctionalThresholdingSAE
+
+        sig = FunctionalThresholdingSAE
+        models = [sig.init(d, n_dict, float(l1), device=device) for l1 in l1_values]
+    else:
+        sig = FunctionalTiedSAE if args.tied else FunctionalSAE
+        models = [sig.init(d, n_dict, float(l1), device=device) for l1 in l1_values]
+    ensemble = FunctionalEnsemble(models, sig, adam, {"lr": 1e-3}, device=device,
+                                  backend=args.backend, no_stacking=no_stacking)
     trainer = DataParallelEnsembleTrainer(ensemble)
+    if args.sig == "topk":
+        from sparse_coding_amd.engine.resample import EnsembleResampler
+
+        resampler = EnsembleResampler(ensemble, n_track=256)
 
     def barrier_sync():
         if distributed:
@@ -110,12 +132,20 @@ def main():
             torch.cuda.synchronize()
 
     for i in range(args.warmup):
-        trainer.step(data_pool[i % n_pool])
+        losses, aux = trainer.step(data_pool[i % n_pool])
+        if resampler is not None:
+            resampler.observe(data_pool[i % n_pool], aux)
 
     barrier_sync()
     t0 = time.perf_counter()
     for i in range(args.steps):
-        trainer.step(data_pool[i % n_pool])
+        losses, aux = trainer.step(data_pool[i % n_pool])
+        if resampler is not None:
+            # config 5 names resampling: tracking runs every step, the
+            # (rare) resample itself once mid-run
+            resampler.observe(data_pool[i % n_pool], aux)
+            if i == args.steps // 2:
+                resampler.resample()
     barrier_sync()
     elapsed = time.perf_counter() - t0
 
@@ -145,6 +175,7 @@ def main():
             "data": "synthetic",
             "config": {
                 "model": "pythia-70m-resid-l2",
+                "sig": args.sig,
                 "d_model": d,
                 "dict_size": n_dict,
                 "n_models": M,
