@@ -114,7 +114,7 @@ class DataParallelEnsembleTrainer:
     broadcast."""
 
     def __init__(self, ensemble, bucket_bytes: int = 64 << 20, group=None,
-                 force_dp_path: bool = False):
+                 force_dp_path: bool = False, graph_capture: Optional[bool] = None):
         self.ensemble = ensemble
         self.group = group
         self._reducer: Optional[GradBucketAllReducer] = None
@@ -124,6 +124,16 @@ class DataParallelEnsembleTrainer:
         # size 1 — lets a single-GPU box exercise the exact RCCL call
         # sequence the 8-GPU scaling run uses (tests/test_full_stack_gpu.py)
         self.force_dp_path = force_dp_path
+        # opt-in hipGraph capture of the WHOLE multi-GPU step including the
+        # RCCL collectives (RCCL supports capture); default off until the
+        # 8-GPU behavior is measured (docs/ROADMAP.md item 2).  Falls back
+        # to eager on capture failure.
+        if graph_capture is None:
+            graph_capture = os.environ.get("SC_AMD_DP_GRAPH") == "1"
+        self.graph_capture = graph_capture
+        self._graph = None
+        self._graph_B = None
+        self._dp_steps = 0
 
     def broadcast_state(self) -> None:
         """One-time parameter/optimizer broadcast from rank 0 (use when
@@ -165,15 +175,44 @@ class DataParallelEnsembleTrainer:
                         works.append((dist.all_reduce(t, async_op=True, group=self.group), t))
 
             use_dp = self.world_size > 1 or self.force_dp_path
-            B = hs.grads_phase(local_batch, on_grads=on_grads if use_dp else None)
-            if use_dp:
-                for w, t in works:
-                    w.wait()
-                if comm_stream is not None:
-                    torch.cuda.current_stream().wait_stream(comm_stream)
-                for _, t in works:
-                    t.div_(self.world_size)
-            hs.update_phase(B)
+
+            def dp_step(x):
+                works.clear()
+                B = hs.grads_phase(x, on_grads=on_grads if use_dp else None)
+                if use_dp:
+                    for w, t in works:
+                        w.wait()
+                    if comm_stream is not None:
+                        torch.cuda.current_stream().wait_stream(comm_stream)
+                    for _, t in works:
+                        t.div_(self.world_size)
+                hs.update_phase(B)
+                return B
+
+            if self.graph_capture and torch.cuda.is_available():
+                B = local_batch.shape[0]
+                if self._graph is not None and self._graph_B == B:
+                    self._x_static.copy_(local_batch)
+                    self._graph.replay()
+                    return hs._loss_data(B), {"c": hs.c}
+                if self._dp_steps >= 2 and self._graph is None:
+                    try:
+                        self._x_static = local_batch.contiguous().clone()
+                        torch.cuda.synchronize()
+                        g = torch.cuda.CUDAGraph()
+                        with torch.cuda.graph(g):
+                            dp_step(self._x_static)
+                        self._graph = g
+                        self._graph_B = B
+                        self._graph.replay()  # capture does not execute
+                        return hs._loss_data(B), {"c": hs.c}
+                    except Exception as e:  # noqa: BLE001 - optimization only
+                        print(f"[dp] graph capture failed ({e}); staying eager")
+                        self.graph_capture = False
+                        self._graph = None
+
+            B = dp_step(local_batch)
+            self._dp_steps += 1
             return hs._loss_data(B), {"c": hs.c}
 
         grads, (loss_data, aux) = self.ensemble.compute_grads(local_batch)

@@ -449,3 +449,43 @@ def test_chunk_broadcast_rccl_one_rank():
         assert torch.allclose(out.cpu(), chunk)
     finally:
         dist.destroy_process_group()
+
+
+def test_dp_rccl_graph_capture_one_rank():
+    """Opt-in hipGraph capture of the whole DP step INCLUDING the RCCL
+    all-reduces (ROADMAP item 2), validated at world size 1: captured
+    replays must keep matching the plain fused step."""
+    import os
+
+    import torch.distributed as dist
+
+    from sparse_coding_amd.engine.ensemble import FunctionalEnsemble
+    from sparse_coding_amd.functional.optim import adam
+    from sparse_coding_amd.models.sae_signatures import FunctionalTiedSAE
+    from sparse_coding_amd.parallel.dp import DataParallelEnsembleTrainer
+
+    os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+    os.environ.setdefault("MASTER_PORT", "29571")
+    dist.init_process_group("nccl", rank=0, world_size=1)
+    try:
+        torch.manual_seed(33)
+        M, B, d, n = 4, 1024, 128, 512
+        models = [FunctionalTiedSAE.init(d, n, 1e-3, device=DEV) for _ in range(M)]
+        ens_dp = FunctionalEnsemble(models, FunctionalTiedSAE, adam, {"lr": 1e-3}, device=DEV, backend="hip")
+        models2 = [({k: v.clone() for k, v in p.items()}, {k: v.clone() for k, v in b.items()})
+                   for p, b in ens_dp.unstack()]
+        ens_ref = FunctionalEnsemble(models2, FunctionalTiedSAE, adam, {"lr": 1e-3}, device=DEV, backend="hip")
+
+        trainer = DataParallelEnsembleTrainer(ens_dp, force_dp_path=True, graph_capture=True)
+        x = torch.randn(B, d, device=DEV)
+        for i in range(6):  # 2 eager + capture + replays
+            losses, _ = trainer.step(x)
+            ens_ref.step_batch(x)
+        torch.cuda.synchronize()
+        assert trainer._graph is not None, "graph capture did not engage"
+        assert torch.isfinite(losses["loss"]).all()
+        for k in ens_ref.params:
+            err = (ens_dp.params[k] - ens_ref.params[k]).abs().max().item()
+            assert err < 1e-6, (k, err)
+    finally:
+        dist.destroy_process_group()
