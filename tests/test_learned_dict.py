@@ -153,3 +153,47 @@ def test_pickle_attribute_parity_with_reference():
                 if missing:
                     problems.append((cls, sorted(missing)))
     assert not problems, problems
+
+
+def test_reference_saved_checkpoint_simulation():
+    """Load a checkpoint AS THE REFERENCE WOULD HAVE SAVED IT: objects built
+    via __new__ with exactly the reference's attribute sets (pickle never
+    calls __init__), then encode/decode/predict must work."""
+    import autoencoders.learned_dict as ald
+
+    d, n = 8, 16
+    enc = torch.randn(n, d)
+    bias = torch.zeros(n)
+
+    # reference TiedSAE attrs: encoder, encoder_bias, norm_encoder,
+    # n_feats, activation_size, center_rot/scale/trans
+    obj = ald.TiedSAE.__new__(ald.TiedSAE)
+    obj.__dict__.update(dict(
+        encoder=enc, encoder_bias=bias, norm_encoder=True,
+        n_feats=n, activation_size=d,
+        center_rot=torch.eye(d), center_scale=torch.ones(d),
+        center_trans=torch.zeros(d),
+    ))
+    buf = io.BytesIO()
+    torch.save([(obj, {"l1_alpha": 1e-3, "dict_size": n})], buf)
+    buf.seek(0)
+    loaded = torch.load(buf, weights_only=False)
+    ld, hp = loaded[0]
+    x = torch.randn(4, d)
+    c = ld.encode(ld.center(x))
+    assert c.shape == (4, n) and (c >= 0).all()
+    x_hat = ld.predict(x)
+    assert x_hat.shape == x.shape and torch.isfinite(x_hat).all()
+    D = ld.get_learned_dict()
+    assert torch.allclose(torch.norm(D, dim=-1), torch.ones(n), atol=1e-5)
+
+    # UntiedSAE: encoder, decoder, encoder_bias, n_feats, activation_size
+    dec = torch.randn(n, d)
+    obj2 = ald.UntiedSAE.__new__(ald.UntiedSAE)
+    obj2.__dict__.update(dict(encoder=enc, decoder=dec, encoder_bias=bias,
+                              n_feats=n, activation_size=d))
+    buf = io.BytesIO()
+    torch.save(obj2, buf)
+    buf.seek(0)
+    ld2 = torch.load(buf, weights_only=False)
+    assert torch.isfinite(ld2.predict(x)).all()
