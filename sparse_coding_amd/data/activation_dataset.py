@@ -272,6 +272,53 @@ def chunk_and_tokenize(dataset, tokenizer, text_key: str = "text", max_length: i
 # capture loop
 # ---------------------------------------------------------------------------
 
+
+class _PinnedStager:
+    """FIFO ring of reusable pinned D2H staging buffers.
+
+    ``stage(t, out)`` async-copies a CUDA tensor into a pinned slot; when the
+    ring is full the oldest slot is retired first (event sync + copy to a
+    pageable tensor appended to ``out``), so ``out`` receives tensors in
+    staging order and pinned memory stays bounded at ~``depth`` batches.
+    """
+
+    def __init__(self, depth: int = 4):
+        import collections
+
+        self.depth = depth
+        self.inflight = collections.deque()  # (flat pinned buf, event, shape)
+        self.free: List[torch.Tensor] = []
+
+    def stage(self, t: torch.Tensor, out: List[torch.Tensor]) -> None:
+        if len(self.inflight) >= self.depth:
+            out.append(self._retire())
+        buf = None
+        for i, f in enumerate(self.free):
+            if f.numel() >= t.numel() and f.dtype == t.dtype:
+                buf = self.free.pop(i)
+                break
+        if buf is None:
+            buf = torch.empty(t.numel(), dtype=t.dtype, pin_memory=True)
+        view = buf[: t.numel()].view(t.shape)
+        view.copy_(t, non_blocking=True)
+        ev = torch.cuda.Event()
+        ev.record()
+        self.inflight.append((buf, ev, t.shape))
+
+    def _retire(self) -> torch.Tensor:
+        import math
+
+        buf, ev, shape = self.inflight.popleft()
+        ev.synchronize()
+        out = buf[: math.prod(shape)].view(shape).clone()  # pageable copy
+        self.free.append(buf)
+        return out
+
+    def retire_all(self, out: List[torch.Tensor]) -> None:
+        while self.inflight:
+            out.append(self._retire())
+
+
 def make_activation_dataset_hf(
     token_batches: Iterable[torch.Tensor],
     model,
@@ -299,6 +346,7 @@ def make_activation_dataset_hf(
     host_parts: Dict[int, List[torch.Tensor]] = {l: [] for l in layers}
     host_counts: Dict[int, int] = {l: 0 for l in layers}
     chunk_idx: Dict[int, int] = {l: 0 for l in layers}
+    stagers: Dict[int, _PinnedStager] = {l: _PinnedStager() for l in layers}
     total = 0
 
     folders = {}
@@ -310,23 +358,23 @@ def make_activation_dataset_hf(
 
     def drain(l: int):
         # NOTE: the capture hook holds a reference to stores[l]; mutate it in
-        # place, never rebind it.  D2H lands in PINNED host buffers so the
-        # non_blocking copies actually overlap the next forward (a pageable
-        # destination would serialize).
+        # place, never rebind it.  D2H goes through a bounded ring of REUSED
+        # pinned staging buffers (pinned destination keeps the non_blocking
+        # copy async; the ring keeps the pinned footprint at ~depth batches
+        # instead of a whole chunk — ADVICE.md round-1).
         while stores[l]:
             t = stores[l].pop(0).to(dtype)
             if t.is_cuda:
-                host = torch.empty(t.shape, dtype=t.dtype, pin_memory=True)
-                host.copy_(t, non_blocking=True)
+                stagers[l].stage(t, host_parts[l])
+                host_counts[l] += t.shape[0]
             else:
-                host = t
-            host_parts[l].append(host)
-            host_counts[l] += t.shape[0]
+                host_parts[l].append(t)
+                host_counts[l] += t.shape[0]
 
     def flush_chunk(l: int, final: bool = False):
         if host_counts[l] == 0:
             return
-        torch.cuda.synchronize() if torch.cuda.is_available() else None
+        stagers[l].retire_all(host_parts[l])
         data = torch.cat(host_parts[l], dim=0)
         while data.shape[0] >= chunk_size or (final and data.shape[0] > 0):
             part = data[:chunk_size]

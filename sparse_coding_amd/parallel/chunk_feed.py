@@ -42,6 +42,11 @@ class BroadcastChunkFeeder:
 
         if self.rank == self.src:
             assert chunk is not None
+            if chunk.dim() != 2:
+                # the receiver-side shape buffer is fixed at 2 entries; a
+                # higher-rank chunk would corrupt the shape exchange — fail
+                # loudly on the source instead
+                raise ValueError(f"BroadcastChunkFeeder.feed wants [N, d] chunks, got shape {tuple(chunk.shape)}")
             chunk = chunk.to(self.device, torch.float32)
             shape = torch.tensor(list(chunk.shape), device=self.device, dtype=torch.long)
         else:

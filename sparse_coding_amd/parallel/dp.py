@@ -86,12 +86,16 @@ class GradBucketAllReducer:
         world = dist.get_world_size(self.group)
 
         works = []
-        if self.comm_stream is not None:
-            self.comm_stream.wait_stream(torch.cuda.current_stream())
         for bucket, assignment in zip(self.buckets, self.assignments):
             for li, off, n in assignment:
                 bucket[off : off + n].copy_(leaves[li].reshape(-1), non_blocking=True)
             if self.comm_stream is not None:
+                # per-bucket event between the (current-stream) pack copies
+                # and the (comm-stream) all_reduce — a single wait_stream at
+                # the top would not order copies issued after it
+                ev = torch.cuda.Event()
+                ev.record()
+                self.comm_stream.wait_event(ev)
                 with torch.cuda.stream(self.comm_stream):
                     works.append((dist.all_reduce(bucket, async_op=True), bucket, assignment))
             else:
@@ -133,6 +137,7 @@ class DataParallelEnsembleTrainer:
         self.graph_capture = graph_capture
         self._graph = None
         self._graph_B = None
+        self._graph_ws = None
         self._dp_steps = 0
 
     def broadcast_state(self) -> None:
@@ -191,11 +196,19 @@ class DataParallelEnsembleTrainer:
 
             if self.graph_capture and torch.cuda.is_available():
                 B = local_batch.shape[0]
+                # a graph is valid only while the step's workspaces are the
+                # ones it captured: an eager step at a different B reallocates
+                # them (HipSAEStep._alloc), after which replay would write
+                # into freed tensors — identity-check the workspace object
+                # (strong ref held at capture, so the id cannot be reused)
+                if self._graph is not None and getattr(hs, "c", None) is not self._graph_ws:
+                    self._graph = None
+                    self._graph_B = None
                 if self._graph is not None and self._graph_B == B:
                     self._x_static.copy_(local_batch)
                     self._graph.replay()
                     return hs._loss_data(B), {"c": hs.c}
-                if self._dp_steps >= 2 and self._graph is None:
+                if self._dp_steps >= 2 and self._graph is None and self._graph_B != B:
                     try:
                         self._x_static = local_batch.contiguous().clone()
                         torch.cuda.synchronize()
@@ -204,6 +217,7 @@ class DataParallelEnsembleTrainer:
                             dp_step(self._x_static)
                         self._graph = g
                         self._graph_B = B
+                        self._graph_ws = hs.c
                         self._graph.replay()  # capture does not execute
                         return hs._loss_data(B), {"c": hs.c}
                     except Exception as e:  # noqa: BLE001 - optimization only
