@@ -182,6 +182,13 @@ class HipSAEStep:
         if name != "adam":
             raise RuntimeError(f"fused HIP step supports adam only, got {name}")
 
+        # per-feature lr multiplier for the post-resample warmup (Anthropic
+        # protocol): the resampler writes ramp values in place, the Adam
+        # kernels read it every step — allocated here (not in _alloc) so a
+        # batch-size change never resets an active warmup, and hipGraph
+        # capture sees a stable pointer
+        self.lr_mult = torch.ones(self.n_models, self.n_dict, device=dev)
+
         # persistent workspaces, sized lazily on first batch
         self._B = None
         # hipGraph capture of the whole step (single-GPU path): replayed
@@ -321,19 +328,20 @@ class HipSAEStep:
             ext.project_adam(p["encoder"], self.gw, self.norms,
                              st["mu"]["encoder"], st["nu"]["encoder"], step_no,
                              self.n_dict, self.lr, self.beta1, self.beta2,
-                             self.eps, EPS_NORM, True)
+                             self.eps, EPS_NORM, True, lr_mult=self.lr_mult)
         else:
             ext.project_adam(p["decoder"], self.gw, self.norms,
                              st["mu"]["decoder"], st["nu"]["decoder"], step_no,
                              self.n_dict, self.lr, self.beta1, self.beta2,
-                             self.eps, EPS_NORM, True)
+                             self.eps, EPS_NORM, True, lr_mult=self.lr_mult)
             ext.project_adam(p["encoder"], self.gw_enc, self.norms,
                              st["mu"]["encoder"], st["nu"]["encoder"], step_no,
                              self.n_dict, self.lr, self.beta1, self.beta2,
-                             self.eps, EPS_NORM, False)
+                             self.eps, EPS_NORM, False, lr_mult=self.lr_mult)
         ext.bias_adam(p["encoder_bias"], self.g_bias, self.bias_decay,
                       st["mu"]["encoder_bias"], st["nu"]["encoder_bias"],
-                      step_no, self.lr, self.beta1, self.beta2, self.eps)
+                      step_no, self.lr, self.beta1, self.beta2, self.eps,
+                      lr_mult=self.lr_mult)
 
     def _loss_data(self, B: int):
         d = self.d_act
@@ -450,10 +458,11 @@ class HipCenteredStep(HipSAEStep):
         ext.project_adam(p["encoder"], self.gw, self.norms,
                          st["mu"]["encoder"], st["nu"]["encoder"], step_no,
                          self.n_dict, self.lr, self.beta1, self.beta2,
-                         self.eps, EPS_NORM, True)
+                         self.eps, EPS_NORM, True, lr_mult=self.lr_mult)
         ext.bias_adam(p["encoder_bias"], self.g_bias, self.zero_decay,
                       st["mu"]["encoder_bias"], st["nu"]["encoder_bias"],
-                      step_no, self.lr, self.beta1, self.beta2, self.eps)
+                      step_no, self.lr, self.beta1, self.beta2, self.eps,
+                      lr_mult=self.lr_mult)
         ext.bias_adam(p["center"], self.g_center, self.zero_decay,
                       st["mu"]["center"], st["nu"]["center"],
                       step_no, self.lr, self.beta1, self.beta2, self.eps)
@@ -528,10 +537,11 @@ class HipPositiveStep(HipSAEStep):
                          st["mu"]["encoder"], st["nu"]["encoder"], step_no,
                          self.n_dict, self.lr, self.beta1, self.beta2,
                          self.eps, EPS_NORM, True,
-                         w_used=self.Wc, clamp_mask=True)
+                         w_used=self.Wc, clamp_mask=True, lr_mult=self.lr_mult)
         ext.bias_adam(p["encoder_bias"], self.g_bias, self.bias_decay,
                       st["mu"]["encoder_bias"], st["nu"]["encoder_bias"],
-                      step_no, self.lr, self.beta1, self.beta2, self.eps)
+                      step_no, self.lr, self.beta1, self.beta2, self.eps,
+                      lr_mult=self.lr_mult)
 
 
 class HipThresholdStep(HipSAEStep):
@@ -603,13 +613,15 @@ class HipThresholdStep(HipSAEStep):
         ext.project_adam(p["encoder"], self.gw, self.norms,
                          st["mu"]["encoder"], st["nu"]["encoder"], step_no,
                          self.n_dict, self.lr, self.beta1, self.beta2,
-                         self.eps, EPS_NORM, True)
+                         self.eps, EPS_NORM, True, lr_mult=self.lr_mult)
         ext.bias_adam(p["activation_scale"], self.g_scale, zero_decay,
                       st["mu"]["activation_scale"], st["nu"]["activation_scale"],
-                      step_no, self.lr, self.beta1, self.beta2, self.eps)
+                      step_no, self.lr, self.beta1, self.beta2, self.eps,
+                      lr_mult=self.lr_mult)
         ext.bias_adam(p["activation_gain"], self.g_gain, zero_decay,
                       st["mu"]["activation_gain"], st["nu"]["activation_gain"],
-                      step_no, self.lr, self.beta1, self.beta2, self.eps)
+                      step_no, self.lr, self.beta1, self.beta2, self.eps,
+                      lr_mult=self.lr_mult)
 
     def _loss_data(self, B: int):
         mse = self.loss_parts[:, 0] / (B * self.d_act)
@@ -1155,6 +1167,7 @@ class HipTopKStep:
         self.ks_i32 = torch.tensor(self.ks, device=dev, dtype=torch.int32)
         self.zero_l1 = torch.zeros(self.n_models, device=dev)
         self.dummy_bias = torch.zeros(self.n_models, self.n_dict, device=dev)
+        self.lr_mult = torch.ones(self.n_models, self.n_dict, device=dev)
 
         opt = ensemble.optimizer_kwargs
         self.lr = float(opt.get("lr", 1e-3))
@@ -1222,7 +1235,7 @@ class HipTopKStep:
         ext.project_adam(ens.params["dict"], self.gw, self.norms,
                          st["mu"]["dict"], st["nu"]["dict"], st["step"],
                          self.n_dict, self.lr, self.beta1, self.beta2,
-                         self.eps, self.EPS, True)
+                         self.eps, self.EPS, True, lr_mult=self.lr_mult)
 
     def _loss_data(self, B):
         mse = self.loss_parts[:, 0] / (B * self.d_act)

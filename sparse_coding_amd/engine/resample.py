@@ -4,21 +4,39 @@ The reference implements resampling only in its DDP experiment
 (huge_batch_size.py:224-254): dead = never fired over a window; dead encoder
 rows re-initialized from the worst-reconstructed examples; Adam state zeroed
 for the resampled slices.  Here the same rule is applied to the stacked
-ensemble, entirely on-device:
+ensemble, entirely on-device, with two selectable protocols:
+
+``protocol="worst"`` (the reference's rule, round-1 default): replacement
+directions come from the top-``n_track`` worst-reconstructed examples.
+
+``protocol="anthropic"`` (the retention protocol of the resampling
+literature; default for long runs): replacement directions are sampled
+WITHOUT replacement with probability proportional to the SQUARED
+per-example reconstruction loss, via an Efraimidis–Spirakis weighted
+reservoir maintained on device (key = log(u)/w, keep the ``n_track``
+largest keys — an exact loss^2-weighted sample of the whole observation
+window, not just its worst tail).  On top of the replacement writes:
+
+* the encoder row is scaled to ``encoder_norm_ratio`` x the mean norm of
+  the ALIVE encoder rows (not all rows — dead rows shrink under decay and
+  would drag the target norm down);
+* the per-feature learning rate of every replaced feature is ramped from
+  ``warmup_start`` x lr back to lr over ``warmup_steps`` steps, so Adam
+  cannot blast the fresh direction away on its first batches.  The ramp is
+  written in place into ``HipSAEStep.lr_mult``, which the fused Adam
+  kernels (k_project_adam / k_bias_adam) read every step — hipGraph-safe
+  because the tensor pointer never changes.  The torch/vmap oracle backend
+  ignores the warmup (it is a per-step numerical reference, not the
+  production trainer).
+
+Mechanics shared by both protocols:
 
 * fired counts come for free from the fused forward kernel (k_enc_fwd
   accumulates them into HipSAEStep.fired) or are recomputed from aux["c"]
   on the torch backend;
-* worst examples are tracked per model with a batched top-K merge
-  (no host sync per step);
-* the replacement writes + Adam-state zeroing are batched index ops on the
-  stacked [M, n, d] tensors.
-
-Replacement scale: unit worst-example direction × 0.2 × mean encoder row
-norm (the convention of the resampling literature; the reference's literal
-expression at :243 divides by the mean norm instead — with its transposed
-weight layout — so the intent, re-scaling new rows well below typical rows,
-is preserved rather than the expression).
+* the example pool is merged per step with a batched top-K (no host sync);
+* the replacement writes + Adam-state zeroing run in one k_resample launch
+  on GPU (batched index ops on the stacked [M, n, d] tensors on CPU).
 """
 
 from __future__ import annotations
@@ -29,24 +47,44 @@ import torch
 
 
 class EnsembleResampler:
-    """Tracks fired counts + worst examples for a FunctionalEnsemble and
-    resamples dead features on demand."""
+    """Tracks fired counts + a replacement-example pool for a
+    FunctionalEnsemble and resamples dead features on demand."""
 
-    def __init__(self, ensemble, n_track: int = 512, encoder_norm_ratio: float = 0.2):
+    def __init__(self, ensemble, n_track: int = 512, encoder_norm_ratio: float = 0.2,
+                 protocol: str = "worst", warmup_steps: int = 1000,
+                 warmup_start: float = 0.1):
+        if protocol not in ("worst", "anthropic"):
+            raise ValueError(f"unknown resample protocol {protocol!r}")
         self.ens = ensemble
         self.n_track = n_track
         self.encoder_norm_ratio = encoder_norm_ratio
+        self.protocol = protocol
+        self.warmup_steps = warmup_steps
+        self.warmup_start = warmup_start
         # TopK ensembles name their weight "dict"; SAEs use "encoder"
         self.w_key = "encoder" if "encoder" in ensemble.params else "dict"
         M, n, d = ensemble.params[self.w_key].shape
         dev = ensemble.params[self.w_key].device
         self.fired = torch.zeros(M, n, device=dev)
-        self.worst_losses = torch.full((M, n_track), -float("inf"), device=dev)
-        self.worst_examples = torch.zeros(M, n_track, d, device=dev)
+        # pool scores: per-example loss ("worst") or reservoir key ("anthropic")
+        self.pool_scores = torch.full((M, n_track), -float("inf"), device=dev)
+        self.pool_examples = torch.zeros(M, n_track, d, device=dev)
+        self._warmup_mask: Optional[torch.Tensor] = None  # [M, n] bool
+        self._warmup_t = 0
+
+    # round-1 attribute names kept as aliases (scripts/tests referenced them)
+    @property
+    def worst_losses(self):
+        return self.pool_scores
+
+    @property
+    def worst_examples(self):
+        return self.pool_examples
 
     @torch.no_grad()
     def observe(self, batch: torch.Tensor, aux: Optional[dict] = None) -> None:
-        """Update fired counts + worst-example pool after a step.
+        """Update fired counts + the example pool after a step, and advance
+        any active post-resample lr warmup.
 
         On the HIP backend, per-example losses come from the residual
         workspace; on the torch backend from aux["c"] and a re-decode.
@@ -68,17 +106,69 @@ class EnsembleResampler:
             x_hat = torch.einsum("mbn,mnd->mbd", c, what)
             per_ex = (x_hat - batch.unsqueeze(0)).pow(2).mean(dim=-1)
 
-        # batched top-K merge of the worst examples
+        if self.protocol == "anthropic":
+            # Efraimidis–Spirakis A-Res: key = log(u)/w with w = loss^2;
+            # keeping the n_track largest keys over the stream is an exact
+            # weighted sample without replacement of the window
+            w = per_ex.pow(2).clamp_min(1e-30)
+            u = torch.rand_like(per_ex).clamp_min(1e-12)
+            scores = torch.log(u) / w
+        else:
+            scores = per_ex
+
+        # batched top-K merge of the pool
         M = per_ex.shape[0]
-        losses = torch.cat([self.worst_losses, per_ex], dim=1)
-        top = torch.topk(losses, self.n_track, dim=1)
-        self.worst_losses = top.values
+        cat_scores = torch.cat([self.pool_scores, scores], dim=1)
+        top = torch.topk(cat_scores, self.n_track, dim=1)
+        self.pool_scores = top.values
         examples = torch.cat(
-            [self.worst_examples, batch.unsqueeze(0).expand(M, *batch.shape)], dim=1
+            [self.pool_examples, batch.unsqueeze(0).expand(M, *batch.shape)], dim=1
         )
-        self.worst_examples = torch.gather(
+        self.pool_examples = torch.gather(
             examples, 1, top.indices[:, :, None].expand(-1, -1, examples.shape[-1])
         )
+        self._tick_warmup()
+
+    @torch.no_grad()
+    def _tick_warmup(self) -> None:
+        if self._warmup_mask is None:
+            return
+        hs = getattr(self.ens, "_hip_step", None)
+        if hs is None or not hasattr(hs, "lr_mult"):
+            self._warmup_mask = None
+            return
+        self._warmup_t += 1
+        if self._warmup_t >= self.warmup_steps:
+            hs.lr_mult.fill_(1.0)
+            self._warmup_mask = None
+            return
+        frac = self._warmup_t / self.warmup_steps
+        factor = self.warmup_start + (1.0 - self.warmup_start) * frac
+        hs.lr_mult.fill_(1.0)
+        hs.lr_mult.masked_fill_(self._warmup_mask, factor)
+
+    def _replaced_mask(self) -> torch.Tensor:
+        """[M, n] bool: the dead features the replacement pass will rewrite
+        (first n_track dead per model, in index order — k_resample's rule)."""
+        dead = self.fired == 0
+        rank = torch.cumsum(dead.to(torch.int32), dim=1) - 1
+        return dead & (rank < self.n_track)
+
+    def _enc_scale(self) -> torch.Tensor:
+        """Per-model replacement norm: ratio x mean ALIVE row norm under the
+        anthropic protocol, ratio x mean row norm under "worst" (round-1 /
+        reference behavior)."""
+        p = self.ens.params
+        norms = torch.norm(p[self.w_key], dim=-1)  # [M, n]
+        if self.protocol == "anthropic":
+            alive = (self.fired > 0).float()
+            denom = alive.sum(dim=1).clamp_min(1.0)
+            mean_norm = (norms * alive).sum(dim=1) / denom
+            # all-dead model: fall back to the all-rows mean
+            mean_norm = torch.where(denom > 1.0, mean_norm, norms.mean(dim=1))
+        else:
+            mean_norm = norms.mean(dim=1)
+        return self.encoder_norm_ratio * mean_norm  # [M]
 
     @torch.no_grad()
     def resample(self) -> torch.Tensor:
@@ -89,12 +179,38 @@ class EnsembleResampler:
         wk = self.w_key
         M, n, d = p[wk].shape
 
+        if self.protocol == "anthropic":
+            replaced = self._replaced_mask()
+
         if p[wk].is_cuda:
             from sparse_coding_amd import ops as _ops
 
             ext = _ops.get_extension(required=True)
-            return self._resample_fused(ext)
+            counts = self._resample_fused(ext)
+        else:
+            counts = self._resample_cpu()
+
+        if self.protocol == "anthropic":
+            hs = getattr(ens, "_hip_step", None)
+            if hs is not None and hasattr(hs, "lr_mult") and bool(replaced.any()):
+                if self._warmup_mask is None:
+                    self._warmup_mask = replaced
+                else:
+                    self._warmup_mask |= replaced
+                self._warmup_t = 0
+                hs.lr_mult.fill_(1.0)
+                hs.lr_mult.masked_fill_(self._warmup_mask, self.warmup_start)
+        return counts
+
+    @torch.no_grad()
+    def _resample_cpu(self) -> torch.Tensor:
+        ens = self.ens
+        p = ens.params
+        st = ens.optim_states
+        wk = self.w_key
+        M, n, d = p[wk].shape
         counts = torch.zeros(M, dtype=torch.long)
+        enc_scale = self._enc_scale()
 
         for m in range(M):
             dead = torch.where(self.fired[m] == 0)[0]
@@ -102,15 +218,14 @@ class EnsembleResampler:
             if k == 0:
                 continue
             dead = dead[:k]
-            worst = self.worst_examples[m, :k]
-            worst_unit = worst / torch.clamp(torch.norm(worst, dim=-1, keepdim=True), 1e-8)
-            avg_norm = torch.norm(p[wk][m], dim=-1).mean()
+            pool = self.pool_examples[m, :k]
+            pool_unit = pool / torch.clamp(torch.norm(pool, dim=-1, keepdim=True), 1e-8)
 
-            p[wk][m, dead] = worst_unit * self.encoder_norm_ratio * avg_norm
+            p[wk][m, dead] = pool_unit * enc_scale[m]
             st["mu"][wk][m, dead] = 0
             st["nu"][wk][m, dead] = 0
             if "decoder" in p:
-                p["decoder"][m, dead] = worst_unit
+                p["decoder"][m, dead] = pool_unit
                 st["mu"]["decoder"][m, dead] = 0
                 st["nu"]["decoder"][m, dead] = 0
             if "encoder_bias" in p:
@@ -120,14 +235,14 @@ class EnsembleResampler:
             counts[m] = k
 
         self.fired.zero_()
-        self.worst_losses.fill_(-float("inf"))
+        self.pool_scores.fill_(-float("inf"))
         return counts
 
     @torch.no_grad()
     def _resample_fused(self, ext) -> torch.Tensor:
         """K14 fully on device (ops k_resample): one launch ranks the dead
-        features per model in index order, rewrites rows from the worst-
-        example pool, and zeroes the Adam slices — no host loop or sync."""
+        features per model in index order, rewrites rows from the example
+        pool, and zeroes the Adam slices — no host loop or sync."""
         ens = self.ens
         p = ens.params
         st = ens.optim_states
@@ -135,10 +250,9 @@ class EnsembleResampler:
         M, n, d = p[wk].shape
         dev = p[wk].device
 
-        worst_unit = self.worst_examples / torch.clamp(
-            torch.norm(self.worst_examples, dim=-1, keepdim=True), 1e-8)
-        avg_norm = torch.norm(p[wk], dim=-1).mean(dim=1)  # [M]
-        enc_scale = (self.encoder_norm_ratio * avg_norm).contiguous()
+        pool_unit = self.pool_examples / torch.clamp(
+            torch.norm(self.pool_examples, dim=-1, keepdim=True), 1e-8)
+        enc_scale = self._enc_scale().contiguous()
         counts = torch.zeros(M, device=dev, dtype=torch.int32)
 
         kwargs = {}
@@ -148,10 +262,10 @@ class EnsembleResampler:
         if "encoder_bias" in p:
             bkw = dict(bias=p["encoder_bias"], mu_b=st["mu"]["encoder_bias"],
                        nu_b=st["nu"]["encoder_bias"])
-        ext.resample(self.fired, worst_unit.contiguous(), enc_scale,
+        ext.resample(self.fired, pool_unit.contiguous(), enc_scale,
                      p[wk], st["mu"][wk], st["nu"][wk],
                      counts_out=counts, **kwargs, **bkw)
 
         self.fired.zero_()
-        self.worst_losses.fill_(-float("inf"))
+        self.pool_scores.fill_(-float("inf"))
         return counts.long().cpu()

@@ -971,11 +971,15 @@ void k_project_adam(float* __restrict__ W,        // [M*n, d]
                     float lr, float b1, float b2, float eps_adam,
                     float eps_norm, int project,
                     const float* __restrict__ w_used_p,  // nullptr -> W
-                    int clamp_mask) {  // 1: g *= [W >= 0] (clamp backward)
+                    int clamp_mask,  // 1: g *= [W >= 0] (clamp backward)
+                    const float* __restrict__ lr_mult) { // [M*n] per-row lr
+                                                         // scale (post-resample
+                                                         // warmup) or nullptr
   const int wave = threadIdx.x / WAVE;
   const int lane = threadIdx.x & (WAVE - 1);
   const int row = blockIdx.x * (NTHREADS / WAVE) + wave;
   if (row >= n_rows_total) return;
+  const float lr_eff = lr_mult ? lr * lr_mult[row] : lr;
 
   float* w_param = W + (long)row * d;
   const float* w = w_used_p ? w_used_p + (long)row * d : w_param;
@@ -1014,7 +1018,7 @@ void k_project_adam(float* __restrict__ W,        // [M*n, d]
     float v1 = b2 * nu_r[j] + (1.0f - b2) * g * g;
     mu_r[j] = m1;
     nu_r[j] = v1;
-    w_param[j] -= lr * (m1 / bc1) / (sqrtf(v1 / bc2) + eps_adam);
+    w_param[j] -= lr_eff * (m1 / bc1) / (sqrtf(v1 / bc2) + eps_adam);
   }
 }
 
@@ -1028,8 +1032,10 @@ void k_bias_adam(float* __restrict__ bias,        // [M, n]
                  const float* __restrict__ bias_decay, // [M]
                  float* __restrict__ mu, float* __restrict__ nu,
                  const float* __restrict__ step_no,
-                 int n, float lr, float b1, float b2, float eps_adam) {
+                 int n, float lr, float b1, float b2, float eps_adam,
+                 const float* __restrict__ lr_mult) { // [M, n] or nullptr
   const int m = blockIdx.x;
+  const float* lrm_m = lr_mult ? lr_mult + (long)m * n : nullptr;
   float* b_m = bias + (long)m * n;
   const float* g_m = g_bias + (long)m * n;
   float* mu_m = mu + (long)m * n;
@@ -1064,7 +1070,8 @@ void k_bias_adam(float* __restrict__ bias,        // [M, n]
     float v1 = b2 * nu_m[j] + (1.0f - b2) * g * g;
     mu_m[j] = m1;
     nu_m[j] = v1;
-    b_m[j] -= lr * (m1 / bc1) / (sqrtf(v1 / bc2) + eps_adam);
+    float lr_eff = lrm_m ? lr * lrm_m[j] : lr;
+    b_m[j] -= lr_eff * (m1 / bc1) / (sqrtf(v1 / bc2) + eps_adam);
   }
 }
 

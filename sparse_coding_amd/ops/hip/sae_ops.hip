@@ -242,7 +242,8 @@ void project_adam(torch::Tensor W, torch::Tensor gw, torch::Tensor norms,
                   torch::Tensor mu, torch::Tensor nu, torch::Tensor step_no,
                   long n_per_model, double lr, double b1, double b2,
                   double eps_adam, double eps_norm, bool project,
-                  c10::optional<torch::Tensor> w_used, bool clamp_mask) {
+                  c10::optional<torch::Tensor> w_used, bool clamp_mask,
+                  c10::optional<torch::Tensor> lr_mult) {
   CHECK_IN(W); CHECK_IN(gw); CHECK_IN(norms); CHECK_IN(mu); CHECK_IN(nu);
   CHECK_IN(step_no);
   const float* wu = nullptr;
@@ -251,6 +252,12 @@ void project_adam(torch::Tensor W, torch::Tensor gw, torch::Tensor norms,
     wu = w_used->data_ptr<float>();
   }
   long rows = W.numel() / W.size(-1);
+  const float* lrm = nullptr;
+  if (lr_mult.has_value()) {
+    CHECK_IN(lr_mult.value());
+    TORCH_CHECK(lr_mult->numel() == rows, "lr_mult must have one entry per row");
+    lrm = lr_mult->data_ptr<float>();
+  }
   int d = W.size(-1);
   dim3 grid(cdiv(rows, NTHREADS / WAVE));
   hipLaunchKernelGGL(k_project_adam, grid, dim3(NTHREADS), 0, cur_stream(),
@@ -259,20 +266,27 @@ void project_adam(torch::Tensor W, torch::Tensor gw, torch::Tensor norms,
                      nu.data_ptr<float>(), step_no.data_ptr<float>(),
                      (int)rows, (int)n_per_model, d, (float)lr, (float)b1,
                      (float)b2, (float)eps_adam, (float)eps_norm,
-                     project ? 1 : 0, wu, clamp_mask ? 1 : 0);
+                     project ? 1 : 0, wu, clamp_mask ? 1 : 0, lrm);
 }
 
 void bias_adam(torch::Tensor bias, torch::Tensor g_bias, torch::Tensor decay,
                torch::Tensor mu, torch::Tensor nu, torch::Tensor step_no,
-               double lr, double b1, double b2, double eps_adam) {
+               double lr, double b1, double b2, double eps_adam,
+               c10::optional<torch::Tensor> lr_mult) {
   CHECK_IN(bias); CHECK_IN(g_bias); CHECK_IN(decay); CHECK_IN(mu);
   CHECK_IN(nu); CHECK_IN(step_no);
   int M = bias.size(0), n = bias.size(1);
+  const float* lrm = nullptr;
+  if (lr_mult.has_value()) {
+    CHECK_IN(lr_mult.value());
+    TORCH_CHECK(lr_mult->numel() == (long)M * n, "lr_mult must match bias shape");
+    lrm = lr_mult->data_ptr<float>();
+  }
   hipLaunchKernelGGL(k_bias_adam, dim3(M), dim3(NTHREADS), 0, cur_stream(),
                      bias.data_ptr<float>(), g_bias.data_ptr<float>(),
                      decay.data_ptr<float>(), mu.data_ptr<float>(),
                      nu.data_ptr<float>(), step_no.data_ptr<float>(), n,
-                     (float)lr, (float)b1, (float)b2, (float)eps_adam);
+                     (float)lr, (float)b1, (float)b2, (float)eps_adam, lrm);
 }
 
 void transpose_scale(torch::Tensor src, torch::Tensor dst,
@@ -473,6 +487,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("nu"), py::arg("step_no"), py::arg("n_per_model"),
         py::arg("lr"), py::arg("b1"), py::arg("b2"), py::arg("eps_adam"),
         py::arg("eps_norm"), py::arg("project"),
-        py::arg("w_used") = py::none(), py::arg("clamp_mask") = false);
-  m.def("bias_adam", &bias_adam, "Adam on bias with L2-norm decay");
+        py::arg("w_used") = py::none(), py::arg("clamp_mask") = false,
+        py::arg("lr_mult") = py::none());
+  m.def("bias_adam", &bias_adam, "Adam on bias with L2-norm decay",
+        py::arg("bias"), py::arg("g_bias"), py::arg("decay"), py::arg("mu"),
+        py::arg("nu"), py::arg("step_no"), py::arg("lr"), py::arg("b1"),
+        py::arg("b2"), py::arg("eps_adam"), py::arg("lr_mult") = py::none());
 }

@@ -489,3 +489,76 @@ def test_dp_rccl_graph_capture_one_rank():
             assert err < 1e-6, (k, err)
     finally:
         dist.destroy_process_group()
+
+
+def test_lr_mult_freezes_rows():
+    """k_project_adam/k_bias_adam per-row lr multiplier: rows at 0 must not
+    move; rows at 1 must match a run without lr_mult."""
+    from sparse_coding_amd.engine.ensemble import FunctionalEnsemble
+    from sparse_coding_amd.functional.optim import adam
+    from sparse_coding_amd.models.sae_signatures import FunctionalTiedSAE
+
+    torch.manual_seed(41)
+    M, B, d, n = 2, 256, 64, 256
+    models = [FunctionalTiedSAE.init(d, n, 1e-3, device=DEV) for _ in range(M)]
+    ens = FunctionalEnsemble(models, FunctionalTiedSAE, adam, {"lr": 1e-3}, device=DEV, backend="hip")
+    models2 = [({k: v.clone() for k, v in p.items()}, {k: v.clone() for k, v in b.items()})
+               for p, b in ens.unstack()]
+    ens_ref = FunctionalEnsemble(models2, FunctionalTiedSAE, adam, {"lr": 1e-3}, device=DEV, backend="hip")
+
+    hs = ens._hip_step
+    hs.lr_mult[:, : n // 2] = 0.0
+    x = torch.randn(B, d, device=DEV)
+    before = ens.params["encoder"][:, : n // 2].clone()
+    before_bias = ens.params["encoder_bias"][:, : n // 2].clone()
+    # one step from identical states: unfrozen rows must match the
+    # all-ones-lr_mult run exactly (after this step the frozen rows make
+    # the forwards diverge, so only the first step is comparable)
+    ens.step_batch(x)
+    ens_ref.step_batch(x)
+    torch.cuda.synchronize()
+    err = (ens.params["encoder"][:, n // 2 :] - ens_ref.params["encoder"][:, n // 2 :]).abs().max()
+    assert err.item() == 0.0
+    for _ in range(2):
+        ens.step_batch(x)
+    torch.cuda.synchronize()
+    assert torch.equal(ens.params["encoder"][:, : n // 2], before)
+    assert torch.equal(ens.params["encoder_bias"][:, : n // 2], before_bias)
+
+
+def test_anthropic_resample_warmup_gpu():
+    """Fused anthropic protocol: resample rewrites dead rows, sets lr_mult to
+    warmup_start on replaced rows, and ramps back to 1 over warmup_steps."""
+    from sparse_coding_amd.engine.ensemble import FunctionalEnsemble
+    from sparse_coding_amd.engine.resample import EnsembleResampler
+    from sparse_coding_amd.functional.optim import adam
+    from sparse_coding_amd.models.sae_signatures import FunctionalTiedSAE
+
+    torch.manual_seed(42)
+    M, B, d, n = 2, 256, 64, 256
+    models = [FunctionalTiedSAE.init(d, n, 1e-3, device=DEV) for _ in range(M)]
+    ens = FunctionalEnsemble(models, FunctionalTiedSAE, adam, {"lr": 1e-3}, device=DEV, backend="hip")
+    rs = EnsembleResampler(ens, n_track=64, protocol="anthropic",
+                           warmup_steps=4, warmup_start=0.25)
+    with torch.no_grad():
+        ens.params["encoder_bias"][:, : n // 2] = -1e6
+    x = torch.randn(B, d, device=DEV)
+    for _ in range(3):
+        _, aux = ens.step_batch(x)
+        rs.observe(x, aux)
+    counts = rs.resample()
+    assert (counts == 64).all()
+    hs = ens._hip_step
+    torch.cuda.synchronize()
+    assert (hs.lr_mult[:, :64] == 0.25).all()
+    assert (hs.lr_mult[:, 64:] == 1.0).all()
+    # replaced encoder rows: unit direction x 0.2 x alive mean norm
+    alive_norm = torch.norm(ens.params["encoder"][:, n // 2 :], dim=-1).mean(dim=1)
+    new_norms = torch.norm(ens.params["encoder"][:, :64], dim=-1)
+    assert torch.allclose(new_norms, (0.2 * alive_norm)[:, None].expand_as(new_norms), rtol=0.08)
+    # ramp: after warmup_steps observes lr_mult returns to 1 everywhere
+    for _ in range(4):
+        _, aux = ens.step_batch(x)
+        rs.observe(x, aux)
+    torch.cuda.synchronize()
+    assert (hs.lr_mult == 1.0).all()
