@@ -136,7 +136,10 @@ def main():
     l1s = np.logspace(args.l1_lo, args.l1_hi, args.n_models)
 
     def make_gen():
+        # generate_rand_feats draws from NUMPY's RNG — seed both so every
+        # make_gen() call rebuilds the SAME ground-truth dictionary
         torch.manual_seed(100)
+        np.random.seed(100)
         return RandomDatasetGenerator(
             activation_dim=d, n_ground_truth_components=args.n_true,
             batch_size=args.batch, feature_num_nonzero=args.nonzero,

@@ -307,6 +307,8 @@ class HipSAEStep:
         elif self.tied:
             ext.grad_w(self.c, self.r, self.gw, gscale, 0.0, bk_gw, prio, bn)
             ext.grad_w(self.gpre, x, self.gw, 1.0, 1.0, bk_gw, prio, bn)
+            if on_grads is not None:  # M == 1: single unchunked grad tensor
+                on_grads([self.gw])
         else:
             ext.grad_w(self.c, self.r, self.gw, gscale, 0.0, bk_gw, prio, bn)
             if on_grads is not None:

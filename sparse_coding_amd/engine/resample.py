@@ -52,7 +52,7 @@ class EnsembleResampler:
 
     def __init__(self, ensemble, n_track: int = 512, encoder_norm_ratio: float = 0.2,
                  protocol: str = "worst", warmup_steps: int = 1000,
-                 warmup_start: float = 0.1):
+                 warmup_start: float = 0.1, dead_threshold: Optional[float] = None):
         if protocol not in ("worst", "anthropic"):
             raise ValueError(f"unknown resample protocol {protocol!r}")
         self.ens = ensemble
@@ -61,6 +61,15 @@ class EnsembleResampler:
         self.protocol = protocol
         self.warmup_steps = warmup_steps
         self.warmup_start = warmup_start
+        # dead = fires on fewer than this fraction of observed examples.
+        # 0 = the reference's literal never-fired rule (huge_batch_size.py:230);
+        # the anthropic protocol defaults to 1e-5 because a feature firing
+        # once per ~1M examples is dead for every practical purpose yet
+        # never trips the ==0 rule over a long window.
+        if dead_threshold is None:
+            dead_threshold = 1e-5 if protocol == "anthropic" else 0.0
+        self.dead_threshold = dead_threshold
+        self.examples_seen = 0
         # TopK ensembles name their weight "dict"; SAEs use "encoder"
         self.w_key = "encoder" if "encoder" in ensemble.params else "dict"
         M, n, d = ensemble.params[self.w_key].shape
@@ -89,6 +98,7 @@ class EnsembleResampler:
         On the HIP backend, per-example losses come from the residual
         workspace; on the torch backend from aux["c"] and a re-decode.
         """
+        self.examples_seen += batch.shape[0]
         hs = getattr(self.ens, "_hip_step", None)
         if hs is not None:
             self.fired += hs.fired
@@ -147,10 +157,18 @@ class EnsembleResampler:
         hs.lr_mult.fill_(1.0)
         hs.lr_mult.masked_fill_(self._warmup_mask, factor)
 
+    def _effective_fired(self) -> torch.Tensor:
+        """fired counts with sub-threshold (rare-firing) features zeroed, so
+        the ==0 dead rule sees them as dead."""
+        if self.dead_threshold <= 0.0 or self.examples_seen == 0:
+            return self.fired
+        min_count = self.dead_threshold * self.examples_seen
+        return torch.where(self.fired < min_count, torch.zeros_like(self.fired), self.fired)
+
     def _replaced_mask(self) -> torch.Tensor:
         """[M, n] bool: the dead features the replacement pass will rewrite
         (first n_track dead per model, in index order — k_resample's rule)."""
-        dead = self.fired == 0
+        dead = self._effective_fired() == 0
         rank = torch.cumsum(dead.to(torch.int32), dim=1) - 1
         return dead & (rank < self.n_track)
 
@@ -161,7 +179,7 @@ class EnsembleResampler:
         p = self.ens.params
         norms = torch.norm(p[self.w_key], dim=-1)  # [M, n]
         if self.protocol == "anthropic":
-            alive = (self.fired > 0).float()
+            alive = (self._effective_fired() > 0).float()
             denom = alive.sum(dim=1).clamp_min(1.0)
             mean_norm = (norms * alive).sum(dim=1) / denom
             # all-dead model: fall back to the all-rows mean
@@ -212,8 +230,9 @@ class EnsembleResampler:
         counts = torch.zeros(M, dtype=torch.long)
         enc_scale = self._enc_scale()
 
+        eff = self._effective_fired()
         for m in range(M):
-            dead = torch.where(self.fired[m] == 0)[0]
+            dead = torch.where(eff[m] == 0)[0]
             k = min(int(dead.numel()), self.n_track)
             if k == 0:
                 continue
@@ -235,6 +254,7 @@ class EnsembleResampler:
             counts[m] = k
 
         self.fired.zero_()
+        self.examples_seen = 0
         self.pool_scores.fill_(-float("inf"))
         return counts
 
@@ -262,10 +282,11 @@ class EnsembleResampler:
         if "encoder_bias" in p:
             bkw = dict(bias=p["encoder_bias"], mu_b=st["mu"]["encoder_bias"],
                        nu_b=st["nu"]["encoder_bias"])
-        ext.resample(self.fired, pool_unit.contiguous(), enc_scale,
+        ext.resample(self._effective_fired().contiguous(), pool_unit.contiguous(), enc_scale,
                      p[wk], st["mu"][wk], st["nu"][wk],
                      counts_out=counts, **kwargs, **bkw)
 
         self.fired.zero_()
+        self.examples_seen = 0
         self.pool_scores.fill_(-float("inf"))
         return counts.long().cpu()

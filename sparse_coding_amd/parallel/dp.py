@@ -110,17 +110,64 @@ class GradBucketAllReducer:
             torch.cuda.current_stream().wait_stream(self.comm_stream)
 
 
+def _reduce_scatter_rows(flat: torch.Tensor, out: torch.Tensor, group) -> "dist.Work":
+    """reduce_scatter_tensor(sum) of a [world*k, ...] tensor into a [k, ...]
+    shard; falls back to all_reduce + local narrow on backends without a
+    native reduce-scatter (gloo) — numerically identical (sum then slice)."""
+    try:
+        return dist.reduce_scatter_tensor(out, flat, op=dist.ReduceOp.SUM,
+                                          group=group, async_op=True)
+    except (RuntimeError, ValueError):
+        work = dist.all_reduce(flat, op=dist.ReduceOp.SUM, group=group, async_op=True)
+
+        class _SliceWork:
+            def wait(self_inner):
+                work.wait()
+                rank = dist.get_rank(group)
+                k = out.shape[0]
+                out.copy_(flat[rank * k : (rank + 1) * k])
+
+        return _SliceWork()
+
+
+def _all_gather_rows(out_flat: torch.Tensor, shard: torch.Tensor, group) -> None:
+    """all_gather of row shards into the full [world*k, ...] tensor."""
+    try:
+        dist.all_gather_into_tensor(out_flat, shard.contiguous(), group=group)
+    except (RuntimeError, ValueError):
+        world = dist.get_world_size(group)
+        k = shard.shape[0]
+        parts = [out_flat[r * k : (r + 1) * k] for r in range(world)]
+        dist.all_gather(parts, shard.contiguous(), group=group)
+
+
 class DataParallelEnsembleTrainer:
     """DP over activation shards: every rank holds a replica of the SAME
-    ensemble; each step consumes a per-rank shard of the global batch,
-    all-reduces the [M, n, d] gradients, and applies the (deterministic)
-    Adam update locally — states stay bit-identical across ranks without a
-    broadcast."""
+    ensemble; each step consumes a per-rank shard of the global batch.
+
+    dp_mode:
+      "allreduce" (default) — all-reduce the [M, n, d] gradients (chunked
+        over model halves, overlapped with the remaining grad GEMMs on a
+        side stream) and apply the full, deterministic Adam update locally;
+        replicas stay bit-identical without a broadcast.
+      "rs_ag" — ZeRO-style: reduce-scatter each weight-gradient chunk into
+        per-rank row shards (half the exposed bytes of an all-reduce on a
+        per-link-bound xGMI ring), run the renorm-projected Adam on the
+        LOCAL shard only (optimizer compute and moments sharded 1/world),
+        then all-gather the updated parameter rows.  Optimizer state is
+        authoritative only on the owning rank — call
+        consolidate_optim_state() before unstack()/checkpointing.
+        Supported for the plain tied/untied HipSAEStep; other fused steps
+        and the torch backend fall back to "allreduce"."""
 
     def __init__(self, ensemble, bucket_bytes: int = 64 << 20, group=None,
-                 force_dp_path: bool = False, graph_capture: Optional[bool] = None):
+                 force_dp_path: bool = False, graph_capture: Optional[bool] = None,
+                 dp_mode: str = "allreduce"):
+        if dp_mode not in ("allreduce", "rs_ag"):
+            raise ValueError(f"unknown dp_mode {dp_mode!r}")
         self.ensemble = ensemble
         self.group = group
+        self.dp_mode = dp_mode
         self._reducer: Optional[GradBucketAllReducer] = None
         self.bucket_bytes = bucket_bytes
         self.world_size = dist.get_world_size(group) if dist.is_initialized() else 1
