@@ -32,8 +32,10 @@ import torch
 def parse_args():
     p = argparse.ArgumentParser()
     p.add_argument("--gpus", type=int, default=1)
-    p.add_argument("--steps", type=int, default=50)
-    p.add_argument("--warmup", type=int, default=10)
+    # defaults give a multi-second timed region (~6 s at the flagship
+    # ~3 ms/step) so driver-observed wall time corroborates the printed value
+    p.add_argument("--steps", type=int, default=2000)
+    p.add_argument("--warmup", type=int, default=50)
     p.add_argument("--batch", type=int, default=2048, help="per-GPU batch of activation vectors")
     p.add_argument("--d-model", type=int, default=512, help="activation width (Pythia-70m residual)")
     p.add_argument("--dict-ratio", type=int, default=8, help="dict size multiple")
@@ -44,6 +46,9 @@ def parse_args():
                    help="model family: sae (tied/untied per --tied; the BASELINE default), "
                         "topk (k=32 + dead-neuron resampling, config 5), thresholding")
     p.add_argument("--backend", choices=["auto", "hip", "torch"], default="auto")
+    p.add_argument("--dp-mode", choices=["allreduce", "rs_ag"], default="allreduce",
+                   help="multi-GPU gradient exchange: chunked overlapped all-reduce, or "
+                        "reduce-scatter + sharded Adam + all-gather (ZeRO-style)")
     p.add_argument("--dtype", choices=["fp32"], default="fp32",
                    help="compute dtype; fp32 matches the reference (fp32 params, BASELINE.md)")
     p.add_argument("--profile-tag", default="", help="extra tag echoed in the JSON config")
@@ -117,7 +122,7 @@ def main():
         models = [sig.init(d, n_dict, float(l1), device=device) for l1 in l1_values]
     ensemble = FunctionalEnsemble(models, sig, adam, {"lr": 1e-3}, device=device,
                                   backend=args.backend, no_stacking=no_stacking)
-    trainer = DataParallelEnsembleTrainer(ensemble)
+    trainer = DataParallelEnsembleTrainer(ensemble, dp_mode=args.dp_mode)
     if args.sig == "topk":
         from sparse_coding_amd.engine.resample import EnsembleResampler
 
@@ -145,7 +150,7 @@ def main():
             # (rare) resample itself once mid-run
             resampler.observe(data_pool[i % n_pool], aux)
             if i == args.steps // 2:
-                resampler.resample()
+                trainer.resample(resampler)
     barrier_sync()
     elapsed = time.perf_counter() - t0
 
@@ -182,7 +187,7 @@ def main():
                 "tied": bool(args.tied),
                 "global_batch": world_size * B,
                 "seq_len": None,
-                "parallelism": f"dp{world_size}",
+                "parallelism": f"dp{world_size}" + ("-rsag" if args.dp_mode == "rs_ag" else ""),
                 "backend": ensemble._hip_step.__class__.__name__ if ensemble._hip_step else "torch-eager",
                 "optimizer": "adam",
                 "tag": args.profile_tag,

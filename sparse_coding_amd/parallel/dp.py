@@ -197,12 +197,26 @@ class DataParallelEnsembleTrainer:
                 if leaf.dtype.is_floating_point or leaf.dtype in (torch.int32, torch.int64):
                     dist.broadcast(leaf, src=0, group=self.group)
 
+    def _rs_ag_supported(self, hs) -> bool:
+        from sparse_coding_amd.engine.hip_step import HipSAEStep
+
+        return (type(hs) is HipSAEStep and not hs.masked and not hs.reverse
+                and hs.n_dict % max(self.world_size, 1) == 0)
+
     def step(self, local_batch: torch.Tensor):
         if self.world_size == 1 and not self.force_dp_path:
             # single-process: take the ensemble's own step (hipGraph-captured
             # on the fused path)
             return self.ensemble.step_batch(local_batch)
         hs = getattr(self.ensemble, "_hip_step", None)
+        if self.dp_mode == "rs_ag":
+            if hs is not None and self._rs_ag_supported(hs):
+                return self._step_rs_ag(hs, local_batch)
+            if hs is None and getattr(self.ensemble.optimizer_func, "__name__", "") == "adam":
+                return self._step_rs_ag_torch(local_batch)
+            if not getattr(self, "_rs_ag_warned", False):
+                print(f"[dp] rs_ag unsupported for {type(hs).__name__}; using allreduce")
+                self._rs_ag_warned = True
         if hs is not None:
             # fused path: gradient tensors are all-reduced as they become
             # final (g_bias after k_gc; the [M,n,d] weight grads in model
@@ -283,6 +297,210 @@ class DataParallelEnsembleTrainer:
             self._reducer.all_reduce_(grads)
         self.ensemble.apply_grads(grads)
         return loss_data, aux
+
+    # -- rs_ag (ZeRO-style) ---------------------------------------------------
+    def _step_rs_ag(self, hs, x: torch.Tensor):
+        """Fused-step rs_ag: weight-grad chunks are reduce-scattered into row
+        shards from the on_grads callback (launched on the comm stream so
+        they overlap the remaining grad GEMMs), the renorm-projected Adam
+        (k_project_adam) runs on the LOCAL rows only, and the updated
+        parameter rows are all-gathered.  Bias-sized grads ([M, n] and
+        smaller) are all-reduced and updated everywhere — they are <0.1% of
+        the bytes and the whole-bias L2-decay term does not shard."""
+        import contextlib
+
+        from sparse_coding_amd.engine.hip_step import EPS_NORM
+
+        ens = self.ensemble
+        world = max(self.world_size, 1)
+        rank = dist.get_rank(self.group) if (dist.is_initialized() and world > 1) else 0
+        d = hs.d_act
+        R = hs.n_models * hs.n_dict
+
+        comm_stream = None
+        if torch.cuda.is_available():
+            if not hasattr(self, "_comm_stream"):
+                self._comm_stream = torch.cuda.Stream()
+            comm_stream = self._comm_stream
+        if not hasattr(self, "_rs_shards"):
+            self._rs_shards = {}
+
+        w_pend = []  # (work|None, grad shard, kind, global row0, chunk rows)
+        b_pend = []  # (work, tensor)
+
+        def locate(t):
+            for kind, base in (("gw", hs.gw), ("gw_enc", getattr(hs, "gw_enc", None))):
+                if base is None:
+                    continue
+                off_b = t.data_ptr() - base.data_ptr()
+                if 0 <= off_b < base.numel() * 4:
+                    return kind, off_b // (4 * d)
+            raise RuntimeError("rs_ag: gradient tensor not in a known workspace")
+
+        def on_grads(tensors):
+            if comm_stream is not None:
+                ev = torch.cuda.Event()
+                ev.record()
+                comm_stream.wait_event(ev)
+            ctx = torch.cuda.stream(comm_stream) if comm_stream is not None else contextlib.nullcontext()
+            with ctx:
+                for t in tensors:
+                    if t.dim() == 2:  # bias-sized
+                        if world > 1:
+                            b_pend.append((dist.all_reduce(t, async_op=True, group=self.group), t))
+                    else:
+                        kind, row0 = locate(t)
+                        rows = t.shape[0] * t.shape[1]
+                        k = rows // world
+                        shard = self._rs_shards.get((kind, row0))
+                        if shard is None or shard.shape[0] != k:
+                            shard = torch.empty(k, d, device=t.device, dtype=torch.float32)
+                            self._rs_shards[(kind, row0)] = shard
+                        flat = t.reshape(rows, d)
+                        if world > 1:
+                            work = _reduce_scatter_rows(flat, shard, self.group)
+                        else:  # world-1 force path: same code shape, local copy
+                            shard.copy_(flat)
+                            work = None
+                        w_pend.append((work, shard, kind, row0, rows))
+
+        B = hs.grads_phase(x, on_grads=on_grads)
+
+        for work, *_ in w_pend:
+            if work is not None:
+                work.wait()
+        for work, _ in b_pend:
+            work.wait()
+        if comm_stream is not None:
+            torch.cuda.current_stream().wait_stream(comm_stream)
+        for _, t in b_pend:
+            t.div_(world)
+
+        st = ens.optim_states
+        st["step"] += 1.0
+        step0 = st["step"].reshape(-1)[0:1]
+        p = ens.params
+        self._rs_consolidate = []
+        norms_flat = hs.norms.reshape(R)
+        lrm_flat = hs.lr_mult.reshape(R)
+        for work, shard, kind, row0, rows in w_pend:
+            if world > 1:
+                shard.div_(world)
+            k = rows // world
+            g0 = row0 + rank * k
+            if hs.tied:
+                pname, project = "encoder", True
+            elif kind == "gw":
+                pname, project = "decoder", True
+            else:
+                pname, project = "encoder", False
+            Wflat = p[pname].reshape(R, d)
+            mu = st["mu"][pname].reshape(R, d)
+            nu = st["nu"][pname].reshape(R, d)
+            sl = slice(g0, g0 + k)
+            hs.ext.project_adam(Wflat[sl], shard, norms_flat[sl], mu[sl], nu[sl],
+                                step0, k, hs.lr, hs.beta1, hs.beta2, hs.eps,
+                                EPS_NORM, project, lr_mult=lrm_flat[sl])
+            if world > 1:
+                _all_gather_rows(Wflat[row0 : row0 + rows], Wflat[sl], self.group)
+            self._rs_consolidate.append((st["mu"][pname], row0, rows))
+            self._rs_consolidate.append((st["nu"][pname], row0, rows))
+        # bias params: identical full update on every rank
+        hs.ext.bias_adam(p["encoder_bias"], hs.g_bias, hs.bias_decay,
+                         st["mu"]["encoder_bias"], st["nu"]["encoder_bias"],
+                         st["step"], hs.lr, hs.beta1, hs.beta2, hs.eps,
+                         lr_mult=hs.lr_mult)
+        return hs._loss_data(B), {"c": hs.c}
+
+    def _step_rs_ag_torch(self, x: torch.Tensor):
+        """Generic-tree rs_ag for the torch backend (and the CPU/gloo tests):
+        weight leaves ([M, n, d] and deeper) are reduce-scattered by row and
+        Adam-updated on the local shard only; small leaves are all-reduced
+        and updated everywhere.  Matches functional.optim.adam's exact fp32
+        operation order so an rs_ag run is bit-identical to an allreduce
+        run (the gloo reduce-scatter fallback sums identically)."""
+        ens = self.ensemble
+        okw = ens.optimizer_kwargs
+        lr = float(okw.get("lr", 1e-3))
+        b1, b2 = okw.get("betas", (0.9, 0.999))
+        eps = float(okw.get("eps", 1e-8))
+        world = max(self.world_size, 1)
+        rank = dist.get_rank(self.group) if (dist.is_initialized() and world > 1) else 0
+
+        grads, (loss_data, aux) = ens.compute_grads(x)
+        st = ens.optim_states
+        st["step"] += 1.0
+        step_t = st["step"].reshape(-1)[0]
+        bc1 = 1.0 - b1 ** step_t
+        bc2 = 1.0 - b2 ** step_t
+
+        g_leaves, _ = tree_flatten(grads)
+        p_leaves, _ = tree_flatten(ens.params)
+        mu_leaves, _ = tree_flatten(st["mu"])
+        nu_leaves, _ = tree_flatten(st["nu"])
+
+        def adam_(pp, gg, mu, nu):
+            # same expressions as functional.optim.adam (bitwise-matching)
+            new_mu = b1 * mu + (1.0 - b1) * gg
+            new_nu = b2 * nu + (1.0 - b2) * gg * gg
+            mu.copy_(new_mu)
+            nu.copy_(new_nu)
+            pp.add_(-lr * (new_mu / bc1) / (torch.sqrt(new_nu / bc2) + eps))
+
+        self._rs_consolidate = []
+        for pp, gg, mu, nu in zip(p_leaves, g_leaves, mu_leaves, nu_leaves):
+            rows = pp.numel() // pp.shape[-1] if pp.dim() >= 1 else 1
+            if pp.dim() >= 3 and world > 1 and rows % world == 0:
+                dl = pp.shape[-1]
+                gflat = gg.reshape(rows, dl).contiguous()
+                k = rows // world
+                shard = torch.empty(k, dl, device=pp.device, dtype=pp.dtype)
+                _reduce_scatter_rows(gflat, shard, self.group).wait()
+                shard.div_(world)
+                sl = slice(rank * k, (rank + 1) * k)
+                pf = pp.reshape(rows, dl)
+                adam_(pf[sl], shard, mu.reshape(rows, dl)[sl], nu.reshape(rows, dl)[sl])
+                _all_gather_rows(pf, pf[sl].clone(), self.group)
+                self._rs_consolidate.append((mu, 0, rows))
+                self._rs_consolidate.append((nu, 0, rows))
+            else:
+                if world > 1:
+                    dist.all_reduce(gg, group=self.group)
+                    gg.div_(world)
+                adam_(pp, gg, mu, nu)
+        return loss_data, aux
+
+    def resample(self, resampler):
+        """DP-correct dead-feature resample: fired counts are summed across
+        ranks (each rank only saw its own batch shard) and the replacement
+        pool is taken from rank 0, so every replica applies the IDENTICAL
+        deterministic rewrite and stays bit-identical — no param broadcast
+        needed.  examples_seen is scaled to the global count for the
+        rate-threshold dead rule."""
+        if dist.is_initialized() and self.world_size > 1:
+            dist.all_reduce(resampler.fired, group=self.group)
+            dist.broadcast(resampler.pool_examples, src=0, group=self.group)
+            resampler.examples_seen *= self.world_size
+        return resampler.resample()
+
+    def consolidate_optim_state(self) -> None:
+        """rs_ag mode shards the Adam moments of the weight rows: each rank's
+        mu/nu are authoritative only for its own rows.  Call this before
+        unstack()/state_dict()/checkpointing to all-gather the full moments
+        onto every rank.  No-op in allreduce mode or at world 1."""
+        if self.dp_mode != "rs_ag" or self.world_size <= 1:
+            return
+        ops = getattr(self, "_rs_consolidate", None)
+        if not ops:
+            return
+        rank = dist.get_rank(self.group)
+        world = self.world_size
+        for mom, row0, rows in ops:
+            k = rows // world
+            d_last = mom.shape[-1]
+            flat = mom.reshape(-1, d_last)
+            sl = slice(row0 + rank * k, row0 + rank * k + k)
+            _all_gather_rows(flat[row0 : row0 + rows], flat[sl].clone(), self.group)
 
 
 def shard_batch(batch: torch.Tensor, rank: int, world_size: int) -> torch.Tensor:

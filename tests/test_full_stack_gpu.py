@@ -562,3 +562,37 @@ def test_anthropic_resample_warmup_gpu():
         rs.observe(x, aux)
     torch.cuda.synchronize()
     assert (hs.lr_mult == 1.0).all()
+
+
+@pytest.mark.parametrize("sig_name", ["tied", "untied"])
+def test_rs_ag_one_rank_matches_step(sig_name):
+    """dp_mode=rs_ag at world 1 (force path: same kernel/collective code
+    shape, local shard copies) must match the plain fused step bit-exactly."""
+    from sparse_coding_amd.engine.ensemble import FunctionalEnsemble
+    from sparse_coding_amd.functional.optim import adam
+    from sparse_coding_amd.models.sae_signatures import FunctionalSAE, FunctionalTiedSAE
+    from sparse_coding_amd.parallel.dp import DataParallelEnsembleTrainer
+
+    sig = FunctionalTiedSAE if sig_name == "tied" else FunctionalSAE
+    torch.manual_seed(51)
+    M, B, d, n = 4, 512, 128, 512
+    models = [sig.init(d, n, 1e-3, device=DEV) for _ in range(M)]
+    ens = FunctionalEnsemble(models, sig, adam, {"lr": 1e-3}, device=DEV, backend="hip")
+    models2 = [({k: v.clone() for k, v in p.items()}, {k: v.clone() for k, v in b.items()})
+               for p, b in ens.unstack()]
+    ens_ref = FunctionalEnsemble(models2, sig, adam, {"lr": 1e-3}, device=DEV, backend="hip")
+
+    trainer = DataParallelEnsembleTrainer(ens, force_dp_path=True, dp_mode="rs_ag")
+    x = torch.randn(B, d, device=DEV)
+    for _ in range(3):
+        losses, _ = trainer.step(x)
+        ens_ref.step_batch(x)
+    torch.cuda.synchronize()
+    assert torch.isfinite(losses["loss"]).all()
+    for k in ens_ref.params:
+        err = (ens.params[k] - ens_ref.params[k]).abs().max().item()
+        assert err == 0.0, (k, err)
+    for k in ("mu", "nu"):
+        for pk in ens_ref.optim_states[k]:
+            err = (ens.optim_states[k][pk] - ens_ref.optim_states[k][pk]).abs().max().item()
+            assert err == 0.0, (k, pk, err)

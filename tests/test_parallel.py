@@ -343,3 +343,125 @@ def test_dispatch_lite_collect():
     collect_lite([handle])
     assert not torch.allclose(ens.params["encoder"], before)
     assert torch.isfinite(ens.params["encoder"]).all()
+
+
+# ---------------------------------------------------------------------------
+# rs_ag (reduce-scatter + sharded Adam + all-gather) DP mode
+# ---------------------------------------------------------------------------
+
+def _rs_ag_worker(rank, world_size, port, batch, dp_mode, out_q):
+    try:
+        os.environ["MASTER_ADDR"] = "127.0.0.1"
+        os.environ["MASTER_PORT"] = str(port)
+        dist.init_process_group("gloo", rank=rank, world_size=world_size)
+
+        from sparse_coding_amd.parallel.dp import DataParallelEnsembleTrainer, shard_batch
+
+        ens = _make_ensemble(seed=0)
+        trainer = DataParallelEnsembleTrainer(ens, dp_mode=dp_mode)
+        for _ in range(3):
+            local = shard_batch(batch, rank, world_size)
+            trainer.step(local)
+        trainer.consolidate_optim_state()
+        if rank == 0:
+            out = {k: v.detach().cpu().numpy() for k, v in ens.params.items()}
+            out["_mu_encoder"] = ens.optim_states["mu"]["encoder"].cpu().numpy()
+            out["_nu_encoder"] = ens.optim_states["nu"]["encoder"].cpu().numpy()
+            out_q.put(out)
+        dist.destroy_process_group()
+    except Exception:  # noqa: BLE001
+        import traceback
+
+        out_q.put({"_error": f"rank {rank}: {traceback.format_exc()}"})
+        raise
+
+
+@pytest.mark.timeout(180)
+def test_rs_ag_matches_allreduce_and_single():
+    """dp_mode=rs_ag == dp_mode=allreduce == single process, params AND
+    (consolidated) Adam moments, on gloo world 2."""
+    torch.manual_seed(7)
+    batch = torch.randn(B, D)
+
+    results = {}
+    for i, mode in enumerate(("rs_ag", "allreduce")):
+        ctx = mp.get_context("spawn")
+        q = ctx.Queue()
+        procs = [ctx.Process(target=_rs_ag_worker, args=(r, 2, 29561 + i, batch, mode, q))
+                 for r in range(2)]
+        for p in procs:
+            p.start()
+        res = q.get(timeout=150)
+        assert "_error" not in res, res.get("_error")
+        for p in procs:
+            p.join(timeout=60)
+            assert p.exitcode == 0
+        results[mode] = res
+
+    for k in results["rs_ag"]:
+        a, b = torch.from_numpy(results["rs_ag"][k]), torch.from_numpy(results["allreduce"][k])
+        assert torch.allclose(a, b, atol=1e-7), (k, (a - b).abs().max())
+
+    ens = _make_ensemble(seed=0)
+    for _ in range(3):
+        ens.step_batch(batch)
+    for k, v in ens.params.items():
+        assert torch.allclose(torch.from_numpy(results["rs_ag"][k]), v, atol=1e-5), k
+    assert torch.allclose(torch.from_numpy(results["rs_ag"]["_mu_encoder"]),
+                          ens.optim_states["mu"]["encoder"], atol=1e-5)
+
+
+def _dp_resample_worker(rank, world_size, port, batch, out_q):
+    try:
+        os.environ["MASTER_ADDR"] = "127.0.0.1"
+        os.environ["MASTER_PORT"] = str(port)
+        dist.init_process_group("gloo", rank=rank, world_size=world_size)
+
+        from sparse_coding_amd.engine.resample import EnsembleResampler
+        from sparse_coding_amd.parallel.dp import DataParallelEnsembleTrainer, shard_batch
+
+        ens = _make_ensemble(seed=0)
+        with torch.no_grad():
+            ens.params["encoder_bias"][:, : N // 2] = -1e6
+        trainer = DataParallelEnsembleTrainer(ens)
+        rs = EnsembleResampler(ens, n_track=8, protocol="anthropic")
+        for _ in range(3):
+            local = shard_batch(batch, rank, world_size)
+            _, aux = trainer.step(local)
+            rs.observe(local, aux)
+        counts = trainer.resample(rs)
+        out_q.put((rank, counts.numpy(),
+                   {k: v.detach().cpu().numpy() for k, v in ens.params.items()}))
+        dist.destroy_process_group()
+    except Exception:  # noqa: BLE001
+        import traceback
+
+        out_q.put((rank, None, {"_error": f"rank {rank}: {traceback.format_exc()}"}))
+        raise
+
+
+@pytest.mark.timeout(180)
+def test_dp_resample_keeps_replicas_identical():
+    """trainer.resample(): fired summed across ranks, pool broadcast from
+    rank 0 — replicas must stay bit-identical after the rewrite."""
+    torch.manual_seed(9)
+    batch = torch.randn(B, D)
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    procs = [ctx.Process(target=_dp_resample_worker, args=(r, 2, 29566, batch, q))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    res = {}
+    for _ in range(2):
+        rank, counts, params = q.get(timeout=150)
+        assert "_error" not in params, params.get("_error")
+        res[rank] = (counts, params)
+    for p in procs:
+        p.join(timeout=60)
+        assert p.exitcode == 0
+    c0, p0 = res[0]
+    c1, p1 = res[1]
+    assert (c0 == c1).all() and c0.sum() > 0
+    for k in p0:
+        assert (p0[k] == p1[k]).all(), k
