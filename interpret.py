@@ -27,6 +27,7 @@ from sparse_coding_amd.interpret.interpret import (
     plot_scores,
     read_results,
 )
+from sparse_coding_amd.interpret import drivers as interp_drivers  # noqa: F401 (API parity)
 
 
 def _load_plugin(spec: str):
@@ -59,6 +60,38 @@ def main(argv=None):
     r = sub.add_parser("read_results")
     r.add_argument("--output-folder", default="interp_results")
     r.add_argument("--plot", default="scores.png")
+    r.add_argument("--score-mode", default="top", choices=["top", "random", "top_random", "all"])
+    r.add_argument("--protocol-layout", action="store_true",
+                   help="read the reference feature_N/explanation.txt layout (drivers.read_scores)")
+
+    # reference CLI surface (interpret.py:764-815): run / run_group / chunks
+    pr = sub.add_parser("run", help="full explain+simulate protocol on one learned dict")
+    pr.add_argument("--learned-dict", required=True)
+    pr.add_argument("--model-name", default="pythia-70m-deduped")
+    pr.add_argument("--layer", type=int, default=2)
+    pr.add_argument("--layer-loc", default="residual")
+    pr.add_argument("--n-feats-explain", type=int, default=10)
+    pr.add_argument("--df-n-feats", type=int, default=200)
+    pr.add_argument("--n-fragments", type=int, default=256)
+    pr.add_argument("--save-loc", default="auto_interp_results/run")
+    pr.add_argument("--client", default="mock", help="mock | hf:<local model path>")
+    pr.add_argument("--device", default="cuda:0" if torch.cuda.is_available() else "cpu")
+
+    g = sub.add_parser("run_group", help="split a learned_dicts.pt and run the protocol per dict")
+    g.add_argument("--results-loc", required=True)
+    g.add_argument("--model-name", default="pythia-70m-deduped")
+    g.add_argument("--layer", type=int, default=2)
+    g.add_argument("--layer-loc", default="residual")
+    g.add_argument("--n-feats-explain", type=int, default=10)
+    g.add_argument("--out-base", default="auto_interp_results")
+    g.add_argument("--device", default="cuda:0" if torch.cuda.is_available() else "cpu")
+
+    c = sub.add_parser("chunks", help="protocol across chunk-count checkpoints of a sweep dir")
+    c.add_argument("--base-dir", required=True)
+    c.add_argument("--save-dir", default="auto_interp_results_overtime")
+    c.add_argument("--l1-val", type=float, default=8.577e-4)
+    c.add_argument("--layer", type=int, default=2)
+    c.add_argument("--device", default="cuda:0" if torch.cuda.is_available() else "cpu")
 
     args = p.parse_args(argv)
 
@@ -91,11 +124,56 @@ def main(argv=None):
         print(f"wrote {len(results)} feature records to {args.output_folder}")
 
     elif args.cmd == "read_results":
-        recs = read_results(args.output_folder)
-        scores = [r["score"] for r in recs.values() if r.get("score") == r.get("score")]
-        print(f"{len(recs)} features, {len(scores)} scored")
-        if scores:
-            plot_scores({"results": scores}, save_path=args.plot)
+        if args.protocol_layout:
+            modes = ["top", "random", "top_random"] if args.score_mode == "all" else [args.score_mode]
+            for mode in modes:
+                by_name = interp_drivers.read_scores(args.output_folder, mode)
+                flat = {name: sc for name, (ndxs, sc) in by_name.items()}
+                print(f"[{mode}] " + ", ".join(f"{k}: n={len(v)}" for k, v in flat.items()))
+                if flat:
+                    plot_scores(flat, save_path=args.plot.replace(".png", f"_{mode}.png"))
+        else:
+            recs = read_results(args.output_folder)
+            scores = [r["score"] for r in recs.values() if r.get("score") == r.get("score")]
+            print(f"{len(recs)} features, {len(scores)} scored")
+            if scores:
+                plot_scores({"results": scores}, save_path=args.plot)
+
+    elif args.cmd in ("run", "run_group", "chunks"):
+        from sparse_coding_amd.config import InterpArgs
+
+        cfg = InterpArgs()
+        cfg.model_name = args.model_name if hasattr(args, "model_name") else cfg.model_name
+        cfg.layer = args.layer
+        cfg.device = args.device
+        if hasattr(args, "layer_loc"):
+            cfg.layer_loc = args.layer_loc
+        if hasattr(args, "n_feats_explain"):
+            cfg.n_feats_explain = args.n_feats_explain
+            cfg.df_n_feats = max(getattr(args, "df_n_feats", 200), args.n_feats_explain)
+        if args.cmd == "run":
+            client = None
+            if args.client.startswith("hf:"):
+                from transformers import AutoModelForCausalLM, AutoTokenizer
+
+                from sparse_coding_amd.interpret.protocol import HFLocalClient
+
+                path = args.client[3:]
+                m = AutoModelForCausalLM.from_pretrained(path).to(args.device)
+                t = AutoTokenizer.from_pretrained(path)
+                client = HFLocalClient(m, t, device=args.device)
+            cfg.save_loc = args.save_loc
+            ld = torch.load(args.learned_dict, map_location="cpu", weights_only=False)
+            if isinstance(ld, list):
+                ld = ld[0][0]
+            scores = interp_drivers.run(ld, cfg, client=client, n_fragments=args.n_fragments)
+            print(f"scored {len(scores)} features -> {cfg.save_loc}")
+        elif args.cmd == "run_group":
+            out = interp_drivers.run_from_grouped(cfg, args.results_loc, out_base=args.out_base)
+            print(f"ran {len(out)} dicts -> {args.out_base}")
+        else:
+            out = interp_drivers.interpret_across_chunks(args.base_dir, args.save_dir, cfg, args.l1_val)
+            print(f"ran {len(out)} (dict, chunk) combos -> {args.save_dir}")
 
 
 if __name__ == "__main__":

@@ -34,6 +34,7 @@ LAYER_LOCS = ["residual", "mlp", "attn", "attn_concat", "mlpout"]
 
 # (family, hidden, n_layer, n_head, intermediate, vocab)
 MODEL_TABLE: Dict[str, Tuple[str, int, int, int, int, int]] = {
+    "tiny-gptneox": ("gptneox", 64, 2, 4, 256, 512),  # test-scale host LM
     "pythia-70m": ("gptneox", 512, 6, 8, 2048, 50304),
     "pythia-70m-deduped": ("gptneox", 512, 6, 8, 2048, 50304),
     "EleutherAI/pythia-70m-deduped": ("gptneox", 512, 6, 8, 2048, 50304),

@@ -1,0 +1,200 @@
+"""The native explain/simulate autointerp protocol (interpret/protocol.py)
+and the folder/grouped/chunk drivers (interpret/drivers.py), fully offline
+via the deterministic MockLLMClient.  Mirrors reference interpret.py:265-688."""
+
+import os
+
+import numpy as np
+import pytest
+import torch
+
+from sparse_coding_amd.interpret.interpret import ActivationRecord
+from sparse_coding_amd.interpret.protocol import (
+    EXAMPLES_PER_SPLIT,
+    TOTAL_EXAMPLES,
+    MockLLMClient,
+    ScoredSimulation,
+    build_explainer_prompt,
+    build_simulator_prompt,
+    calculate_max_activation,
+    explain_and_score,
+    interpret_protocol,
+    parse_simulation,
+    split_records,
+)
+
+FRAG = 16
+
+
+def _selective_records(token: str, n: int, active: bool, seed: int = 0):
+    """Fragments whose feature activates exactly on `token`."""
+    rng = np.random.default_rng(seed)
+    recs = []
+    for i in range(n):
+        toks = [f"tok{j}" for j in rng.integers(0, 20, FRAG)]
+        if active:
+            toks[int(rng.integers(0, FRAG))] = token
+        acts = [5.0 if t == token else 0.0 for t in toks]
+        recs.append(ActivationRecord(tokens=toks, activations=acts))
+    return recs
+
+
+def test_split_records_shapes():
+    top = _selective_records("X", TOTAL_EXAMPLES + 5, True)
+    rand = _selective_records("X", TOTAL_EXAMPLES, True, seed=1)
+    split = split_records(top, rand)
+    assert len(split.train) == TOTAL_EXAMPLES - EXAMPLES_PER_SPLIT
+    assert len(split.valid_top) == EXAMPLES_PER_SPLIT
+    assert len(split.valid_random) == EXAMPLES_PER_SPLIT
+    assert len(split.valid) == 10  # reference asserts 10 scored sequences
+
+
+def test_prompts_and_parse_roundtrip():
+    recs = _selective_records("needle", 3, True)
+    prompt = build_explainer_prompt(recs, calculate_max_activation(recs))
+    assert "needle\t10" in prompt
+    sim_prompt = build_simulator_prompt(" tokens: 'needle'", recs[0].tokens)
+    assert sim_prompt.count("<level>") == FRAG
+    fake = "\n".join(f"{t}\t{7 if t == 'needle' else 0}" for t in recs[0].tokens)
+    levels = parse_simulation(fake, recs[0].tokens)
+    assert len(levels) == FRAG
+    assert max(levels) == 7.0
+
+
+def test_parse_simulation_malformed():
+    toks = ["a", "b", "c"]
+    assert parse_simulation("garbage no tabs", toks) == [0.0, 0.0, 0.0]
+    assert parse_simulation("a\t3\nb\tnope\nc\t99", toks) == [3.0, 0.0, 10.0]
+
+
+def test_selective_feature_scores_high_noise_scores_low():
+    client = MockLLMClient()
+    top = _selective_records("needle", TOTAL_EXAMPLES, True)
+    rand = _selective_records("needle", TOTAL_EXAMPLES, True, seed=2)
+    expl, sim = explain_and_score(split_records(top, rand), client)
+    assert "'needle'" in expl
+    assert sim.score() > 0.8
+    assert sim.top_only_score() > 0.8
+
+    # noise feature: activations uncorrelated with any token identity
+    rng = np.random.default_rng(3)
+    noise = [ActivationRecord(tokens=[f"tok{j}" for j in rng.integers(0, 20, FRAG)],
+                              activations=list(rng.random(FRAG)))
+             for _ in range(2 * TOTAL_EXAMPLES)]
+    _, sim_n = explain_and_score(split_records(noise[:TOTAL_EXAMPLES], noise[TOTAL_EXAMPLES:]), client)
+    assert sim_n.score() < 0.5
+
+
+def _toy_table(n_frag=200, n_feats=4, seed=0):
+    """Activation table where feature 0 fires on token id 7, feature 1 on
+    token id 13, the rest never fire."""
+    rng = np.random.default_rng(seed)
+    tokens = torch.from_numpy(rng.integers(0, 32, (n_frag, FRAG)))
+    acts = torch.zeros(n_frag, FRAG, n_feats)
+    acts[:, :, 0] = (tokens == 7).float() * 4.0
+    acts[:, :, 1] = (tokens == 13).float() * 2.0
+    return acts, tokens
+
+
+def test_interpret_protocol_end_to_end(tmp_path):
+    acts, tokens = _toy_table()
+    scores = interpret_protocol(acts, tokens, [0, 1, 2], MockLLMClient(), str(tmp_path))
+    # features 0/1 are selective -> scored; feature 2 never fires -> skipped
+    assert scores[0] > 0.8 and scores[1] > 0.8
+    assert 2 not in scores
+    # reference on-disk layout
+    txt = open(tmp_path / "feature_0" / "explanation.txt").read()
+    assert "Score: " in txt and "Top only score:" in txt and "Random only score:" in txt
+    assert (tmp_path / "feature_0" / "scored_simulation.pkl").exists()
+    assert (tmp_path / "feature_0" / "neuron_record.pkl").exists()
+    assert (tmp_path / "feature_2").exists()  # placeholder skip marker
+    # resumability: second call skips everything
+    scores2 = interpret_protocol(acts, tokens, [0, 1, 2], MockLLMClient(), str(tmp_path))
+    assert scores2 == {}
+
+
+def test_score_readers(tmp_path):
+    from sparse_coding_amd.interpret.drivers import get_score, read_scores, read_transform_scores
+
+    acts, tokens = _toy_table()
+    tdir = tmp_path / "sparse_coding"
+    interpret_protocol(acts, tokens, [0, 1], MockLLMClient(), str(tdir))
+    ndxs, scores = read_transform_scores(str(tdir), "top_random")
+    assert sorted(ndxs) == [0, 1]
+    assert all(s > 0.8 for s in scores)
+    for mode in ("top", "random", "top_random"):
+        lines = open(tdir / "feature_0" / "explanation.txt").read().split("\n")
+        assert isinstance(get_score(lines, mode), float)
+    allscores = read_scores(str(tmp_path), "top")
+    assert list(allscores.keys())[0] == "sparse_coding"
+
+
+def _tiny_cfg(tmp_path):
+    from sparse_coding_amd.config import InterpArgs
+
+    cfg = InterpArgs()
+    cfg.model_name = "tiny-gptneox"
+    cfg.layer = 1
+    cfg.layer_loc = "residual"
+    cfg.device = "cpu"
+    cfg.df_n_feats = 16
+    cfg.n_feats_explain = 2
+    cfg.save_loc = str(tmp_path / "out")
+    return cfg
+
+
+def _tiny_learned_dict(d=64, n=32):
+    from sparse_coding_amd.models.learned_dict import TiedSAE
+
+    torch.manual_seed(0)
+    return TiedSAE(torch.randn(n, d), torch.zeros(n))
+
+
+@pytest.mark.timeout(300)
+def test_run_driver_tiny_model(tmp_path):
+    """run(): fragment table from a tiny host LM + protocol over 2 features."""
+    from sparse_coding_amd.interpret.drivers import run
+
+    cfg = _tiny_cfg(tmp_path)
+    run(_tiny_learned_dict(), cfg, n_fragments=16)
+    assert os.path.exists(cfg.save_loc + "/fragment_table.pt")
+    feats = [f for f in os.listdir(cfg.save_loc) if f.startswith("feature_")]
+    assert len(feats) == 2
+
+
+@pytest.mark.timeout(300)
+def test_run_from_grouped(tmp_path):
+    from sparse_coding_amd.interpret.drivers import make_tag_name, run_from_grouped
+
+    lds = [(_tiny_learned_dict(), {"tied": True, "dict_size": 32, "l1_alpha": 8.5e-4})]
+    loc = tmp_path / "learned_dicts.pt"
+    torch.save(lds, loc)
+    cfg = _tiny_cfg(tmp_path)
+    out = run_from_grouped(cfg, str(loc), out_base=str(tmp_path / "grouped"), n_fragments=8)
+    assert len(out) == 1
+    name = make_tag_name(lds[0][1]) + ".pt"
+    assert name in out
+
+
+def test_parse_folder_name():
+    from sparse_coding_amd.interpret.drivers import parse_folder_name
+
+    assert parse_folder_name("tied_residual_l2_r4") == ("tied", "residual", 2, 4.0, "")
+    assert parse_folder_name("untied_mlp_l5_r0") == ("untied", "mlp", 5, 0.5, "")
+    assert parse_folder_name("tied_residual_l2_r2_long") == ("tied", "residual", 2, 2.0, "long")
+
+
+@pytest.mark.timeout(300)
+def test_interpret_across_chunks(tmp_path):
+    """Chunk-count driver over a synthetic sweep layout."""
+    from sparse_coding_amd.interpret.drivers import interpret_across_chunks
+
+    base = tmp_path / "sweep"
+    for nc in (1, 4):
+        d = base / "tied_residual_l1_r2" / f"_{nc - 1}"
+        os.makedirs(d, exist_ok=True)
+        torch.save([(_tiny_learned_dict(), {"l1_alpha": 8.5e-4})], d / "learned_dicts.pt")
+    cfg = _tiny_cfg(tmp_path)
+    out = interpret_across_chunks(str(base), str(tmp_path / "res"), cfg, 8.5e-4,
+                                  chunks=(1, 4), n_fragments=8)
+    assert len(out) == 2
