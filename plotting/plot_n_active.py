@@ -82,3 +82,72 @@ def main():
 
 if __name__ == "__main__":
     main()
+
+
+# ---------------------------------------------------------------------------
+# shared helpers for the named n-active family scripts (reference
+# plot_n_active*.py all compute: per saved checkpoint, per learned dict,
+# the fraction of features active >threshold times on one chunk sample)
+# ---------------------------------------------------------------------------
+
+def frac_alive_series(dicts_path: str, sample: torch.Tensor, device: str = "cpu",
+                      threshold: int = 10, batch_size: int = 16384):
+    """[(l1_alpha, frac_alive, n_alive, n_feats), ...] for one
+    learned_dicts.pt (reference plot_n_active_over_time.py:52-77)."""
+    out = []
+    dicts = torch.load(dicts_path, map_location="cpu", weights_only=False)
+    for ld, hp in dicts:
+        ld.to_device(device)
+        s = sample.to(device)
+        n_active_count = torch.zeros(ld.n_feats, device=device)
+        for i in range(0, len(s), batch_size):
+            code = ld.encode(s[i : i + batch_size])
+            n_active_count += (code > 0).sum(dim=0).float()
+        n_alive = int((n_active_count > threshold).sum().item())
+        l1a = hp.get("l1_alpha", 0) or 8e-5  # reference maps l1=0 -> 8e-5
+        out.append((l1a, n_alive / ld.n_feats, n_alive, ld.n_feats))
+    return out
+
+
+def series_over_checkpoints(load_dir: str, folder: str, epochs, sample,
+                            device: str = "cpu", threshold: int = 10):
+    """[(epoch, [(l1, frac), ...]), ...] over _{epoch}/learned_dicts.pt."""
+    data = []
+    for epoch in epochs:
+        path = os.path.join(load_dir, folder, f"_{epoch}", "learned_dicts.pt")
+        if not os.path.exists(path):
+            continue
+        rows = frac_alive_series(path, sample, device, threshold)
+        data.append((epoch, [(l1, frac) for l1, frac, *_ in rows]))
+    return data
+
+
+def two_panel_alive_plot(series, save_path: str, title: str, abs_scale: float = 1.0):
+    """Reference's two-panel layout (:100-119): fraction + absolute counts
+    vs l1, one line per series key."""
+    import matplotlib
+
+    matplotlib.use("Agg")
+    import matplotlib.pyplot as plt
+
+    fig, (ax, ax2) = plt.subplots(1, 2, figsize=plt.figaspect(1 / 3))
+    for key, rows in series:
+        if not rows:
+            continue
+        ax.plot(*zip(*rows), label=str(key))
+        ax2.plot(*zip(*[(l1, abs_scale * f) for l1, f in rows]), label=str(key))
+    for a, ylab in ((ax, "Fraction of features alive"), (ax2, "Number of features alive")):
+        a.set_xscale("log")
+        a.set_xlabel("L1 Alpha")
+        a.set_ylabel(ylab)
+        a.legend(fontsize=7)
+    ax.set_title(title)
+    fig.tight_layout()
+    fig.savefig(save_path)
+    return fig
+
+
+def load_sample(chunk_path: str, n_samples: int = 50000) -> torch.Tensor:
+    chunk = torch.load(chunk_path, map_location="cpu", weights_only=False).float()
+    idx = np.random.choice(len(chunk), size=min(n_samples, len(chunk)), replace=False)
+    return chunk[idx]

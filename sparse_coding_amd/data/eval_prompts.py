@@ -101,3 +101,26 @@ def generate_induction_dataset(tokenizer, n_prompts: int = 64, seq_len: int = 32
     vocab = tokenizer.vocab_size if tokenizer is not None else 50257
     half = torch.randint(0, vocab, (n_prompts, seq_len // 2), generator=g)
     return torch.cat([half, half], dim=1)
+
+
+def gender_prompt_batch(n: int = 256, vocab_size: int = 50304, seq_len: int = 16,
+                        pool_size: int = 10, seed: int = 0) -> Tuple[torch.Tensor, torch.Tensor]:
+    """Tokenizer-free synthetic version of the gender prompt set for the
+    erasure study (reference preprocess_gender_dataset.py feeds real names;
+    this no-network environment substitutes two disjoint token-id pools as
+    the "names", which makes the concept linearly decodable from the host
+    LM's embeddings — exactly what the erasure pipeline needs).
+
+    Returns (tokens [n, seq_len], labels [n]) where the name token sits at a
+    fixed template position and the label is the pool it came from."""
+    rng = np.random.default_rng(seed)
+    lo = 100  # avoid special-token ids
+    female_pool = rng.choice(np.arange(lo, vocab_size // 2), size=pool_size, replace=False)
+    male_pool = rng.choice(np.arange(vocab_size // 2, vocab_size - 1), size=pool_size, replace=False)
+    tokens = rng.integers(lo, vocab_size - 1, size=(n, seq_len))
+    labels = rng.integers(0, 2, size=n)
+    name_pos = seq_len // 2
+    for i in range(n):
+        pool = female_pool if labels[i] == 1 else male_pool
+        tokens[i, name_pos] = rng.choice(pool)
+    return torch.from_numpy(tokens).long(), torch.from_numpy(labels).long()

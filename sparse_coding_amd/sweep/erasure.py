@@ -39,23 +39,74 @@ def ablate_top_features(learned_dict, acts: torch.Tensor, labels: torch.Tensor, 
 
 
 def leace_erase(acts: torch.Tensor, labels: torch.Tensor, rank: int = 1) -> torch.Tensor:
-    """Rank-k least-squares concept erasure: project out the top directions
-    of the class-mean difference in whitened space (LEACE, Belrose et al.)."""
-    x = acts - acts.mean(dim=0)
-    y = labels.float() - labels.float().mean()
-    # cross-covariance direction(s)
-    sigma = (x.T @ x) / x.shape[0] + 1e-4 * torch.eye(x.shape[1])
-    cross = (x * y[:, None]).mean(dim=0, keepdim=True).T  # [d, 1]
-    # whitened projection directions
-    evals, evecs = torch.linalg.eigh(sigma)
-    w_inv_half = evecs @ torch.diag(evals.clamp_min(1e-6).rsqrt()) @ evecs.T
-    z = w_inv_half @ cross
-    q, _ = torch.linalg.qr(z)
-    q = q[:, :rank]
-    # erase in whitened space, unwhiten
-    w_half = evecs @ torch.diag(evals.clamp_min(1e-6).sqrt()) @ evecs.T
-    proj = w_inv_half @ q @ q.T @ w_half
-    return acts - x @ proj.T
+    """One-shot LEACE erasure (fit + apply on the same data); see
+    :class:`LeaceEraser` for the estimator."""
+    return LeaceEraser.fit(acts, labels, rank=rank)(acts)
+
+
+class LeaceEraser:
+    """The full LEACE estimator (Belrose et al. 2023, arXiv:2306.03819):
+    the least-squares-optimal affine eraser
+        r(x) = x - Sigma^{1/2} P Sigma^{-1/2} (x - mu)
+    with P the orthogonal projection onto the whitened cross-covariance
+    colspace W @ Sigma_xz.  Fit once, apply anywhere (train/eval split —
+    the reference's erasure studies fit on one prompt set and score
+    transfer on another, plotting/erasure_plot.py:199-215).
+
+    ``labels`` may be binary {0,1} (rank-1 eraser) or integer classes
+    (one-hot z, rank up to C-1)."""
+
+    def __init__(self, eraser: torch.Tensor, mu: torch.Tensor):
+        self.eraser = eraser  # [d, d]
+        self.mu = mu
+
+    @staticmethod
+    def fit(acts: torch.Tensor, labels: torch.Tensor, rank: Optional[int] = None) -> "LeaceEraser":
+        x = acts - acts.mean(dim=0)
+        classes = labels.unique()
+        if len(classes) <= 2:
+            z = (labels.float() - labels.float().mean()).unsqueeze(1)
+        else:
+            z = torch.nn.functional.one_hot(labels.long()).float()
+            z = z - z.mean(dim=0)
+        sigma = (x.T @ x) / x.shape[0] + 1e-4 * torch.eye(x.shape[1])
+        cross = (x.T @ z) / x.shape[0]  # [d, C]
+        evals, evecs = torch.linalg.eigh(sigma)
+        w_inv_half = evecs @ torch.diag(evals.clamp_min(1e-6).rsqrt()) @ evecs.T
+        w_half = evecs @ torch.diag(evals.clamp_min(1e-6).sqrt()) @ evecs.T
+        q, _ = torch.linalg.qr(w_inv_half @ cross)
+        if rank is not None:
+            q = q[:, :rank]
+        eraser = w_half @ q @ q.T @ w_inv_half  # Sigma^{1/2} P Sigma^{-1/2}
+        return LeaceEraser(eraser, acts.mean(dim=0))
+
+    def __call__(self, acts: torch.Tensor) -> torch.Tensor:
+        return acts - (acts - self.mu) @ self.eraser.T
+
+    def mean_edit(self, acts: torch.Tensor) -> float:
+        return (self(acts) - acts).norm(dim=-1).mean().item()
+
+
+def mean_erase(acts: torch.Tensor, labels: torch.Tensor,
+               fit_acts: Optional[torch.Tensor] = None,
+               fit_labels: Optional[torch.Tensor] = None,
+               affine: bool = False) -> torch.Tensor:
+    """The reference study's "Mean" erasers (plotting/erasure_plot.py:140-152):
+    project out the class-mean-difference direction; ``affine=True`` instead
+    translates every class's mean onto the global mean (an affine edit)."""
+    fa = acts if fit_acts is None else fit_acts
+    fl = labels if fit_labels is None else fit_labels
+    mu0 = fa[fl == 0].mean(dim=0)
+    mu1 = fa[fl == 1].mean(dim=0)
+    if affine:
+        mu = fa.mean(dim=0)
+        out = acts.clone()
+        out[labels == 0] += mu - mu0
+        out[labels == 1] += mu - mu1
+        return out
+    v = mu1 - mu0
+    v = v / v.norm().clamp_min(1e-8)
+    return acts - (acts @ v)[:, None] * v[None, :]
 
 
 def erasure_curves(
