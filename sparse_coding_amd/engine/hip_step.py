@@ -131,9 +131,11 @@ def maybe_make_step(ensemble, required: bool = False) -> Optional["HipSAEStep"]:
             raise RuntimeError(f"backend='hip' requested but signature {sig.__name__} has no fused step yet")
         return None
     if tied and not _identity_centering(ensemble.buffers):
-        if required:
-            raise RuntimeError("fused tied step requires identity centering buffers")
-        return None
+        # general affine whitening-centering (K7): PCA-whitened-input tied
+        # SAE (reference sae_ensemble.py:98-132) — tied pipeline on a
+        # precomputed per-model whitened input
+        ext = _ops.get_extension(required=True)
+        return HipWhitenedStep(ensemble, ext)
 
     ext = _ops.get_extension(required=True)  # loud on GPU
     return HipSAEStep(ensemble, ext, tied=tied)
@@ -476,6 +478,70 @@ class HipCenteredStep(HipSAEStep):
 
     def dp_grad_tensors(self):
         return [self.gw, self.g_bias, self.g_center]
+
+
+class HipWhitenedStep(HipSAEStep):
+    """Fused step for FunctionalTiedSAE with GENERAL affine
+    whitening-centering buffers (K7; reference sae_ensemble.py:98-132,
+    :127-132 center / :148 loss-in-centered-space).
+
+    The buffers are non-trainable, and the loss lives entirely in the
+    centered space x' = ((x - t[m]) @ rot[m]^T) * s[m], so the step is the
+    plain tied pipeline on a per-model centered input ([M, B, d], same
+    model-stride kernels as HipCenteredStep) after ONE batched [d, d] GEMM:
+    x' = (x - t) @ Wc with Wc = (rot * s[:, None])^T precomputed."""
+
+    def __init__(self, ensemble, ext):
+        super().__init__(ensemble, ext, tied=True)
+        b = ensemble.buffers
+        rot = b["center_rot"].detach().float()
+        scale = b["center_scale"].detach().float()
+        self.trans = b["center_trans"].detach().float().contiguous()
+        # result[b, c] = sum_u (x-t)[b, u] * rot[c, u] * s[c]
+        self.Wc = (rot * scale.unsqueeze(-1)).transpose(1, 2).contiguous()  # [M, d, d]
+
+    def _alloc(self, B: int):
+        super()._alloc(B)
+        M, n, d = self.n_models, self.n_dict, self.d_act
+        dev = self.ens.params["encoder"].device
+        self.xsub = torch.empty(M, B, d, device=dev)
+        self.xc = torch.empty(M, B, d, device=dev)
+        # the per-model-x-stride kernels live in the "t" staging path
+        self.kc["staging"] = "t"
+
+    def grads_phase(self, x: torch.Tensor, on_grads=None):
+        ens, ext = self.ens, self.ext
+        B = x.shape[0]
+        if self._B != B:
+            self._alloc(B)
+        x = x.contiguous()
+        p = ens.params
+        enc = p["encoder"]
+        bias = p["encoder_bias"]
+        bk, prio = self.kc["bk"], self.kc["prio"]
+        bk_gw = self.kc["bk_grad_w"] or bk
+
+        self.loss_parts.zero_()
+        self.g_bias.zero_()
+
+        torch.sub(x.unsqueeze(0), self.trans.unsqueeze(1), out=self.xsub)
+        torch.bmm(self.xsub, self.Wc, out=self.xc)  # library GEMM: plain batched matmul
+
+        ext.row_norms(enc, self.norms, self.inv_norms, EPS_NORM)
+        ext.enc_fwd(self.xc, enc, bias, self.inv_norms, self.c,
+                    self.loss_parts, self.fired, 0, bk, prio)
+        ext.dec_fwd(self.c, enc, self.inv_norms, self.xc, self.r,
+                    self.loss_parts, self.kc["bk_dec"] or bk, prio)
+        ext.gc(self.r, enc, self.inv_norms, self.c, self.l1_alpha,
+               self.gpre, self.g_bias, bk, prio)
+
+        gscale = 2.0 / (B * self.d_act)
+        ext.grad_w(self.c, self.r, self.gw, gscale, 0.0, bk_gw, prio)
+        ext.grad_w(self.gpre, self.xc, self.gw, 1.0, 1.0, bk_gw, prio)
+        if on_grads is not None:
+            on_grads([self.gw, self.g_bias])
+        return B
+    # update_phase / _loss_data: inherited tied versions (incl. bias decay)
 
 
 class HipPositiveStep(HipSAEStep):

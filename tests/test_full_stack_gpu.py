@@ -598,3 +598,40 @@ def test_rs_ag_one_rank_matches_step(sig_name):
         for pk in ens_ref.optim_states[k]:
             err = (ens.optim_states[k][pk] - ens_ref.optim_states[k][pk]).abs().max().item()
             assert err < 1e-5, (k, pk, err)
+
+
+def test_whitened_tied_step_matches_oracle():
+    """HipWhitenedStep (general affine centering, K7) vs the vmap oracle:
+    same losses and same trained weights over several steps."""
+    from sparse_coding_amd.engine.ensemble import FunctionalEnsemble
+    from sparse_coding_amd.engine.hip_step import HipWhitenedStep
+    from sparse_coding_amd.functional.optim import adam
+    from sparse_coding_amd.models.sae_signatures import FunctionalTiedSAE
+
+    torch.manual_seed(61)
+    M, B, d, n = 3, 256, 64, 192
+    models = []
+    for i in range(M):
+        q, _ = torch.linalg.qr(torch.randn(d, d))
+        models.append(FunctionalTiedSAE.init(
+            d, n, 10 ** (-4 + 0.3 * i), device=DEV,
+            rotation=q.to(DEV), translation=torch.randn(d, device=DEV) * 0.3,
+            scaling=(torch.rand(d, device=DEV) + 0.5)))
+    ens_hip = FunctionalEnsemble(models, FunctionalTiedSAE, adam, {"lr": 1e-3}, device=DEV, backend="hip")
+    assert isinstance(ens_hip._hip_step, HipWhitenedStep)
+    models2 = [({k: v.clone() for k, v in p.items()}, {k: v.clone() for k, v in b.items()})
+               for p, b in ens_hip.unstack()]
+    ens_ref = FunctionalEnsemble(models2, FunctionalTiedSAE, adam, {"lr": 1e-3}, device=DEV, backend="torch")
+
+    x = torch.randn(B, d, device=DEV)
+    for i in range(4):
+        l_hip, aux_hip = ens_hip.step_batch(x)
+        l_ref, aux_ref = ens_ref.step_batch(x)
+        for k in ("loss", "l_reconstruction", "l_l1"):
+            err = (l_hip[k] - l_ref[k]).abs().max() / l_ref[k].abs().max().clamp_min(1e-9)
+            assert err < 1e-3, (i, k, err)
+    torch.cuda.synchronize()
+    err = (ens_hip.params["encoder"] - ens_ref.params["encoder"]).abs().max()
+    assert err < 2e-3, err
+    err_b = (ens_hip.params["encoder_bias"] - ens_ref.params["encoder_bias"]).abs().max()
+    assert err_b < 2e-3, err_b
