@@ -403,3 +403,124 @@ def scan_layer_moments(
         return [_layer_moments_job(j) for j in jobs]
     with mp.get_context("spawn").Pool(min(n_procs, len(jobs))) as pool:
         return pool.map(_layer_moments_job, jobs)
+
+
+# ---------------------------------------------------------------------------
+# round-2 stragglers (VERDICT.md missing #5): reference :811-842, :382-409,
+# :534-568, :582-619
+# ---------------------------------------------------------------------------
+
+def run_mmcs_with_larger(learned_dicts, threshold: float = 0.9, device="cpu"):
+    """Hungarian-matched MMCS of each dict against the next-larger dict in a
+    (n_l1 x n_sizes) grid of raw dictionaries (reference :811-842).
+
+    ``learned_dicts[l1][size]`` is a [n_feats, d] tensor (or LearnedDict);
+    returns (mean-matched-cos [n_l1, n_sizes], %-features-above-threshold
+    [n_l1, n_sizes], per-cell matched-cos arrays for histograms)."""
+    from scipy.optimize import linear_sum_assignment
+
+    def as_tensor(x):
+        return x.get_learned_dict() if hasattr(x, "get_learned_dict") else x
+
+    n_l1, n_sizes = len(learned_dicts), len(learned_dicts[0])
+    av = np.zeros((n_l1, n_sizes))
+    above = np.zeros((n_l1, n_sizes))
+    hists = np.empty((n_l1, max(n_sizes - 1, 1)), dtype=object)
+    for l1_ndx in range(n_l1):
+        for size_ndx in range(n_sizes - 1):
+            small = as_tensor(learned_dicts[l1_ndx][size_ndx]).to(device).float()
+            large = as_tensor(learned_dicts[l1_ndx][size_ndx + 1]).to(device).float()
+            sn = small / small.norm(dim=-1, keepdim=True).clamp_min(1e-8)
+            ln = large / large.norm(dim=-1, keepdim=True).clamp_min(1e-8)
+            cos = (sn @ ln.T).cpu().numpy()  # one GEMM, not a python loop
+            row_ind, col_ind = linear_sum_assignment(1 - cos)
+            matched = cos[row_ind, col_ind]
+            av[l1_ndx, size_ndx] = matched.mean()
+            above[l1_ndx, size_ndx] = (matched > threshold).sum() / small.shape[0] * 100
+            hists[l1_ndx][size_ndx] = matched
+    return av, above, hists
+
+
+def plot_capacity_scatter(dicts: List[Tuple[LearnedDict, Dict[str, Any]]],
+                          show: bool = False, save_name: str = "capacity_scatter") -> None:
+    """Per-dict capacity-per-feature scatter + pooled histogram
+    (reference :382-409)."""
+    import matplotlib
+
+    matplotlib.use("Agg")
+    import matplotlib.pyplot as plt
+
+    all_capacities = []
+    for i, (ld, _hparams) in enumerate(dicts):
+        capacities = capacity_per_feature(ld)
+        fig, ax = plt.subplots()
+        ax.scatter(range(len(capacities)), capacities.cpu())
+        ax.set_xlabel("Learned feature")
+        ax.set_ylabel("Capacity")
+        ax.set_title(f"Capacity per feature - {save_name}")
+        fig.savefig(f"{save_name}_{i}.png")
+        plt.close(fig)
+        all_capacities.append(capacities)
+    fig, ax = plt.subplots()
+    ax.hist(torch.cat(all_capacities).flatten().cpu(), bins=80)
+    ax.set_xlabel("Capacity")
+    ax.set_ylabel("Frequency")
+    ax.set_title(f"Capacity histogram - {save_name}")
+    fig.savefig(f"{save_name}_hist.png")
+    plt.close(fig)
+
+
+def cluster_vectors(model: LearnedDict, n_clusters: int = 1000, top_clusters: int = 10,
+                    save_loc: str = "outputs/top_clusters.txt", perplexity: float = 30.0):
+    """t-SNE -> k-means over dictionary directions; writes the member ids of
+    the most-populated clusters, one cluster per line (reference :534-568)."""
+    import os
+
+    from sklearn.cluster import KMeans
+    from sklearn.manifold import TSNE
+
+    d = model.get_learned_dict().detach().cpu().numpy()
+    n = d.shape[0]
+    n_clusters = min(n_clusters, n)
+    tsne = TSNE(n_components=2, random_state=0,
+                perplexity=min(perplexity, max(n - 1, 1)))
+    emb = tsne.fit_transform(d)
+    km = KMeans(n_clusters=n_clusters, random_state=0, n_init=4).fit(emb)
+    ids, counts = np.unique(km.labels_, return_counts=True)
+    order = np.argsort(counts)[::-1]
+    top_points = [np.where(km.labels_ == cid)[0] for cid in ids[order][:top_clusters]]
+    os.makedirs(os.path.dirname(save_loc) or ".", exist_ok=True)
+    with open(save_loc, "w") as f:
+        for cluster in top_points:
+            f.write(f"{[int(i) for i in cluster]}\n")
+    return top_points
+
+
+def make_one_chunk_per_layer(model_name: str = "pythia-70m-deduped",
+                             out_root: str = "single_chunks",
+                             layer_locs=("residual", "mlp", "mlpout", "attn"),
+                             n_layers: int = 6, device: str = "cuda:0",
+                             dataset_name: str = "synthetic",
+                             chunk_size_gb: float = 0.05, **kw) -> None:
+    """One activation chunk per (layer, loc) in the l{N}_{loc} folder layout
+    the plotting/baseline scripts consume (reference :582-601; the Pile
+    stream is replaced by setup_data's network-free token source)."""
+    from sparse_coding_amd.data.activation_dataset import load_model, setup_data
+
+    model = load_model(model_name, device=device)
+    for layer_loc in layer_locs:
+        for layer in range(n_layers):
+            setup_data(None, model, dataset_name,
+                       f"{out_root}/l{layer}_{layer_loc}",
+                       layer=layer, layer_loc=layer_loc, n_chunks=1,
+                       chunk_size_gb=chunk_size_gb, device=device,
+                       model_name=model_name, **kw)
+
+
+def make_one_chunk_per_layer_gpt2sm(out_root: str = "single_chunks_gpt2sm",
+                                    device: str = "cuda:0",
+                                    chunk_size_gb: float = 0.05, **kw) -> None:
+    """GPT-2-small variant (reference :603-619): 12 residual layers."""
+    make_one_chunk_per_layer(model_name="gpt2", out_root=out_root,
+                             layer_locs=("residual",), n_layers=12,
+                             device=device, chunk_size_gb=chunk_size_gb, **kw)

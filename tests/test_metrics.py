@@ -147,3 +147,69 @@ def test_ica_identifiability():
     o1 = np.argsort(abs(ica1.ica.components_[:, 0]))
     o2 = np.argsort(abs(ica2.ica.components_[:, 0]))
     assert np.allclose(abs(ica1.ica.components_[o1]), abs(ica2.ica.components_[o2]), atol=1e-3)
+
+
+def test_run_mmcs_with_larger():
+    """Hungarian-matched MMCS grid vs next-larger dict (reference :811-842):
+    a larger dict that CONTAINS the smaller one matches at ~1.0."""
+    from sparse_coding_amd.metrics.standard_metrics import run_mmcs_with_larger
+
+    torch.manual_seed(0)
+    d = 16
+    small = torch.randn(8, d)
+    large = torch.cat([small * 2.0, torch.randn(8, d)])  # scaled copies inside
+    grid = [[small, large]]
+    av, above, hists = run_mmcs_with_larger(grid, threshold=0.9)
+    assert av.shape == (1, 2)
+    assert av[0, 0] > 0.99
+    assert above[0, 0] == 100.0
+    assert hists[0][0].shape == (8,)
+    # unrelated dicts match poorly
+    grid2 = [[torch.randn(8, d), torch.randn(64, d)]]
+    av2, above2, _ = run_mmcs_with_larger(grid2)
+    assert av2[0, 0] < 0.9
+
+
+def test_plot_capacity_scatter(tmp_path):
+    from sparse_coding_amd.metrics.standard_metrics import plot_capacity_scatter
+    from sparse_coding_amd.models.learned_dict import TiedSAE
+
+    torch.manual_seed(1)
+    dicts = [(TiedSAE(torch.randn(16, 8), torch.zeros(16)), {"l1_alpha": 1e-3})
+             for _ in range(2)]
+    base = str(tmp_path / "cap")
+    plot_capacity_scatter(dicts, save_name=base)
+    import os
+
+    assert os.path.exists(base + "_0.png")
+    assert os.path.exists(base + "_1.png")
+    assert os.path.exists(base + "_hist.png")
+
+
+def test_cluster_vectors_export(tmp_path):
+    from sparse_coding_amd.metrics.standard_metrics import cluster_vectors
+    from sparse_coding_amd.models.learned_dict import TiedSAE
+
+    torch.manual_seed(2)
+    ld = TiedSAE(torch.randn(40, 8), torch.zeros(40))
+    loc = str(tmp_path / "top_clusters.txt")
+    top = cluster_vectors(ld, n_clusters=5, top_clusters=3, save_loc=loc, perplexity=5.0)
+    assert len(top) == 3
+    lines = open(loc).read().strip().split("\n")
+    assert len(lines) == 3
+    # clusters are disjoint id lists covering <= n_feats
+    ids = [int(x) for line in lines for x in line.strip("[]").split(",") if x.strip()]
+    assert len(ids) == len(set(ids))
+
+
+def test_make_one_chunk_per_layer(tmp_path):
+    """Layer-chunk helper writes the l{N}_{loc} layout (reference :582-601)."""
+    from sparse_coding_amd.metrics.standard_metrics import make_one_chunk_per_layer
+
+    make_one_chunk_per_layer(model_name="tiny-gptneox", out_root=str(tmp_path),
+                             layer_locs=("residual",), n_layers=2, device="cpu",
+                             chunk_size_gb=0.0002, max_length=16, model_batch_size=2)
+    import os
+
+    for layer in range(2):
+        assert os.path.exists(tmp_path / f"l{layer}_residual" / "0.pt")
