@@ -210,7 +210,8 @@ def test_example_notebooks_execute(tmp_path, monkeypatch):
     monkeypatch.chdir(tmp_path)
     root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
     for name in ("dict_across_time", "dict_compare", "interpreting_sparse_dictionaries",
-                 "case_studies_loop", "inter_layer_comparison"):
+                 "case_studies_loop", "inter_layer_comparison",
+                 "inter_dict_connections", "feature_interp"):
         nb = json.load(open(os.path.join(root, "examples", f"{name}.ipynb")))
         src = "\n".join("".join(c["source"]) for c in nb["cells"] if c["cell_type"] == "code")
         exec(compile(src, name, "exec"), {})  # noqa: S102 - our own notebooks
