@@ -88,6 +88,109 @@ def sharded_sweep(init_func_for_rank: Callable, cfg):
     return learned_dicts
 
 
+def lm_chunk_generator(cfg, device):
+    """Infinite stream of [chunk_activations, d] fp32 activation chunks from
+    the host LM, entirely in the generator rank's HBM (no disk round-trip).
+    The LM runs bf16 on GPU (ROADMAP item 7: ~2x generation throughput;
+    activations are cast fp32 for training as the reference trains fp32)."""
+    from sparse_coding_amd.data.activation_dataset import (
+        capture_activation_hook,
+        load_model,
+        synthetic_token_batches,
+    )
+
+    dtype = torch.bfloat16 if str(device).startswith("cuda") else None
+    model = load_model(cfg.model_name, device=device, dtype=dtype)
+    vocab = model.config.vocab_size
+    bsz = getattr(cfg, "model_batch_size", 4)
+    max_len = getattr(cfg, "max_length", 256)
+    per_batch = bsz * max_len
+    chunk_acts = getattr(cfg, "chunk_activations", 1 << 18)
+    batches_per_chunk = max(1, (chunk_acts + per_batch - 1) // per_batch)
+    layer, layer_loc = cfg.layer, getattr(cfg, "layer_loc", "residual")
+
+    while True:
+        parts = []
+        for toks in synthetic_token_batches(vocab, bsz, max_len, batches_per_chunk):
+            store = []
+            with torch.no_grad(), capture_activation_hook(model, layer, layer_loc, store):
+                model(input_ids=toks.to(device))
+            parts.append(store[0].float())
+        yield torch.cat(parts)[:chunk_acts].contiguous()
+
+
+def generator_trainer_sweep(init_func_for_rank: Callable, cfg,
+                            make_generator: Callable = None):
+    """Config-4/5 data plane: RANK 0 IS A DEDICATED
+    GENERATOR — it runs the host LM and streams each activation chunk
+    straight from its HBM to every trainer rank over one RCCL broadcast
+    (xGMI p2p; the chunk never touches disk or host RAM) — while ranks
+    >= 1 train their own ensembles on the previous chunk.  The pipeline
+    overlap is the natural one: rank 0 generates chunk i+1 while the
+    trainers are still stepping through chunk i; the broadcast is the only
+    sync point.
+
+    make_generator(cfg, device) -> iterator of [N, d] fp32 chunks
+    (default: lm_chunk_generator — the host-LM stream).
+    Returns the gathered learned_dicts on rank 0, None elsewhere.
+    """
+    from torch.utils.data import BatchSampler, RandomSampler
+
+    from sparse_coding_amd.parallel.chunk_feed import BroadcastChunkFeeder
+
+    rank, local_rank, world = init_distributed()
+    if world < 2:
+        raise RuntimeError("generator_trainer_sweep needs world_size >= 2 "
+                           "(one generator rank + trainers)")
+    device = f"cuda:{local_rank}" if torch.cuda.is_available() else "cpu"
+    cfg.device = device
+    torch.manual_seed(0)
+    np.random.seed(0)
+    os.makedirs(cfg.output_folder, exist_ok=True)
+
+    feeder = BroadcastChunkFeeder(device, src=0)
+    logger = RunLogger(cfg.output_folder, name="generator_trainer_sweep",
+                       use_wandb=getattr(cfg, "use_wandb", False)) if rank == 0 else None
+
+    ensemble, args, name = (None, None, None)
+    if rank == 0:
+        gen = (make_generator or lm_chunk_generator)(cfg, device)
+    else:
+        ensemble, args, name = init_func_for_rank(cfg, rank, world)
+
+    class _Counter:  # mp.Value-compatible progress stub
+        value = 0
+
+    n_steps = int(getattr(cfg, "n_chunks", 4)) * int(getattr(cfg, "n_repetitions", 1) or 1)
+    learned_dicts = None
+    for i in range(n_steps):
+        if rank == 0:
+            chunk = next(gen)
+            feeder.feed(chunk)
+            if logger is not None:
+                logger.log({"chunk": i, "chunk_rows": int(chunk.shape[0])})
+        else:
+            local = feeder.feed(None)
+            sampler = BatchSampler(RandomSampler(range(local.shape[0])),
+                                   batch_size=args.get("batch_size", 256), drop_last=False)
+            ensemble_train_loop(ensemble, cfg, args, name, sampler, local, _Counter())
+
+        if i == n_steps - 1 or (i + 1) in [2**j for j in range(3, 10)]:
+            local_dicts = [] if rank == 0 else unstacked_to_learned_dicts(
+                ensemble, args, cfg.ensemble_hyperparams, cfg.buffer_hyperparams)
+            gathered = [None] * world if rank == 0 else None
+            dist.gather_object(local_dicts, gathered, dst=0)
+            if rank == 0:
+                learned_dicts = [ld for part in gathered for ld in part]
+                iter_folder = os.path.join(cfg.output_folder, f"_{i}")
+                os.makedirs(iter_folder, exist_ok=True)
+                torch.save(learned_dicts, os.path.join(iter_folder, "learned_dicts.pt"))
+
+    if logger is not None:
+        logger.close()
+    return learned_dicts
+
+
 def _demo_init_for_rank(cfg, rank: int, world: int):
     """Default grid slice: rank r trains an 8-way L1 ensemble at dict ratio
     2^r (so 8 ranks cover ratios 1..128)."""

@@ -465,3 +465,79 @@ def test_dp_resample_keeps_replicas_identical():
     assert (c0 == c1).all() and c0.sum() > 0
     for k in p0:
         assert (p0[k] == p1[k]).all(), k
+
+
+# ---------------------------------------------------------------------------
+# generator-rank -> trainer-rank chunk streaming
+# ---------------------------------------------------------------------------
+
+def _gen_trainer_worker(rank, world_size, port, tmpdir, out_q):
+    try:
+        os.environ["MASTER_ADDR"] = "127.0.0.1"
+        os.environ["MASTER_PORT"] = str(port)
+        os.environ["WORLD_SIZE"] = str(world_size)
+        os.environ["RANK"] = str(rank)
+        os.environ["LOCAL_RANK"] = str(rank)
+        dist.init_process_group("gloo", rank=rank, world_size=world_size)
+
+        from sparse_coding_amd.config import EnsembleArgs
+        from sparse_coding_amd.sweep.sharded_sweep import generator_trainer_sweep
+
+        cfg = EnsembleArgs()
+        cfg.model_name = "tiny-gptneox"
+        cfg.layer = 1
+        cfg.layer_loc = "residual"
+        cfg.model_batch_size = 2
+        cfg.max_length = 16
+        cfg.chunk_activations = 256
+        cfg.n_chunks = 2
+        cfg.n_repetitions = 1
+        cfg.batch_size = 64
+        cfg.lr = 1e-3
+        cfg.output_folder = os.path.join(tmpdir, "out")
+        cfg.use_wandb = False
+        cfg.ensemble_hyperparams = ["dict_size"]
+        cfg.buffer_hyperparams = ["l1_alpha"]
+
+        def init_for_rank(c, r, w):
+            from sparse_coding_amd.engine.ensemble import FunctionalEnsemble
+            from sparse_coding_amd.functional.optim import adam
+            from sparse_coding_amd.models.sae_signatures import FunctionalTiedSAE
+
+            torch.manual_seed(100 + r)
+            d = 64  # tiny-gptneox hidden
+            models = [FunctionalTiedSAE.init(d, 2 * d, l1) for l1 in (1e-4, 1e-3)]
+            ens = FunctionalEnsemble(models, FunctionalTiedSAE, adam, {"lr": 1e-3},
+                                     backend="torch")
+            return ens, {"batch_size": c.batch_size, "device": "cpu", "dict_size": 2 * d}, f"r{r}"
+
+        dicts = generator_trainer_sweep(init_for_rank, cfg)
+        if rank == 0:
+            out_q.put(("ok", len(dicts), cfg.output_folder))
+        dist.destroy_process_group()
+    except Exception:  # noqa: BLE001
+        import traceback
+
+        out_q.put(("error", f"rank {rank}: {traceback.format_exc()}", None))
+        raise
+
+
+@pytest.mark.timeout(300)
+def test_generator_trainer_sweep_gloo(tmp_path):
+    """Rank 0 generates host-LM activation chunks, rank 1 trains; checkpoints
+    gather to rank 0 in the reference layout."""
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    procs = [ctx.Process(target=_gen_trainer_worker, args=(r, 2, 29581, str(tmp_path), q))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    status, n_dicts, out_folder = q.get(timeout=250)
+    assert status == "ok", n_dicts
+    for p in procs:
+        p.join(timeout=60)
+        assert p.exitcode == 0
+    assert n_dicts == 2  # rank 1's two-l1 ensemble
+    final = torch.load(os.path.join(out_folder, "_1", "learned_dicts.pt"), weights_only=False)
+    assert len(final) == 2
+    assert type(final[0][0]).__module__ == "autoencoders.learned_dict"
