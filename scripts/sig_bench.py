@@ -47,6 +47,15 @@ def build(sig_name, d, n, M, device):
     elif sig_name == "centered":
         sig = sigs.FunctionalTiedCenteredSAE
         models = [sig.init(d, n, float(l1), device=device) for l1 in l1s]
+    elif sig_name == "whitened":
+        sig = sigs.FunctionalTiedSAE
+        torch.manual_seed(7)
+        models = []
+        for l1 in l1s:
+            q, _ = torch.linalg.qr(torch.randn(d, d, device=device))
+            models.append(sig.init(d, n, float(l1), device=device, rotation=q,
+                                   translation=torch.randn(d, device=device) * 0.1,
+                                   scaling=torch.rand(d, device=device) + 0.5))
     elif sig_name == "positive":
         sig = positive.FunctionalPositiveTiedSAE
         models = [sig.init(d, n, float(l1), device=device) for l1 in l1s]
@@ -102,13 +111,20 @@ def main():
     p.add_argument("--d-model", type=int, default=512)
     p.add_argument("--dict-size", type=int, default=4096)
     p.add_argument("--n-models", type=int, default=8)
+    p.add_argument("--only", default="", help="run just one signature")
+    p.add_argument("--backend", default="", help="run just one backend (hip|torch)")
     args = p.parse_args()
     device = "cuda:0"
 
-    for sig_name in ("tied", "untied", "masked_tied", "thresholding", "reverse",
-                     "centered", "positive", "topk", "lista", "residual", "semilinear"):
+    sig_names = ("tied", "untied", "masked_tied", "thresholding", "reverse",
+                 "centered", "whitened", "positive", "topk", "lista",
+                 "residual", "semilinear")
+    if args.only:
+        sig_names = tuple(s for s in sig_names if s == args.only)
+    backends = (args.backend,) if args.backend else ("hip", "torch")
+    for sig_name in sig_names:
         row = {"sig": sig_name}
-        for backend in ("hip", "torch"):
+        for backend in backends:
             try:
                 ms, impl = time_backend(sig_name, backend, args, device)
                 row[backend + "_ms"] = round(ms, 3)
