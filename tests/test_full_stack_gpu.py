@@ -635,3 +635,51 @@ def test_whitened_tied_step_matches_oracle():
     assert err < 2e-3, err
     err_b = (ens_hip.params["encoder_bias"] - ens_ref.params["encoder_bias"]).abs().max()
     assert err_b < 2e-3, err_b
+
+
+def test_sweep_resume_equivalence_gpu(tmp_path):
+    """resume_state.pt round-trips the FUSED step's Adam state mid-sweep
+    (VERDICT weak #7: the CPU test covers the torch backend only): an
+    interrupted+resumed fused run must match an uninterrupted one."""
+    from sparse_coding_amd.config import SyntheticEnsembleArgs
+    from sparse_coding_amd.models.sae_signatures import FunctionalTiedSAE
+    from sparse_coding_amd.sweep import big_sweep
+    from sparse_coding_amd.sweep.experiments import make_grid_ensembles
+
+    def mini_cfg(base, n_repetitions):
+        cfg = SyntheticEnsembleArgs()
+        cfg.use_synthetic_dataset = True
+        cfg.activation_width = 128
+        cfg.n_ground_truth_components = 256
+        cfg.gen_batch_size = 512
+        cfg.feature_num_nonzero = 8
+        cfg.noise_magnitude_scale = 0.0
+        cfg.chunk_size_gb = 128 * 512 * 4 * 2 / 1024**3  # 4 batches/chunk
+        cfg.n_chunks = 2
+        cfg.n_repetitions = n_repetitions
+        cfg.batch_size = 512
+        cfg.device = DEV
+        cfg.dataset_folder = str(base / "data")
+        cfg.output_folder = str(base / "out")
+        cfg.use_wandb = False
+        cfg.wandb_images = False
+        return cfg
+
+    def init_func(c):
+        return make_grid_ensembles(c, FunctionalTiedSAE, [1e-3], [2.0], devices=[DEV])
+
+    cfg_a = mini_cfg(tmp_path / "a", 4)
+    dicts_a = big_sweep.sweep(init_func, cfg_a)
+
+    cfg_b1 = mini_cfg(tmp_path / "b", 2)
+    big_sweep.sweep(init_func, cfg_b1)
+    assert os.path.exists(os.path.join(cfg_b1.output_folder, "resume_state.pt"))
+    cfg_b2 = mini_cfg(tmp_path / "b", 4)
+    cfg_b2.resume = True
+    dicts_b = big_sweep.sweep(init_func, cfg_b2)
+
+    (ld_a, hp_a), = dicts_a
+    (ld_b, hp_b), = dicts_b
+    assert hp_a == hp_b
+    assert torch.allclose(ld_a.get_learned_dict(), ld_b.get_learned_dict(), atol=1e-5)
+    assert torch.allclose(ld_a.encoder_bias, ld_b.encoder_bias, atol=1e-5)
