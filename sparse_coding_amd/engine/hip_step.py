@@ -414,7 +414,6 @@ class HipCenteredStep(HipSAEStep):
         dev = self.ens.params["encoder"].device
         self.xc = torch.empty(M, B, d, device=dev)
         self.g_center = torch.empty(M, d, device=dev)
-        self.ones_row = torch.ones(M, 1, B, device=dev)
         self.zero_decay = torch.zeros_like(self.bias_decay)
         self.kc["staging"] = "t"
 
@@ -447,9 +446,8 @@ class HipCenteredStep(HipSAEStep):
         ext.grad_w(self.c, self.r, self.gw, gscale, 0.0, bk_gw, prio)
         ext.grad_w(self.gpre, self.xc, self.gw, 1.0, 1.0, bk_gw, prio)
         # center gradient from existing reductions (docstring derivation);
-        # column sum as a skinny batched GEMM (full-HBM-rate stream)
-        torch.bmm(self.ones_row, self.r, out=self.g_center.view(-1, 1, self.d_act))
-        self.g_center.mul_(gscale)
+        # column sum via the coalesced k_colsum kernel
+        ext.colsum(self.r, self.g_center, gscale)
         self.g_center.sub_(torch.einsum("mn,mnd->md", self.g_bias * self.inv_norms, enc))
         if on_grads is not None:
             on_grads([self.gw, self.g_bias, self.g_center])
@@ -973,7 +971,6 @@ class HipResidualDenoisingStep:
         self.rr = f(M, B, d)
         self.g_x = f(M, B, n)
         self.g_h = f(M, B, n)
-        self.ones_row = torch.ones(M, 1, B, device=dev)
         self.gA = f(M, n, d)
         self.gW = [f(M, n, n) for _ in range(L)]
         self.g_theta = [f(M, n) for _ in range(L)]
@@ -1034,11 +1031,11 @@ class HipResidualDenoisingStep:
                         self.scratch_lp, bk_dec, prio)
             # relu(x + theta) backward
             self.g_h.mul_((self.hs[l] > 0).float())
-            # column sum as a ones-vector GEMM: torch.sum(dim=1) dispatches a
-            # strided reduce at ~140 GB/s (rocprof r02_residual_kernel_stats:
-            # 1.79 ms = 5% of the step); the skinny batched GEMM streams the
-            # same bytes at full HBM rate
-            torch.bmm(self.ones_row, self.g_h, out=self.g_theta[l].view(-1, 1, self.n_dict))
+            # column sum via k_colsum: torch.sum(dim=1) dispatches a strided
+            # reduce at ~140 GB/s (rocprof r02_residual_kernel_stats: 1.79 ms
+            # = 5% of the step); the coalesced-row kernel streams the same
+            # bytes at HBM rate
+            ext.colsum(self.g_h, self.g_theta[l])
             if on_grads is not None:
                 on_grads([self.g_theta[l]])
             self.g_x.add_(self.g_h)  # residual skip + through-layer paths
@@ -1133,7 +1130,6 @@ class HipSemilinearStep:
         self.rr = f(M, B, d)
         self.g_c = f(M, B, n)
         self.g_h = f(M, B, h)
-        self.ones_row = torch.ones(M, 1, B, device=dev)
         self.gA = f(M, n, d)
         self.gW2 = f(M, n, h)
         self.gW1 = f(M, h, d)
@@ -1177,9 +1173,9 @@ class HipSemilinearStep:
         ext.dec_fwd(self.g_c, l2p["weight"], self.ones_mh, self.zeros_bh,
                     self.g_h, self.scratch_lp, bk_dec, prio)
         self.g_h.mul_((self.h1 > 0).float())
-        # strided-reduce -> ones-vector GEMM (9% of the semilinear step,
+        # strided-reduce -> k_colsum (9% of the semilinear step,
         # rocprof r02_semilinear_kernel_stats)
-        torch.bmm(self.ones_row, self.g_h, out=self.g_b1.view(-1, 1, self.hidden))
+        ext.colsum(self.g_h, self.g_b1)
         ext.grad_w(self.g_h, b, self.gW1, 1.0, 0.0, bk_gw, prio)
         if on_grads is not None:
             on_grads([self.gA, self.gW2, self.gW1, self.g_b2, self.g_b1])

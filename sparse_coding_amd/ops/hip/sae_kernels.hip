@@ -1163,6 +1163,31 @@ void k_resample(const float* __restrict__ fired,      // [M, n]
 }
 
 // ---------------------------------------------------------------------------
+// k_colsum: out[m, j] = alpha * sum_b X[m, b, j] for X [M, B, n].
+// The column-sum tail of the residual/semilinear backward: torch's strided
+// reduce runs this at ~140 GB/s (profiles/r02_*_kernel_stats.csv); here each
+// 256-thread block owns 256 consecutive columns of one (model, B-slice) so
+// every row read is one fully-coalesced 1 KB line burst, and the B axis is
+// split over grid.z (partials combined with atomicAdd; caller zero-fills
+// out) to keep >= 4k waves in flight on the 256-CU chip.
+#define COLSUM_T 256
+extern "C" __global__ __launch_bounds__(COLSUM_T)
+void k_colsum(const float* __restrict__ X, float* __restrict__ out,
+              int B, int n, float alpha) {
+  const int j = blockIdx.x * COLSUM_T + threadIdx.x;
+  const int m = blockIdx.y;
+  const int nsplit = gridDim.z;
+  if (j >= n) return;
+  const long base = (long)m * B * n;
+  const int b0 = (int)(((long)blockIdx.z * B) / nsplit);
+  const int b1 = (int)(((long)(blockIdx.z + 1) * B) / nsplit);
+  float acc = 0.f;
+  #pragma unroll 4
+  for (int b = b0; b < b1; ++b) acc += X[base + (long)b * n + j];
+  atomicAdd(&out[(long)m * n + j], acc * alpha);
+}
+
+// ---------------------------------------------------------------------------
 // k_transpose_scale: dst[c][r] = src[r][c] * (scale ? scale[r] : 1), batched
 // over grid.z with explicit strides.  64x64 LDS tiles, coalesced both sides.
 // Feeds the all-direct-staged GEMM variants below (x^T, r^T, What^T).
