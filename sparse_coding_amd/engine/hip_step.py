@@ -775,6 +775,7 @@ class HipLISTAStep:
         self.gW = [f(M, n, d) for _ in range(L)]
         self.g_theta = [f(M, n) for _ in range(L)]
         self.g_rho = [f(M) for _ in range(L)]
+        self.scratch_mn = f(M, n)
         self.zero_decay = torch.zeros(M, device=dev)
         self._B = B
         self._graph = None
@@ -891,7 +892,8 @@ class HipLISTAStep:
 
     def _loss_data(self, B: int):
         mse = self.loss_parts[:, 0] / (B * self.d_act)
-        l1 = self.l1_alpha * self.y[self.n_layers].abs().sum(dim=(1, 2)) / B
+        self.ext.colsum(self.y[self.n_layers], self.scratch_mn, absval=True)
+        l1 = self.l1_alpha * self.scratch_mn.sum(dim=1) / B
         return {"loss": mse + l1, "l_reconstruction": mse, "l_l1": l1}
 
     def _capture(self, x: torch.Tensor) -> None:
@@ -975,6 +977,7 @@ class HipResidualDenoisingStep:
         self.gW = [f(M, n, n) for _ in range(L)]
         self.g_theta = [f(M, n) for _ in range(L)]
         self.g_bias = f(M, n)
+        self.scratch_mn = f(M, n)
         self.zero_decay = torch.zeros(M, device=dev)
         self._B = B
 
@@ -1012,7 +1015,11 @@ class HipResidualDenoisingStep:
         self.c.clamp_(min=0.0)
 
         ext.dec_fwd(self.c, A, self.inv_norms, b, self.rr, self.loss_parts, bk_dec, prio)
-        self._l1_sum = self.c.sum(dim=(1, 2))  # c >= 0
+        # c >= 0, so the L1 sum is a column sum + tiny [M, n] reduce; the
+        # torch sum(dim=(1,2)) dispatched a 1.79 ms strided reduce
+        # (profiles/r02_residual_kernel_stats.csv)
+        ext.colsum(self.c, self.scratch_mn)
+        self._l1_sum = self.scratch_mn.sum(dim=1)
 
         # backward: g_c via k_gc (relu mask + l1 term + bias colsum, K=d)
         ext.gc(self.rr, A, self.inv_norms, self.c, self.l1_alpha,
@@ -1135,6 +1142,7 @@ class HipSemilinearStep:
         self.gW1 = f(M, h, d)
         self.g_b2 = f(M, n)
         self.g_b1 = f(M, h)
+        self.scratch_mn = f(M, n)
         self.zero_decay = torch.zeros(M, device=dev)
         self._B = B
 
@@ -1161,7 +1169,10 @@ class HipSemilinearStep:
         ext.enc_fwd(self.h1, l2p["weight"], l2p["bias"], None, self.c,
                     self.scratch_lp, self.fired, 0, bk, prio)
         ext.dec_fwd(self.c, A, self.inv_norms, b, self.rr, self.loss_parts, bk_dec, prio)
-        self._l1_sum = self.c.sum(dim=(1, 2))
+        # c >= 0: L1 via the coalesced column-sum kernel (the torch
+        # sum(dim=(1,2)) was 9%% of this step)
+        ext.colsum(self.c, self.scratch_mn)
+        self._l1_sum = self.scratch_mn.sum(dim=1)
 
         # backward
         ext.gc(self.rr, A, self.inv_norms, self.c, self.l1_alpha,

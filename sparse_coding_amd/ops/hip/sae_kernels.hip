@@ -1173,7 +1173,7 @@ void k_resample(const float* __restrict__ fired,      // [M, n]
 #define COLSUM_T 256
 extern "C" __global__ __launch_bounds__(COLSUM_T)
 void k_colsum(const float* __restrict__ X, float* __restrict__ out,
-              int B, int n, float alpha) {
+              int B, int n, float alpha, int absval) {
   const int j = blockIdx.x * COLSUM_T + threadIdx.x;
   const int m = blockIdx.y;
   const int nsplit = gridDim.z;
@@ -1182,8 +1182,13 @@ void k_colsum(const float* __restrict__ X, float* __restrict__ out,
   const int b0 = (int)(((long)blockIdx.z * B) / nsplit);
   const int b1 = (int)(((long)(blockIdx.z + 1) * B) / nsplit);
   float acc = 0.f;
-  #pragma unroll 4
-  for (int b = b0; b < b1; ++b) acc += X[base + (long)b * n + j];
+  if (absval) {
+    #pragma unroll 4
+    for (int b = b0; b < b1; ++b) acc += fabsf(X[base + (long)b * n + j]);
+  } else {
+    #pragma unroll 4
+    for (int b = b0; b < b1; ++b) acc += X[base + (long)b * n + j];
+  }
   atomicAdd(&out[(long)m * n + j], acc * alpha);
 }
 

@@ -289,7 +289,7 @@ void bias_adam(torch::Tensor bias, torch::Tensor g_bias, torch::Tensor decay,
                      (float)lr, (float)b1, (float)b2, (float)eps_adam, lrm);
 }
 
-void colsum(torch::Tensor X, torch::Tensor out, double alpha) {
+void colsum(torch::Tensor X, torch::Tensor out, double alpha, bool absval) {
   CHECK_IN(X); CHECK_IN(out);
   TORCH_CHECK(X.dim() == 3, "colsum wants [M, B, n]");
   int M = X.size(0), B = X.size(1), n = X.size(2);
@@ -300,7 +300,7 @@ void colsum(torch::Tensor X, torch::Tensor out, double alpha) {
   dim3 grid(cdiv(n, 256), M, nsplit);
   hipLaunchKernelGGL(k_colsum, grid, dim3(256), 0, cur_stream(),
                      X.data_ptr<float>(), out.data_ptr<float>(), B, n,
-                     (float)alpha);
+                     (float)alpha, absval ? 1 : 0);
 }
 
 void transpose_scale(torch::Tensor src, torch::Tensor dst,
@@ -504,7 +504,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("w_used") = py::none(), py::arg("clamp_mask") = false,
         py::arg("lr_mult") = py::none());
   m.def("colsum", &colsum, "out[m,j] = alpha * sum_b X[m,b,j] (coalesced column sum)",
-        py::arg("X"), py::arg("out"), py::arg("alpha") = 1.0);
+        py::arg("X"), py::arg("out"), py::arg("alpha") = 1.0,
+        py::arg("absval") = false);
   m.def("bias_adam", &bias_adam, "Adam on bias with L2-norm decay",
         py::arg("bias"), py::arg("g_bias"), py::arg("decay"), py::arg("mu"),
         py::arg("nu"), py::arg("step_no"), py::arg("lr"), py::arg("b1"),
