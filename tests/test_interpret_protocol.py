@@ -198,3 +198,20 @@ def test_interpret_across_chunks(tmp_path):
     out = interpret_across_chunks(str(base), str(tmp_path / "res"), cfg, 8.5e-4,
                                   chunks=(1, 4), n_fragments=8)
     assert len(out) == 2
+
+
+@pytest.mark.timeout(300)
+def test_interpret_across_baselines(tmp_path):
+    """Baseline-folder driver over the reference l{N}_{loc} layout."""
+    from sparse_coding_amd.interpret.drivers import interpret_across_baselines
+
+    base = tmp_path / "baselines"
+    for layer in (0, 1):
+        d = base / f"l{layer}_residual"
+        os.makedirs(d, exist_ok=True)
+        torch.save(_tiny_learned_dict(), d / "pca.pt")
+        torch.save(_tiny_learned_dict(), d / "nmf.pt")  # must be skipped (reference :563)
+    cfg = _tiny_cfg(tmp_path)
+    out = interpret_across_baselines(str(base), str(tmp_path / "res"), cfg, n_fragments=8)
+    assert set(out) == {"l0_residual/pca.pt", "l1_residual/pca.pt"}
+    assert os.path.isdir(tmp_path / "res" / "l0_residual" / "pca")
