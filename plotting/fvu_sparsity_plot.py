@@ -90,19 +90,24 @@ def plot_fvu_sparsity(
     return fig
 
 
-def main():
+def main(argv=None, default_out="fvu_sparsity.png", center_default=False):
     p = argparse.ArgumentParser()
     p.add_argument("--learned-dicts", required=True)
     p.add_argument("--chunk", required=True, help="held-out activation chunk .pt")
     p.add_argument("--device", default="cuda:0" if torch.cuda.is_available() else "cpu")
     p.add_argument("--n-samples", type=int, default=20000)
-    p.add_argument("--out", default="fvu_sparsity.png")
+    p.add_argument("--out", default=default_out)
     p.add_argument("--baseline-dir", default="", help="folder of saved baseline dicts")
-    args = p.parse_args()
+    p.add_argument("--center", action="store_true", default=center_default,
+                   help="subtract the chunk mean before scoring (the mean-centered "
+                        "MLP variant, reference fvu_sparsity_plot_mlp_center.py)")
+    args = p.parse_args(argv)
 
     chunk = torch.load(args.chunk, map_location="cpu").float()
     idx = np.random.choice(len(chunk), size=min(args.n_samples, len(chunk)), replace=False)
     sample = chunk[idx]
+    if args.center:
+        sample = sample - chunk.mean(dim=0)
 
     curves = score_learned_dicts(args.learned_dicts, sample, device=args.device)
 

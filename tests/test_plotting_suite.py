@@ -236,3 +236,17 @@ def test_leace_multiclass():
     before = torch.stack([acts[labels == c].mean(dim=0) for c in range(3)])
     spread_before = (before - before.mean(dim=0)).norm(dim=1).max()
     assert spread < 0.25 * spread_before
+
+
+def test_fvu_sparsity_named_variants(tmp_path, chunk_file):
+    import fvu_sparsity_plot_gpt2sm as g
+    import fvu_sparsity_plot_mlp_center as c
+
+    _sweep_layout(tmp_path, ["sweep"], epochs=(9,))
+    ld_path = f"{tmp_path}/sweep/_9/learned_dicts.pt"
+    out1, out2 = str(tmp_path / "g.png"), str(tmp_path / "c.png")
+    g.main(["--learned-dicts", ld_path, "--chunk", chunk_file,
+            "--device", "cpu", "--out", out1])
+    c.main(["--learned-dicts", ld_path, "--chunk", chunk_file,
+            "--device", "cpu", "--out", out2])
+    assert os.path.exists(out1) and os.path.exists(out2)
