@@ -27,7 +27,7 @@ _sys.path.insert(0, _os.path.dirname(_os.path.dirname(_os.path.abspath(__file__)
 
 import os
 
-import numpy as np
+
 import torch
 
 

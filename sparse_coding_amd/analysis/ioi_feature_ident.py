@@ -17,7 +17,7 @@ semantics (data/ioi_counterfact.py) and the ablation machinery:
 
 from __future__ import annotations
 
-from typing import Dict, List, Optional, Tuple
+from typing import Dict, List, Optional
 
 import torch
 

@@ -23,7 +23,7 @@ from __future__ import annotations
 import copy
 import os
 from datetime import datetime
-from typing import Callable, Dict, List, Optional, Tuple
+from typing import Dict, List, Tuple
 
 import torch
 
