@@ -716,3 +716,46 @@ def test_sweep_resume_equivalence_gpu(tmp_path):
     assert err < 1e-5, err
     err_m = (ens_a.optim_states["nu"]["encoder"] - ens_b.optim_states["nu"]["encoder"]).abs().max().item()
     assert err_m < 1e-6, err_m
+
+
+def test_colsum_matches_torch():
+    """k_colsum vs torch.sum(dim=1), plain and absval, edge n not a
+    multiple of the 256-column block."""
+    from sparse_coding_amd import ops
+
+    ext = ops.get_extension(required=True)
+    torch.manual_seed(91)
+    for M, B, n in ((3, 517, 384), (8, 2048, 4096), (2, 64, 100)):
+        x = torch.randn(M, B, n, device=DEV)
+        out = torch.empty(M, n, device=DEV)
+        ext.colsum(x, out)
+        ref = x.sum(dim=1)
+        assert torch.allclose(out, ref, atol=1e-3 * B ** 0.5), (M, B, n)
+        ext.colsum(x, out, 2.5, True)
+        ref = 2.5 * x.abs().sum(dim=1)
+        assert torch.allclose(out, ref, rtol=1e-5, atol=1e-3 * B ** 0.5), (M, B, n)
+
+
+def test_rs_ag_respects_lr_mult():
+    """The sharded Adam must read the same per-row lr_mult slices the full
+    update does: frozen rows stay frozen under dp_mode=rs_ag."""
+    from sparse_coding_amd.engine.ensemble import FunctionalEnsemble
+    from sparse_coding_amd.functional.optim import adam
+    from sparse_coding_amd.models.sae_signatures import FunctionalTiedSAE
+    from sparse_coding_amd.parallel.dp import DataParallelEnsembleTrainer
+
+    torch.manual_seed(92)
+    M, B, d, n = 2, 256, 64, 256
+    models = [FunctionalTiedSAE.init(d, n, 1e-3, device=DEV) for _ in range(M)]
+    ens = FunctionalEnsemble(models, FunctionalTiedSAE, adam, {"lr": 1e-3}, device=DEV, backend="hip")
+    trainer = DataParallelEnsembleTrainer(ens, force_dp_path=True, dp_mode="rs_ag")
+    hs = ens._hip_step
+    hs.lr_mult[:, : n // 2] = 0.0
+    before = ens.params["encoder"][:, : n // 2].clone()
+    x = torch.randn(B, d, device=DEV)
+    for _ in range(3):
+        trainer.step(x)
+    torch.cuda.synchronize()
+    assert torch.equal(ens.params["encoder"][:, : n // 2], before)
+    assert not torch.equal(ens.params["encoder"][:, n // 2 :],
+                           torch.zeros_like(before))  # others moved
