@@ -42,7 +42,10 @@ def parse_args():
     p.add_argument("--n-models", type=int, default=8, help="L1-ensemble size")
     p.add_argument("--tied", action="store_true", default=True)
     p.add_argument("--untied", dest="tied", action="store_false")
-    p.add_argument("--sig", choices=["sae", "topk", "thresholding"], default="sae",
+    # --family is the torchrun-safe alias: torchrun's own parser abbrev-
+    # matches "--sig" to --signals-to-handle and steals it from script args
+    p.add_argument("--sig", "--family", dest="sig",
+                   choices=["sae", "topk", "thresholding"], default="sae",
                    help="model family: sae (tied/untied per --tied; the BASELINE default), "
                         "topk (k=32 + dead-neuron resampling, config 5), thresholding")
     p.add_argument("--backend", choices=["auto", "hip", "torch"], default="auto")
