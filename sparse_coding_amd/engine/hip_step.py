@@ -261,6 +261,7 @@ class HipSAEStep:
         kc = self.kc
         bk, prio = kc["bk"], kc["prio"]
         bn = kc.get("bn", 128)
+        bn_enc = kc.get("bn_enc") or bn
         bk_dec = kc["bk_dec"] or bk
         bk_gw = kc["bk_grad_w"] or bk
 
@@ -284,11 +285,11 @@ class HipSAEStep:
             # transpose-in-staging GEMMs (no separate transpose kernels)
             enc_inv = self.inv_norms if self.tied else None
             mode = 3 if self.reverse else 0
-            ext.enc_fwd(x, enc, bias, enc_inv, self.c, self.loss_parts, self.fired, mode, bk, prio, bn,
+            ext.enc_fwd(x, enc, bias, enc_inv, self.c, self.loss_parts, self.fired, mode, bk, prio, bn_enc,
                         dict_sizes=self.dict_sizes)
             ext.dec_fwd(self.c, dict_w, self.inv_norms, x, self.r, self.loss_parts, bk_dec, prio, bn)
             ext.gc(self.r, dict_w, self.inv_norms, self.c, self.l1_alpha, self.gpre, self.g_bias, bk, prio,
-                   gc_mode=1 if self.reverse else 0, bn=bn)
+                   gc_mode=1 if self.reverse else 0, bn=bn_enc)
 
         if on_grads is not None:
             on_grads([self.g_bias])  # final after k_gc

@@ -42,6 +42,11 @@ DEFAULTS = {
     # acts/s on the flagship bench).
     "bk_grad_w": None,
     "bk_dec": 32,
+    # per-kernel output-tile width for the SHORT-K kernels (enc_fwd/gc,
+    # K = d): None -> "bn".  PMC r02: enc/gc sit at 0.52-0.53 MFMA-busy vs
+    # dec's 0.83; the 128x256 tile halves their column panels (fewer
+    # barriers per FLOP) at 2 blocks/CU — sweep via SC_AMD_BN_ENC.
+    "bn_enc": None,
 }
 
 _cfg = dict(DEFAULTS)
@@ -58,6 +63,8 @@ if os.environ.get("SC_AMD_BK_GRAD_W"):
     _cfg["bk_grad_w"] = int(os.environ["SC_AMD_BK_GRAD_W"])
 if os.environ.get("SC_AMD_BK_DEC"):
     _cfg["bk_dec"] = int(os.environ["SC_AMD_BK_DEC"])
+if os.environ.get("SC_AMD_BN_ENC"):
+    _cfg["bn_enc"] = int(os.environ["SC_AMD_BN_ENC"])
 
 
 def kernel_config() -> dict:
@@ -71,3 +78,4 @@ def set_kernel_config(**kwargs) -> None:
         _cfg[k] = v
     assert _cfg["bk"] in (16, 32) and _cfg["staging"] in ("pre", "t")
     assert _cfg["bn"] in (128, 256)
+    assert _cfg["bn_enc"] in (None, 128, 256)
