@@ -186,3 +186,30 @@ def test_sweep_image_metrics(tmp_path):
     names = os.listdir(img_dir)
     assert any("sparsity_hist" in n for n in names)
     assert any("mmcs_grid" in n for n in names)
+
+
+@pytest.mark.timeout(300)
+def test_big_sweep_experiments_cli(tmp_path):
+    """The reference-named big_sweep_experiments.py entry runs any catalogue
+    experiment as a subcommand (the reference's __main__ hard-codes one)."""
+    import subprocess
+    import sys
+
+    root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    out = subprocess.run([sys.executable, os.path.join(root, "big_sweep_experiments.py"), "list"],
+                         capture_output=True, text=True, cwd=root)
+    assert out.returncode == 0 and "run_dense_l1_range" in out.stdout
+
+    r = subprocess.run(
+        [sys.executable, os.path.join(root, "big_sweep_experiments.py"), "run_synthetic",
+         "--use_synthetic_dataset", "True",
+         "--activation_width", "32", "--n_ground_truth_components", "48",
+         "--gen_batch_size", "128", "--feature_num_nonzero", "4",
+         "--chunk_size_gb", str(32 * 128 * 4 * 2 / 1024**3), "--n_chunks", "2",
+         "--batch_size", "64", "--device", "cpu",
+         "--dataset_folder", str(tmp_path / "data"),
+         "--output_folder", str(tmp_path / "out"),
+         "--use_wandb", "False"],
+        capture_output=True, text=True, cwd=root, timeout=280)
+    assert r.returncode == 0, r.stderr[-800:]
+    assert os.path.exists(tmp_path / "out" / "_1" / "learned_dicts.pt")
