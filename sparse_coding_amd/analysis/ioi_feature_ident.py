@@ -51,7 +51,6 @@ def logit_diff(model, dataset, device: str = "cpu",
                edit_fn=None) -> float:
     """Mean (IO logit - S logit) at the end position, optionally with an
     activation edit patched in at one layer (the ablation eval hook)."""
-    import contextlib
 
     from sparse_coding_amd.data.activation_dataset import resolve_hook_point
 

@@ -10,7 +10,7 @@ config.py:15-21, which breaks embedding).
 from __future__ import annotations
 
 import argparse
-from dataclasses import dataclass, field, fields
+from dataclasses import dataclass, field
 from typing import Any, List, Optional
 
 import torch

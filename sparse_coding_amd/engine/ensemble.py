@@ -24,7 +24,7 @@ The writeback in ``step_batch`` copies new optimizer state into the existing
 
 from __future__ import annotations
 
-from typing import Any, List, Optional, Type
+from typing import List, Type
 
 import torch
 

@@ -13,7 +13,7 @@ Shared math lives in module-level helpers instead of being repeated per class.
 
 from __future__ import annotations
 
-from typing import Any, Dict, Optional, Tuple
+from typing import Dict
 
 import torch
 import torch.nn as nn

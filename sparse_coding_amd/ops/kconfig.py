@@ -22,7 +22,7 @@ combinations on a GPU box and the measured winner is baked into DEFAULTS.
 from __future__ import annotations
 
 import os
-from typing import Literal
+
 
 # Measured winners (MI355X, flagship bench shape; see profiles/README.md).
 DEFAULTS = {
