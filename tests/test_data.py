@@ -170,3 +170,22 @@ def test_get_activation_size_and_tensor_names():
     assert get_activation_size("pythia-1.4b", "residual") == 2048
     assert make_tensor_name(2, "residual", "pythia-70m") == "blocks.2.hook_resid_post"
     assert make_tensor_name(5, "mlpout", "gpt2") == "blocks.5.hook_mlp_out"
+
+
+@pytest.mark.timeout(300)
+def test_generate_test_data_cli(tmp_path):
+    """The reference-named generate_test_data.py CLI (C25) writes the chunk
+    layout end to end."""
+    import subprocess
+    import sys
+
+    root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    r = subprocess.run(
+        [sys.executable, os.path.join(root, "generate_test_data.py"),
+         "--model_name", "tiny-gptneox", "--layers", "1",
+         "--layer_loc", "residual", "--dataset_folder", str(tmp_path / "gtd"),
+         "--n_chunks", "1", "--chunk_size_gb", "0.0002", "--device", "cpu"],
+        capture_output=True, text=True, cwd=root, timeout=280)
+    assert r.returncode == 0, r.stderr[-500:]
+    chunk = torch.load(tmp_path / "gtd" / "0.pt", weights_only=False)
+    assert chunk.dtype == torch.float16 and chunk.shape[1] == 64
