@@ -88,6 +88,23 @@ def ensemble_train_loop(ensemble, cfg, args, ensemble_name, sampler, dataset, pr
 
     logger = getattr(cfg, "logger", None)
 
+    # opt-in dead-feature resampling inside the sweep (the reference wires
+    # resampling only into its DDP experiment; here any sweep can use the
+    # anthropic/worst protocols — engine/resample.py).  The dead window is
+    # one chunk (the reference's rule, huge_batch_size.py:230); the rewrite
+    # lands in the shared-memory params, so the parent sees it like any
+    # training update.
+    resample_every = int(getattr(cfg, "resample_every_chunks", 0) or 0)
+    resampler = None
+    if resample_every:
+        from sparse_coding_amd.engine.resample import EnsembleResampler
+
+        resampler = EnsembleResampler(
+            ensemble,
+            n_track=int(getattr(cfg, "resample_n_track", 512)),
+            protocol=getattr(cfg, "resample_protocol", "anthropic"),
+            warmup_steps=int(getattr(cfg, "resample_warmup_steps", 1000)))
+
     # stage the whole chunk into this GPU's HBM once: a 2 GB chunk costs one
     # H2D copy amortized over ~1000 batches instead of a per-batch unpinned
     # H2D stall (288 GB per GPU — whole-chunk residency is the design)
@@ -105,6 +122,8 @@ def ensemble_train_loop(ensemble, cfg, args, ensemble_name, sampler, dataset, pr
         batch = dataset[batch_idxs].to(args["device"])
         losses, aux = ensemble.step_batch(batch)
         n_acts += batch.shape[0]
+        if resampler is not None:
+            resampler.observe(batch, aux)
 
         if logger is not None and i % getattr(cfg, "log_every", 10) == 0:
             num_nonzero = aux["c"].count_nonzero(dim=-1).float().mean(dim=-1)
@@ -122,6 +141,11 @@ def ensemble_train_loop(ensemble, cfg, args, ensemble_name, sampler, dataset, pr
             logger.log(log)
 
         progress_counter.value = i
+
+    if resampler is not None and (getattr(cfg, "_chunk_i", 0) + 1) % resample_every == 0:
+        counts = resampler.resample()
+        if logger is not None:
+            logger.log({f"{ensemble_name}_resampled": int(sum(counts))})
 
     if device.type == "cuda":
         torch.cuda.synchronize(device)
@@ -335,6 +359,7 @@ def sweep(ensemble_init_func, cfg):
         for ensemble, arg, _ in ensembles:
             learned_dicts.extend(unstacked_to_learned_dicts(ensemble, arg, ensemble_hyperparams, buffer_hyperparams))
     for i, chunk_idx in enumerate(chunk_order):
+        cfg._chunk_i = i  # read by ensemble_train_loop's resample schedule
         if i < start_chunk:
             continue
         print(f"Chunk {i + 1}/{len(chunk_order)}")

@@ -213,3 +213,32 @@ def test_big_sweep_experiments_cli(tmp_path):
         capture_output=True, text=True, cwd=root, timeout=280)
     assert r.returncode == 0, r.stderr[-800:]
     assert os.path.exists(tmp_path / "out" / "_1" / "learned_dicts.pt")
+
+
+def test_sweep_with_resampling(tmp_path):
+    """cfg.resample_every_chunks wires the anthropic resampler into the
+    dispatched train loop: dead features are rewritten mid-sweep and the
+    rewrite is visible through the shared-memory params."""
+    cfg = _mini_cfg(tmp_path, 2)
+    cfg.resample_every_chunks = 1
+    cfg.resample_n_track = 8
+    cfg.resample_warmup_steps = 4
+
+    def init_func(c):
+        out = make_grid_ensembles(c, FunctionalTiedSAE, [1e-3], [1.0], devices=["cpu"])
+        for ens, _, _ in out[0]:
+            with torch.no_grad():
+                ens.params["encoder_bias"][:, :8] = -1e6  # guaranteed dead
+        return out
+
+    dicts = big_sweep.sweep(init_func, cfg)
+    (ld, _), = dicts
+    # the poisoned features were resampled: bias reset from -1e6
+    assert (ld.encoder_bias[:8] > -1e5).all()
+    import json
+
+    log_path = os.path.join(cfg.output_folder, "metrics.jsonl")
+    if os.path.exists(log_path):
+        logged = [json.loads(l) for l in open(log_path)]
+        assert any(any(k.endswith("_resampled") and v > 0 for k, v in row.items())
+                   for row in logged)
