@@ -64,6 +64,7 @@ def sharded_sweep(init_func_for_rank: Callable, cfg):
 
     learned_dicts = None
     for i, chunk_idx in enumerate(chunk_order):
+        cfg._chunk_i = i  # resample schedule (big_sweep.ensemble_train_loop)
         chunk = None
         if rank == 0:
             chunk = torch.load(os.path.join(cfg.dataset_folder, f"{chunk_idx}.pt"),
@@ -164,6 +165,7 @@ def generator_trainer_sweep(init_func_for_rank: Callable, cfg,
     n_steps = int(getattr(cfg, "n_chunks", 4)) * int(getattr(cfg, "n_repetitions", 1) or 1)
     learned_dicts = None
     for i in range(n_steps):
+        cfg._chunk_i = i  # resample schedule (big_sweep.ensemble_train_loop)
         if rank == 0:
             chunk = next(gen)
             feeder.feed(chunk)
