@@ -101,6 +101,11 @@ class EnsembleArgs(TrainArgs):
     activation_width: int = 512
     use_synthetic_dataset: bool = False
     bias_decay: float = 0.0
+    # in-sweep dead-feature resampling (engine/resample.py); 0 = off
+    resample_every_chunks: int = 0
+    resample_protocol: str = "anthropic"
+    resample_n_track: int = 512
+    resample_warmup_steps: int = 1000
 
 
 @dataclass
