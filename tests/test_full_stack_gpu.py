@@ -759,3 +759,44 @@ def test_rs_ag_respects_lr_mult():
     torch.cuda.synchronize()
     assert torch.equal(ens.params["encoder"][:, : n // 2], before)
     assert not torch.equal(ens.params["encoder"][:, n // 2 :], before_live)  # others moved
+
+
+def test_sweep_with_resampling_gpu(tmp_path):
+    """In-sweep anthropic resampling through the dispatched FUSED path:
+    k_resample + lr_mult run inside the spawned child against shared CUDA
+    tensors; the rewrite must be visible to the parent."""
+    from sparse_coding_amd.config import SyntheticEnsembleArgs
+    from sparse_coding_amd.models.sae_signatures import FunctionalTiedSAE
+    from sparse_coding_amd.sweep import big_sweep
+    from sparse_coding_amd.sweep.experiments import make_grid_ensembles
+
+    cfg = SyntheticEnsembleArgs()
+    cfg.use_synthetic_dataset = True
+    cfg.activation_width = 128
+    cfg.n_ground_truth_components = 256
+    cfg.gen_batch_size = 512
+    cfg.feature_num_nonzero = 8
+    cfg.noise_magnitude_scale = 0.0
+    cfg.chunk_size_gb = 128 * 512 * 4 * 2 / 1024**3
+    cfg.n_chunks = 2
+    cfg.batch_size = 512
+    cfg.device = DEV
+    cfg.dataset_folder = str(tmp_path / "data")
+    cfg.output_folder = str(tmp_path / "out")
+    cfg.use_wandb = False
+    cfg.wandb_images = False
+    cfg.resample_every_chunks = 1
+    cfg.resample_n_track = 8
+    cfg.resample_warmup_steps = 4
+
+    def init_func(c):
+        out = make_grid_ensembles(c, FunctionalTiedSAE, [1e-3], [1.0], devices=[DEV])
+        for ens, _, _ in out[0]:
+            with torch.no_grad():
+                ens.params["encoder_bias"][:, :8] = -1e6  # guaranteed dead
+        return out
+
+    dicts = big_sweep.sweep(init_func, cfg)
+    (ld, _), = dicts
+    assert (ld.encoder_bias[:8] > -1e5).all()  # rewritten by k_resample
+    assert torch.isfinite(ld.get_learned_dict()).all()
