@@ -219,3 +219,25 @@ def plot_scores(results_by_name: Dict[str, List[float]], save_path: Optional[str
     if save_path:
         fig.savefig(save_path)
     return fig
+
+
+def fragment_table_to_dataframe(acts, tokens, feature_indices, tokenizer=None):
+    """The reference's activation-DataFrame schema (interpret.py:131-212 /
+    read at :271-283): one row per fragment with ``fragment_token_ids``,
+    ``fragment_token_strs`` and, per kept feature i,
+    ``feature_{i}_activation_{j}`` (j < fragment length) + ``feature_{i}_max``
+    columns.  The tensor table stays the native format; this converter gives
+    reference-analysis code the exact columns it indexes."""
+    import pandas as pd
+
+    n_frag, frag_len, n_feats = acts.shape
+    cols = {
+        "fragment_token_ids": [tokens[i].tolist() for i in range(n_frag)],
+        "fragment_token_strs": [_decode_tokens(tokenizer, tokens[i]) for i in range(n_frag)],
+    }
+    acts_np = acts.cpu().numpy()
+    for fi, feature in enumerate(feature_indices):
+        for j in range(frag_len):
+            cols[f"feature_{feature}_activation_{j}"] = acts_np[:, j, fi]
+        cols[f"feature_{feature}_max"] = acts_np[:, :, fi].max(axis=1)
+    return pd.DataFrame(cols)

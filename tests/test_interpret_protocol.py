@@ -215,3 +215,24 @@ def test_interpret_across_baselines(tmp_path):
     out = interpret_across_baselines(str(base), str(tmp_path / "res"), cfg, n_fragments=8)
     assert set(out) == {"l0_residual/pca.pt", "l1_residual/pca.pt"}
     assert os.path.isdir(tmp_path / "res" / "l0_residual" / "pca")
+
+
+def test_fragment_table_to_dataframe():
+    """Reference DF schema round-trip (interpret.py:131-212 columns)."""
+    from sparse_coding_amd.interpret.interpret import fragment_table_to_dataframe
+
+    acts, tokens = _toy_table(n_frag=20)
+    df = fragment_table_to_dataframe(acts, tokens, [0, 1, 2, 3])
+    assert len(df) == 20
+    assert "fragment_token_strs" in df.columns
+    for f in range(4):
+        assert f"feature_{f}_max" in df.columns
+        assert f"feature_{f}_activation_0" in df.columns
+    # max column consistent with per-position columns
+    import numpy as np
+
+    j_cols = [f"feature_0_activation_{j}" for j in range(acts.shape[1])]
+    assert np.allclose(df[j_cols].max(axis=1), df["feature_0_max"])
+    # the reference's sort-by-max record selection works on this schema
+    top = df.sort_values(by="feature_0_max", ascending=False).head(5)
+    assert top["feature_0_max"].iloc[0] == df["feature_0_max"].max()
