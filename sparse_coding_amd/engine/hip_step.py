@@ -1271,6 +1271,9 @@ class HipTopKStep:
         self._eager_steps = 0
 
     def _alloc(self, B):
+        from sparse_coding_amd.ops.kconfig import kernel_config
+
+        self.kc = kernel_config()
         M, n, d = self.n_models, self.n_dict, self.d_act
         dev = self.ens.params["dict"].device
         f = lambda *shape: torch.empty(shape, device=dev, dtype=torch.float32)
@@ -1299,19 +1302,23 @@ class HipTopKStep:
         self.loss_parts.zero_()
         self.g_bias_scratch.zero_()
 
+        kc = self.kc
+        bk, prio = kc["bk"], kc["prio"]
+        bk_dec = kc["bk_dec"] or bk
+        bk_gw = kc["bk_grad_w"] or bk
         ext.row_norms(W, self.norms, self.inv_norms, self.EPS)
         ext.enc_fwd(x, W, self.dummy_bias, self.inv_norms,
-                    self.scores, self.loss_parts, self.fired, 1)
+                    self.scores, self.loss_parts, self.fired, 1, bk, prio)
         # exact per-row radix top-k + scatter + fired counts in ONE kernel
         # (k_topk_select) — replaces the torch.topk/scatter chain
         ext.topk_select(self.scores, self.c, self.fired, self.ks_i32)
 
-        ext.dec_fwd(self.c, W, self.inv_norms, x, self.r, self.loss_parts)
+        ext.dec_fwd(self.c, W, self.inv_norms, x, self.r, self.loss_parts, bk_dec, prio)
         ext.gc(self.r, W, self.inv_norms, self.c, self.zero_l1,
-               self.gpre, self.g_bias_scratch)
+               self.gpre, self.g_bias_scratch, bk, prio)
         gscale = 2.0 / (B * self.d_act)
-        ext.grad_w(self.c, self.r, self.gw, gscale, 0.0)
-        ext.grad_w(self.gpre, x, self.gw, 1.0, 1.0)
+        ext.grad_w(self.c, self.r, self.gw, gscale, 0.0, bk_gw, prio)
+        ext.grad_w(self.gpre, x, self.gw, 1.0, 1.0, bk_gw, prio)
         return B
 
     def update_phase(self, B):
