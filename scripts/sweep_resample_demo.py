@@ -42,6 +42,9 @@ def main():
     cfg.n_ground_truth_components = 8192
     cfg.gen_batch_size = 4096
     cfg.feature_num_nonzero = 40
+    # flat inclusion probability: the default 0.99 decay over 8192
+    # components renormalizes early-feature probs past 1 (degenerate data)
+    cfg.feature_prob_decay = 1.0
     cfg.noise_magnitude_scale = 0.0
     cfg.chunk_size_gb = args.chunk_gb
     cfg.n_chunks = args.n_chunks
