@@ -236,3 +236,45 @@ def test_fragment_table_to_dataframe():
     # the reference's sort-by-max record selection works on this schema
     top = df.sort_values(by="feature_0_max", ascending=False).head(5)
     assert top["feature_0_max"].iloc[0] == df["feature_0_max"].max()
+
+
+class _FakeBatch(dict):
+    def to(self, device):
+        return self
+
+
+class _FakeTok:
+    """Minimal HF-tokenizer surface for HFLocalClient: batch __call__ with
+    padding, pad/eos tokens, decode."""
+
+    eos_token = "<eos>"
+    pad_token = None
+    pad_token_id = 0
+
+    def __call__(self, texts, return_tensors=None, padding=True,
+                 truncation=True, max_length=64):
+        ids = [[(hash(w) % 400) + 2 for w in t.split()[:max_length]] for t in texts]
+        L = max(len(x) for x in ids)
+        input_ids = torch.tensor([x + [self.pad_token_id] * (L - len(x)) for x in ids])
+        return _FakeBatch(input_ids=input_ids,
+                          attention_mask=(input_ids != self.pad_token_id).long())
+
+    def decode(self, ids, skip_special_tokens=True):
+        return " ".join(f"<{int(t)}>" for t in ids if int(t) != self.pad_token_id)
+
+
+@pytest.mark.timeout(300)
+def test_hf_local_client_batches():
+    """HFLocalClient: batched greedy generation through a local causal LM,
+    one response per prompt, in order."""
+    from sparse_coding_amd.data.activation_dataset import load_model
+    from sparse_coding_amd.interpret.protocol import HFLocalClient
+
+    torch.manual_seed(0)
+    model = load_model("tiny-gptneox", device="cpu")
+    client = HFLocalClient(model, _FakeTok(), device="cpu",
+                           max_new_tokens=4, batch_size=2)
+    prompts = [f"prompt number {i} with some tokens" for i in range(5)]
+    outs = client.batch_complete(prompts)
+    assert len(outs) == 5
+    assert all(isinstance(o, str) for o in outs)
