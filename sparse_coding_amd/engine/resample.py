@@ -71,7 +71,16 @@ class EnsembleResampler:
         self.dead_threshold = dead_threshold
         self.examples_seen = 0
         # TopK ensembles name their weight "dict"; SAEs use "encoder"
-        self.w_key = "encoder" if "encoder" in ensemble.params else "dict"
+        if "encoder" in ensemble.params:
+            self.w_key = "encoder"
+        elif "dict" in ensemble.params:
+            self.w_key = "dict"
+        else:
+            raise ValueError(
+                "EnsembleResampler supports single-matrix encoders (tied/untied "
+                "SAEs, TopK); this ensemble's params "
+                f"({sorted(ensemble.params)}) have no 'encoder'/'dict' weight "
+                "(multi-layer encoders have no per-feature row to resample)")
         M, n, d = ensemble.params[self.w_key].shape
         dev = ensemble.params[self.w_key].device
         self.fired = torch.zeros(M, n, device=dev)

@@ -109,3 +109,13 @@ def test_replaced_mask_matches_rule():
     assert mask[0, [1, 2, 4, 5]].all()
     assert not mask[0, [0, 3, 6, 7]].any() or not mask[0, [6, 7]].any()
     assert mask.sum() == 4
+
+
+def test_resampler_rejects_multilayer_encoders():
+    from sparse_coding_amd.models.lista import FunctionalLISTADenoisingSAE
+
+    models = [FunctionalLISTADenoisingSAE.init(16, 32, 2, 1e-3)]
+    ens = FunctionalEnsemble(models, FunctionalLISTADenoisingSAE, adam, {"lr": 1e-3},
+                             backend="torch")
+    with pytest.raises(ValueError, match="no 'encoder'/'dict'"):
+        EnsembleResampler(ens)
