@@ -33,7 +33,9 @@
 //   k_gc_t/k_gc2_t: gpre = (c>0) .* (gs * r @ Wdec_hat^T + l1/B) (+ bias-grad)
 //   k_grad_w_t   : gw = beta*gw + alpha * P^T @ Q   (K = batch contraction)
 //   k_row_norms, k_project_adam (gradient of w/max(||w||,eps)), k_bias_adam,
-//   k_transpose_scale
+//   k_transpose_scale, k_colsum (HBM-rate column sums / L1 reductions),
+//   k_topk_select (exact per-row radix top-k), k_resample (K14 on-device),
+//   k_lista_bwd_elem (one-pass LISTA backward elementwise chain)
 
 #include <hip/hip_runtime.h>
 
